@@ -1,0 +1,295 @@
+"""DTensor core tests on CPU/gloo world_size=2.
+
+Mirrors the reference test pyramid tier 1: distribute /
+redistribute / op parity vs a single-device reference.
+"""
+import pytest
+import torch
+import torch.nn.functional as F
+
+from tests.common import spawn
+
+from vescale_amd import (
+    DTensor,
+    InterleavedShard,
+    Partial,
+    RaggedShard,
+    Replicate,
+    Shard,
+    distribute_tensor,
+    init_device_mesh,
+)
+
+
+# ---------------------------------------------------------------------------
+# pure placement math (no comm)
+# ---------------------------------------------------------------------------
+def test_shard_chunk_math():
+    assert Shard.chunk_size(10, 4, 0) == 3
+    assert Shard.chunk_size(10, 4, 3) == 1
+    assert [Shard.chunk_size(10, 4, i) for i in range(4)] == [3, 3, 3, 1]
+    assert [Shard.chunk_offset(10, 4, i) for i in range(4)] == [0, 3, 6, 9]
+    # chunk semantics match torch.chunk
+    t = torch.arange(10)
+    chunks = Shard(0).split_tensor(t, 4)
+    ref = list(torch.chunk(t, 4))
+    for c, r in zip(chunks, ref):
+        assert torch.equal(c, r)
+
+
+def test_interleaved_shard_split():
+    t = torch.arange(12)
+    p = InterleavedShard(0, 2)
+    chunks = p.split_tensor(t, 3)
+    # view as [2,6], shard cols into 3 -> rank0 gets cols 0,1 of each row
+    assert torch.equal(chunks[0], torch.tensor([0, 1, 6, 7]))
+    assert torch.equal(chunks[2], torch.tensor([4, 5, 10, 11]))
+
+
+def test_ragged_shard_split():
+    t = torch.arange(24).reshape(6, 4)
+    p = RaggedShard((0,), (1, 2))  # units over flattened leading dim
+    chunks = p.split_tensor(t, 2)
+    assert chunks[0].numel() == 8 and chunks[1].numel() == 16
+    assert torch.equal(chunks[0], torch.arange(8))
+    rec = RaggedShard.reconstruct(torch.cat(chunks), (6, 4))
+    assert torch.equal(rec, t)
+
+
+# ---------------------------------------------------------------------------
+# ws=2 distributed behavior
+# ---------------------------------------------------------------------------
+def _t_distribute(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(5)
+    g = torch.randn(9, 4)  # uneven on purpose
+    d = distribute_tensor(g, mesh, [Shard(0)])
+    assert d.shape == (9, 4)
+    expect = torch.chunk(g, ws)[rank] if rank < ws else None
+    assert torch.equal(d._local_tensor, expect)
+    full = d.full_tensor()
+    assert torch.equal(full, g)
+
+
+def test_distribute_uneven():
+    spawn(2, _t_distribute)
+
+
+def _t_redistribute_matrix(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(7)
+    g = torch.randn(8, 6)
+    for src in ([Shard(0)], [Shard(1)], [Replicate()]):
+        for dst in ([Shard(0)], [Shard(1)], [Replicate()]):
+            d = distribute_tensor(g, mesh, src)
+            r = d.redistribute(placements=dst)
+            assert torch.equal(r.full_tensor(), g), f"{src}->{dst}"
+    # Partial -> Replicate / Shard
+    local = torch.full((4, 4), float(rank + 1))
+    d = DTensor.from_local(local, mesh, [Partial()], shape=torch.Size((4, 4)))
+    rep = d.redistribute(placements=[Replicate()])
+    assert torch.equal(rep._local_tensor, torch.full((4, 4), 3.0))
+    d = DTensor.from_local(local, mesh, [Partial()], shape=torch.Size((4, 4)))
+    sh = d.redistribute(placements=[Shard(0)])
+    assert sh._local_tensor.shape == (2, 4)
+    assert torch.equal(sh.full_tensor(), torch.full((4, 4), 3.0))
+
+
+def test_redistribute_matrix():
+    spawn(2, _t_redistribute_matrix)
+
+
+def _t_redistribute_uneven(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    g = torch.arange(7 * 3, dtype=torch.float32).reshape(7, 3)
+    d = distribute_tensor(g, mesh, [Shard(0)])
+    r = d.redistribute(placements=[Replicate()])
+    assert torch.equal(r._local_tensor, g)
+    r2 = r.redistribute(placements=[Shard(1)])
+    assert torch.equal(r2.full_tensor(), g)
+
+
+def test_redistribute_uneven():
+    spawn(2, _t_redistribute_uneven)
+
+
+def _t_ragged_redistribute(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    g = torch.arange(12, dtype=torch.float32)
+    p = RaggedShard((0,), (1, 3))
+    d = distribute_tensor(g, mesh, [p])
+    assert d._local_tensor.numel() == (3 if rank == 0 else 9)
+    # RS -> R
+    full = d.full_tensor()
+    assert torch.equal(full, g)
+    # RS -> RS' (interval-intersection all_to_all)
+    p2 = RaggedShard((0,), (2, 2))
+    d2 = d.redistribute(placements=[p2])
+    assert d2._local_tensor.numel() == 6
+    assert torch.equal(d2.full_tensor(), g)
+
+
+def test_ragged_redistribute():
+    spawn(2, _t_ragged_redistribute)
+
+
+def _t_mm_parity(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(11)
+    a = torch.randn(8, 6)
+    b = torch.randn(6, 4)
+    ref = a @ b
+    for pa in ([Shard(0)], [Shard(1)], [Replicate()]):
+        for pb in ([Shard(0)], [Shard(1)], [Replicate()]):
+            da = distribute_tensor(a, mesh, pa)
+            db = distribute_tensor(b, mesh, pb)
+            dc = da @ db
+            assert torch.allclose(dc.full_tensor(), ref, atol=1e-5), f"{pa}x{pb}"
+
+
+def test_mm_parity():
+    spawn(2, _t_mm_parity)
+
+
+def _t_linear_training_step(rank, ws):
+    """2-layer MLP with Shard(0)/Shard(1) weights (colwise->rowwise TP) —
+    the BASELINE.json plumbing config."""
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(3)
+    x_g = torch.randn(4, 8)
+    w1_g = torch.randn(16, 8)
+    w2_g = torch.randn(8, 16)
+    # reference
+    xr = x_g.clone().requires_grad_(True)
+    w1r = w1_g.clone().requires_grad_(True)
+    w2r = w2_g.clone().requires_grad_(True)
+    loss_r = F.linear(F.relu(F.linear(xr, w1r)), w2r).pow(2).sum()
+    loss_r.backward()
+
+    x = distribute_tensor(x_g, mesh, [Replicate()]).requires_grad_(True)
+    w1 = distribute_tensor(w1_g, mesh, [Shard(0)]).requires_grad_(True)
+    w2 = distribute_tensor(w2_g, mesh, [Shard(1)]).requires_grad_(True)
+    h = F.relu(F.linear(x, w1))
+    out = F.linear(h, w2)
+    loss = out.pow(2).sum()
+    lf = loss.redistribute(placements=[Replicate()])
+    assert torch.allclose(lf.to_local(), loss_r.detach(), atol=1e-4)
+    loss.backward()
+    gw1 = w1.grad.full_tensor() if isinstance(w1.grad, DTensor) else w1.grad
+    assert torch.allclose(gw1, w1r.grad, atol=1e-4)
+    gw2 = w2.grad.full_tensor() if isinstance(w2.grad, DTensor) else w2.grad
+    assert torch.allclose(gw2, w2r.grad, atol=1e-4)
+
+
+def test_linear_training_step():
+    spawn(2, _t_linear_training_step)
+
+
+def _t_reductions(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(13)
+    g = torch.randn(8, 6)
+    d = distribute_tensor(g, mesh, [Shard(0)])
+    checks = [
+        (d.sum(), g.sum()),
+        (d.mean(), g.mean()),
+        (d.sum(dim=0), g.sum(dim=0)),
+        (d.sum(dim=1), g.sum(dim=1)),
+        (d.amax(), g.amax()),
+        (d.pow(2).sum(), g.pow(2).sum()),
+    ]
+    for got, want in checks:
+        gf = got.redistribute(placements=[Replicate()]).to_local()
+        assert torch.allclose(gf, want, atol=1e-5)
+    n = torch.linalg.vector_norm(d)
+    assert torch.allclose(n.to_local(), torch.linalg.vector_norm(g), atol=1e-5)
+
+
+def test_reductions():
+    spawn(2, _t_reductions)
+
+
+def _t_embedding_vocab_parallel(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(17)
+    w = torch.randn(10, 4)
+    idx = torch.randint(0, 10, (3, 5))
+    ref = F.embedding(idx, w)
+    dw = distribute_tensor(w, mesh, [Shard(0)])
+    didx = distribute_tensor(idx, mesh, [Replicate()])
+    out = F.embedding(didx, dw)
+    assert any(p.is_partial() for p in out.placements)
+    assert torch.allclose(out.full_tensor(), ref, atol=1e-5)
+
+
+def test_embedding_vocab_parallel():
+    spawn(2, _t_embedding_vocab_parallel)
+
+
+def _t_cross_entropy_batch_sharded(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(19)
+    logits = torch.randn(8, 12)
+    tgt = torch.randint(0, 12, (8,))
+    ref = F.cross_entropy(logits, tgt)
+    dl = distribute_tensor(logits, mesh, [Shard(0)]).requires_grad_(True)
+    dt = distribute_tensor(tgt, mesh, [Shard(0)])
+    loss = F.cross_entropy(dl, dt)
+    lv = loss.redistribute(placements=[Replicate()]).to_local()
+    assert torch.allclose(lv, ref, atol=1e-5)
+    loss.backward()
+    logits_r = logits.clone().requires_grad_(True)
+    F.cross_entropy(logits_r, tgt).backward()
+    assert torch.allclose(dl.grad.full_tensor(), logits_r.grad, atol=1e-5)
+
+
+def test_cross_entropy_batch_sharded():
+    spawn(2, _t_cross_entropy_batch_sharded)
+
+
+def _t_2d_mesh(rank, ws):
+    mesh = init_device_mesh("cpu", (2, 2), mesh_dim_names=("DP", "TP"))
+    torch.manual_seed(23)
+    g = torch.randn(8, 8)
+    d = distribute_tensor(g, mesh, [Shard(0), Shard(1)])
+    coord = mesh.get_coordinate()
+    assert d._local_tensor.shape == (4, 4)
+    assert torch.equal(d.full_tensor(), g)
+    r = d.redistribute(placements=[Replicate(), Shard(0)])
+    assert torch.equal(r.full_tensor(), g)
+
+
+@pytest.mark.skipif(False, reason="")
+def test_2d_mesh():
+    spawn(4, _t_2d_mesh)
+
+
+def _t_submesh(rank, ws):
+    mesh = init_device_mesh("cpu", (2, 2), mesh_dim_names=("DP", "TP"))
+    tp = mesh["TP"]
+    assert tp.ndim == 1 and tp.size() == 2
+    dp = mesh["DP"]
+    assert dp.size() == 2
+    g = torch.randn(4, 4)
+    d = distribute_tensor(g, tp, [Shard(0)])
+    assert torch.equal(d.full_tensor(), g)
+
+
+def test_submesh():
+    spawn(4, _t_submesh)
+
+
+def _t_softmax_dropout(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(29)
+    g = torch.randn(6, 8)
+    d = distribute_tensor(g, mesh, [Shard(0)])
+    s = torch.softmax(d, dim=-1)
+    assert torch.allclose(s.full_tensor(), torch.softmax(g, dim=-1), atol=1e-6)
+    # dropout p=0 keeps values
+    o = F.dropout(d, p=0.0, training=True)
+    assert torch.allclose(o.full_tensor(), g)
+
+
+def test_softmax_dropout():
+    spawn(2, _t_softmax_dropout)
