@@ -1,0 +1,147 @@
+"""Flagship benchmark — Llama-3-8B veScale-FSDP-style training step on
+MI355X (BASELINE.json metric: tokens/sec whole node at 1/2/4/8 GPUs).
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+  (N>1 launched via torch.distributed.run, one rank per GPU over RCCL)
+
+Synthetic data (random tokens), random-init weights, bf16 compute, weak
+scaling (per-GPU batch fixed).  Times exactly K steps bracketed by
+barrier + torch.cuda.synchronize on both sides; MAX over ranks; rank 0
+prints ONE JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--model", type=str, default="llama3_8b",
+                    choices=["llama3_8b", "llama3_70b", "llama_tiny"])
+    ap.add_argument("--batch", type=int, default=2, help="per-GPU batch size")
+    ap.add_argument("--seq", type=int, default=8192)
+    ap.add_argument("--lr", type=float, default=3e-4)
+    ap.add_argument("--activation-checkpointing", action="store_true")
+    ap.add_argument("--no-master-weights", action="store_true")
+    args = ap.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(world_size, 1)
+
+    on_gpu = torch.cuda.is_available()
+    device = torch.device(f"cuda:{local_rank}") if on_gpu else torch.device("cpu")
+    if on_gpu:
+        torch.cuda.set_device(device)
+
+    if world_size > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group("nccl" if on_gpu else "gloo")
+
+    from vescale_amd.dtensor import init_device_mesh
+    from vescale_amd.fsdp import FSDP, FlatAdamW
+    from vescale_amd.models import llama as M
+
+    cfg = getattr(M, args.model)() if args.model != "llama_tiny" else M.llama_tiny()
+    if not on_gpu and args.model != "llama_tiny":
+        # CPU smoke: shrink so the default invocation finishes in minutes
+        cfg = M.llama_tiny()
+        args.batch, args.seq = 2, 64
+
+    torch.manual_seed(1234)
+    mesh = init_device_mesh(device.type, (world_size,), mesh_dim_names=("DP",)) if world_size > 1 else None
+
+    with torch.device("meta"):
+        model = M.LlamaModel(cfg)
+    model = model.to_empty(device=device)
+    model.rope_table.copy_(
+        M.build_rope_table(cfg.max_seq_len, cfg.head_dim, cfg.rope_theta).to(device)
+    )
+    model.init_weights()
+    model.activation_checkpointing = args.activation_checkpointing
+    dtype = torch.bfloat16 if on_gpu else torch.float32
+
+    if world_size > 1:
+        eng = FSDP(model, mesh, param_dtype=dtype, device=device)
+    else:
+        eng = FSDP(model, None, param_dtype=dtype, device=device)
+    opt = FlatAdamW(eng, lr=args.lr, grad_clip=1.0,
+                    use_master_weights=not args.no_master_weights)
+
+    B, S = args.batch, args.seq
+    gen = torch.Generator(device="cpu").manual_seed(4321 + rank)
+    x = torch.randint(0, cfg.vocab_size, (B, S), generator=gen).to(device)
+    y = torch.roll(x, -1, dims=1)
+
+    def step():
+        loss = eng(x, y)
+        loss.backward()
+        opt.step()
+        return loss
+
+    for _ in range(args.warmup):
+        step()
+
+    if world_size > 1:
+        dist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = step()
+    if on_gpu:
+        torch.cuda.synchronize()
+    if world_size > 1:
+        dist.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64)
+    if world_size > 1:
+        dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
+    sec = float(elapsed.item())
+    ms_per_step = sec / args.steps * 1e3
+    tokens_per_step = B * S * n_gpus
+    tok_s = tokens_per_step * args.steps / sec
+
+    if rank == 0:
+        out = {
+            "metric": "tokens/sec (whole node) Llama-3-8B veScale-FSDP",
+            "value": tok_s,
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if on_gpu else "fp32-cpu-smoke",
+            "data": "synthetic",
+            "config": {
+                "model": args.model if on_gpu else "llama_tiny(cpu-smoke)",
+                "global_batch": B * n_gpus,
+                "seq_len": S,
+                "parallelism": f"fsdp{n_gpus}",
+                "final_loss": float(loss.detach().float().cpu()),
+                "activation_checkpointing": args.activation_checkpointing,
+            },
+        }
+        print(json.dumps(out))
+
+    if world_size > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

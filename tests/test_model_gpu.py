@@ -1,0 +1,53 @@
+"""GPU model tests: tiny-Llama training step on MI355X (bf16, HIP kernels)
+vs CPU fp32 reference; FSDP ws=1 engine on GPU."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+
+
+@requires_gpu
+def test_llama_tiny_forward_parity():
+    from vescale_amd.models.llama import LlamaModel, llama_tiny
+
+    torch.manual_seed(0)
+    cfg = llama_tiny(vocab=512, seq=128)
+    model = LlamaModel(cfg)
+    model.init_weights()
+    x = torch.randint(0, cfg.vocab_size, (2, 64))
+    y = torch.roll(x, -1, dims=1)
+    loss_cpu = model(x, y)  # fp32 CPU reference path
+
+    gm = LlamaModel(cfg)
+    gm.load_state_dict(model.state_dict())
+    gm = gm.cuda().to(torch.bfloat16)
+    gm.rope_table.data = gm.rope_table.data.float()
+    loss_gpu = gm(x.cuda(), y.cuda())
+    assert abs(float(loss_gpu) - float(loss_cpu)) / float(loss_cpu) < 0.02, (
+        float(loss_gpu), float(loss_cpu),
+    )
+
+
+@requires_gpu
+def test_fsdp_gpu_train_decreases():
+    from vescale_amd.fsdp import FSDP, FlatAdamW
+    from vescale_amd.models.llama import LlamaModel, llama_tiny
+
+    torch.manual_seed(0)
+    cfg = llama_tiny(vocab=512, seq=128)
+    model = LlamaModel(cfg).cuda()
+    model.init_weights()
+    eng = FSDP(model, None, param_dtype=torch.bfloat16, device=torch.device("cuda"))
+    opt = FlatAdamW(eng, lr=1e-3, grad_clip=1.0)
+    x = torch.randint(0, cfg.vocab_size, (4, 128), device="cuda")
+    y = torch.roll(x, -1, dims=1)
+    losses = []
+    for _ in range(10):
+        loss = eng(x, y)
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] * 0.7, losses
+    assert all(l == l for l in losses), f"NaN in {losses}"

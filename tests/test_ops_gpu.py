@@ -1,0 +1,197 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference
+(SURVEY.md §4 — "numerics tests for a HIP kernel compare it against a plain
+PyTorch fp32 reference of the same op")."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+
+
+def _ext():
+    import vescale_amd.ops as ops
+
+    return ops.require_ext()
+
+
+@requires_gpu
+def test_ext_loaded():
+    import vescale_amd.ops as ops
+
+    assert ops.has_ext(), f"HIP extension must be present on GPU: {ops._import_error}"
+
+
+@requires_gpu
+def test_rmsnorm_fwd_bwd():
+    from vescale_amd.ops import rmsnorm
+
+    torch.manual_seed(0)
+    x = torch.randn(64, 4096, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(4096, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    out = rmsnorm(x, w, 1e-5)
+    dy = torch.randn_like(out)
+    out.backward(dy)
+
+    xf = x.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    rrms = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-5)
+    ref = xf * rrms * wf
+    ref.backward(dy.float())
+
+    assert torch.allclose(out.float(), ref.detach(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(x.grad.float(), xf.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(w.grad.float(), wf.grad, atol=1.0, rtol=2e-2)
+
+
+@requires_gpu
+def test_rope():
+    from vescale_amd.ops import build_rope_table, rope_apply
+    from vescale_amd.ops.functional import _rope_ref
+
+    torch.manual_seed(1)
+    tab = build_rope_table(128, 128, 500000.0, device="cuda")
+    x = torch.randn(2, 128, 8, 128, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    out = rope_apply(x, tab)
+    ref = _rope_ref(x.detach().float(), tab, 0, False)
+    assert torch.allclose(out.float(), ref, atol=2e-2, rtol=2e-2)
+    dy = torch.randn_like(out)
+    out.backward(dy)
+    ref_grad = _rope_ref(dy.float(), tab, 0, True)
+    assert torch.allclose(x.grad.float(), ref_grad, atol=2e-2, rtol=2e-2)
+
+
+@requires_gpu
+def test_swiglu():
+    from vescale_amd.ops import swiglu
+
+    torch.manual_seed(2)
+    g = torch.randn(256, 1024, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    u = torch.randn(256, 1024, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    out = swiglu(g, u)
+    dy = torch.randn_like(out)
+    out.backward(dy)
+    gf = g.detach().float().requires_grad_(True)
+    uf = u.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.silu(gf) * uf
+    ref.backward(dy.float())
+    assert torch.allclose(out.float(), ref.detach(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(g.grad.float(), gf.grad, atol=3e-2, rtol=3e-2)
+    assert torch.allclose(u.grad.float(), uf.grad, atol=3e-2, rtol=3e-2)
+
+
+@requires_gpu
+def test_fused_cross_entropy():
+    from vescale_amd.ops import fused_cross_entropy
+
+    torch.manual_seed(3)
+    N, V = 512, 128256
+    logits = torch.randn(N, V, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    tgt = torch.randint(0, V, (N,), device="cuda")
+    tgt[:7] = -100
+    loss = fused_cross_entropy(logits, tgt)
+    loss.backward()
+
+    lf = logits.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(lf, tgt, ignore_index=-100)
+    ref.backward()
+    assert torch.allclose(loss.float(), ref.detach(), atol=1e-2, rtol=1e-3), (float(loss), float(ref))
+    assert torch.allclose(logits.grad.float(), lf.grad, atol=1e-3, rtol=5e-2)
+
+
+@requires_gpu
+def test_adamw_flat():
+    from vescale_amd.ops import adamw_step_flat
+
+    torch.manual_seed(4)
+    n = 4096 * 9 + 17
+    p32 = torch.randn(n, device="cuda")
+    p = p32.bfloat16().clone()
+    master = p.float().clone()
+    g = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    m = torch.zeros(n, device="cuda")
+    v = torch.zeros(n, device="cuda")
+    for step in range(1, 4):
+        adamw_step_flat(p, master, g, m, v, lr=1e-3, beta1=0.9, beta2=0.95,
+                        eps=1e-8, weight_decay=0.1, step=step)
+    # fp32 torch reference
+    pr = p32.bfloat16().float().clone()
+    mr = torch.zeros(n, device="cuda")
+    vr = torch.zeros(n, device="cuda")
+    gf = g.float()
+    for step in range(1, 4):
+        pr.mul_(1 - 1e-3 * 0.1)
+        mr.mul_(0.9).add_(gf, alpha=0.1)
+        vr.mul_(0.95).addcmul_(gf, gf, value=0.05)
+        bc1 = 1 - 0.9**step
+        bc2 = 1 - 0.95**step
+        pr.addcdiv_(mr, (vr / bc2).sqrt().add_(1e-8), value=-1e-3 / bc1)
+    assert torch.allclose(master, pr, atol=1e-5, rtol=1e-4)
+    assert torch.allclose(m, mr, atol=1e-6)
+    assert torch.allclose(v, vr, atol=1e-7)
+
+
+@requires_gpu
+def test_l2norm_and_scale():
+    from vescale_amd.ops import l2norm_sq, scale_flat_
+
+    torch.manual_seed(5)
+    x = torch.randn(123457, device="cuda", dtype=torch.bfloat16)
+    got = l2norm_sq(x)
+    ref = x.float().pow(2).sum()
+    assert torch.allclose(got, ref, rtol=1e-3)
+    y = x.clone()
+    scale_flat_(y, 0.5)
+    assert torch.allclose(y.float(), (x.float() * 0.5), atol=1e-2)
+
+
+@requires_gpu
+def test_philox_shard_parity():
+    """The headline RNG property: a sharded fill is bitwise-identical to the
+    single-GPU fill (reference patch #2 semantics)."""
+    C = _ext()
+    n = 1 << 16
+    seed, off = 1234, 7
+    full = torch.empty(n, device="cuda", dtype=torch.bfloat16)
+    C.philox_uniform_(full, [n], [n], [0], 0, True, seed, off, 0.0, 1.0)
+    # shard = second quarter, filled independently with flat offset
+    q = n // 4
+    shard = torch.empty(q, device="cuda", dtype=torch.bfloat16)
+    C.philox_uniform_(shard, [n], [q], [q], q, True, seed, off, 0.0, 1.0)
+    assert torch.equal(shard, full[q : 2 * q])
+    # 2-D sharding: rows [8:16) of a [32, 64] matrix
+    g = torch.empty(32 * 64, device="cuda", dtype=torch.bfloat16)
+    C.philox_uniform_(g, [32 * 64], [32 * 64], [0], 0, True, seed, off, 0.0, 1.0)
+    g = g.view(32, 64)
+    sh = torch.empty(8, 64, device="cuda", dtype=torch.bfloat16)
+    C.philox_uniform_(sh.view(-1), [32, 64], [8, 64], [8, 0], 0, False, seed, off, 0.0, 1.0)
+    assert torch.equal(sh, g[8:16])
+    # column shard: cols [16:32)
+    sc = torch.empty(32, 16, device="cuda", dtype=torch.bfloat16)
+    C.philox_uniform_(sc.view(-1), [32, 64], [32, 16], [0, 16], 0, False, seed, off, 0.0, 1.0)
+    assert torch.equal(sc, g[:, 16:32])
+
+
+@requires_gpu
+def test_philox_normal_stats():
+    C = _ext()
+    n = 1 << 22
+    out = torch.empty(n, device="cuda", dtype=torch.float32)
+    C.philox_normal_(out, [n], [n], [0], 0, True, 99, 0, 0.0, 1.0)
+    assert abs(out.mean().item()) < 5e-3
+    assert abs(out.std().item() - 1.0) < 5e-3
+
+
+@requires_gpu
+def test_philox_dropout():
+    C = _ext()
+    torch.manual_seed(6)
+    x = torch.ones(1 << 20, device="cuda", dtype=torch.bfloat16)
+    out, mask = C.philox_dropout(x, [x.numel()], [x.numel()], [0], 0, True, 7, 0, 0.1, True)
+    keep_frac = mask.float().mean().item()
+    assert abs(keep_frac - 0.9) < 5e-3
+    # kept elements scaled by 1/(1-p)
+    kept = out[mask.bool()]
+    assert torch.allclose(kept.float(), torch.full_like(kept.float(), 1.0 / 0.9), atol=1e-2)
+    dropped = out[~mask.bool()]
+    assert (dropped == 0).all()

@@ -1,0 +1,481 @@
+"""veScale-FSDP-style engine, MI355X-native.
+
+Design (BASELINE.json north star; reference: the RaggedShard DTensor core
+in vescale/ + docs/texts/raggedshard.md — the FSDP wrapper itself is not
+open-sourced, so this engine is a from-scratch design):
+
+  - parameters are grouped into UNITS (one per transformer block + one root
+    unit); each unit's params live in ONE flat bf16 buffer, 256-element
+    aligned per param, padded so the flat extent divides evenly by the DP
+    world — each rank owns a contiguous flat shard (RaggedShard layout with
+    equal units; exposed as DTensor(RaggedShard) for checkpointing).
+  - unshard = ONE all_gather_into_tensor per unit on a dedicated HIP
+    all-gather stream, prefetched one unit ahead of compute (xGMI note:
+    a single fused large all-gather saturates the 7 p2p links; many small
+    per-param collectives would be latency-bound).
+  - gradients accumulate in-place into a lazily-allocated flat bf16 grad
+    buffer (param.grad pre-pointed at views — no copy); when a unit's last
+    grad lands, ONE reduce_scatter_tensor runs on a dedicated
+    reduce-scatter stream (separate RCCL communicator so AG/RS overlap
+    without cross-rank ordering hazards).
+  - the optimizer (FlatAdamW) steps on flat shards with the fused CDNA4
+    AdamW kernel; fp32 master shards optional.
+  - world_size == 1 degenerates to flat-buffer training with zero comm.
+"""
+from __future__ import annotations
+
+import logging
+from typing import Callable, Dict, List, Optional, Sequence, Tuple, Type, Union
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..dtensor.device_mesh import DeviceMesh
+from ..dtensor.dtensor import DTensor
+from ..dtensor.placement_types import RaggedShard, TensorMeta
+from ..dtensor._dtensor_spec import DTensorSpec
+
+logger = logging.getLogger(__name__)
+
+ALIGN = 256  # per-param alignment in elements (vector/bank friendly)
+
+
+def _all_gather_flat(out: torch.Tensor, shard: torch.Tensor, pg) -> None:
+    """all_gather_into_tensor with a gloo-safe fallback (CPU tests)."""
+    try:
+        dist.all_gather_into_tensor(out, shard, group=pg)
+    except RuntimeError:
+        ws = dist.get_world_size(pg)
+        bufs = list(out.chunk(ws))
+        dist.all_gather(bufs, shard.contiguous(), group=pg)
+
+
+def _align(n: int, a: int = ALIGN) -> int:
+    return (n + a - 1) // a * a
+
+
+class FSDPUnit:
+    """One flat-parameter group."""
+
+    def __init__(
+        self,
+        name: str,
+        module: nn.Module,
+        params: List[Tuple[str, nn.Parameter]],
+        world_size: int,
+        rank: int,
+        device: torch.device,
+        param_dtype: torch.dtype = torch.bfloat16,
+    ):
+        self.name = name
+        self.module = module
+        self.world_size = world_size
+        self.rank = rank
+        self.device = device
+        self.param_dtype = param_dtype
+
+        self.param_infos: List[Tuple[str, torch.Size, int, int]] = []  # (fqn, shape, offset, numel)
+        off = 0
+        for fqn, p in params:
+            n = p.numel()
+            self.param_infos.append((fqn, p.shape, off, n))
+            off += _align(n)
+        self.flat_numel = _align(off, ALIGN * world_size)
+        self.shard_numel = self.flat_numel // world_size
+        self.shard_off = self.rank * self.shard_numel
+
+        # build the shard from current param values
+        full = torch.zeros(self.flat_numel, dtype=param_dtype, device=device)
+        for (fqn, shape, o, n), (_, p) in zip(self.param_infos, params):
+            full[o : o + n].copy_(p.detach().reshape(-1).to(param_dtype))
+        if world_size > 1:
+            self.shard = full[self.shard_off : self.shard_off + self.shard_numel].clone()
+            del full
+            self.full: Optional[torch.Tensor] = None
+        else:
+            self.shard = full  # shard IS the full buffer
+            self.full = full
+
+        self.params: List[nn.Parameter] = [p for _, p in params]
+        self.grad_full: Optional[torch.Tensor] = None
+        self.grad_shard: Optional[torch.Tensor] = None
+        self._grads_ready = 0
+        self._ag_event: Optional[torch.cuda.Event] = None
+        self._rs_work = None
+        self._is_unsharded = False
+        if world_size == 1:
+            self._attach_param_views()
+            self._is_unsharded = True
+
+    # ------------------------------------------------------------------
+    def _attach_param_views(self):
+        assert self.full is not None
+        for (fqn, shape, o, n), p in zip(self.param_infos, self.params):
+            p.data = self.full[o : o + n].view(shape)
+
+    def _detach_param_views(self):
+        for (fqn, shape, o, n), p in zip(self.param_infos, self.params):
+            p.data = torch.empty(0, dtype=self.param_dtype, device=self.device)
+
+    def attach_grad_views(self):
+        """Point param.grad at views of the flat grad buffer so autograd
+        accumulates in place (no copy)."""
+        if self.grad_full is None:
+            self.grad_full = torch.zeros(
+                self.flat_numel, dtype=self.param_dtype, device=self.device
+            )
+        for (fqn, shape, o, n), p in zip(self.param_infos, self.params):
+            p.grad = self.grad_full[o : o + n].view(shape)
+
+    # ------------------------------------------------------------------
+    def dtensor_spec(self, mesh: DeviceMesh, shape: torch.Size) -> DTensorSpec:
+        units = tuple([1] * self.world_size)
+        placement = RaggedShard((0,), units)
+        tm = TensorMeta(shape, (1,), self.param_dtype)
+        return DTensorSpec(mesh, (placement,), tm)
+
+
+_UNIT_CLASSES_DEFAULT = ("TransformerBlock", "Block", "DecoderLayer")
+
+
+class FSDP(nn.Module):
+    def __init__(
+        self,
+        module: nn.Module,
+        mesh: Optional[DeviceMesh] = None,
+        *,
+        process_group: Optional[dist.ProcessGroup] = None,
+        unit_classes: Sequence[str] = _UNIT_CLASSES_DEFAULT,
+        param_dtype: torch.dtype = torch.bfloat16,
+        reduce_dtype: torch.dtype = torch.bfloat16,
+        prefetch: bool = True,
+        reshard_after_forward: bool = True,
+        device: Optional[torch.device] = None,
+    ):
+        super().__init__()
+        self.module = module
+        self.mesh = mesh
+        if mesh is not None:
+            self.pg = mesh.get_group(mesh.ndim - 1) if process_group is None else process_group
+        else:
+            self.pg = process_group
+        self.world_size = dist.get_world_size(self.pg) if (self.pg is not None and dist.is_initialized()) else 1
+        self.rank = dist.get_rank(self.pg) if (self.pg is not None and dist.is_initialized()) else 0
+        self.prefetch = prefetch
+        self.reshard_after_forward = reshard_after_forward and self.world_size > 1
+        self.param_dtype = param_dtype
+        dev = device
+        if dev is None:
+            dev = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+        self.device = dev
+        self._on_gpu = dev.type == "cuda"
+
+        # separate communicators so AG and RS streams can't reorder against
+        # each other across ranks
+        if self.world_size > 1:
+            ranks = dist.get_process_group_ranks(self.pg)
+            self.ag_pg = self.pg
+            self.rs_pg = dist.new_group(ranks=ranks) if dist.is_initialized() else self.pg
+        else:
+            self.ag_pg = self.rs_pg = None
+
+        if self._on_gpu:
+            self._ag_stream = torch.cuda.Stream()
+            self._rs_stream = torch.cuda.Stream()
+        else:
+            self._ag_stream = self._rs_stream = None
+
+        module.to(dev)
+        if param_dtype is not None:
+            module.to(dtype=param_dtype)
+            # keep fp32 buffers (rope table) in fp32
+            for b in module.buffers():
+                if b.dtype == param_dtype and b.is_floating_point():
+                    pass
+        self.units: List[FSDPUnit] = []
+        self._build_units(unit_classes)
+        self._install_hooks()
+        self._exec_order: List[int] = []  # recorded forward order of units
+        self._in_backward = False
+
+    # ------------------------------------------------------------------
+    def _build_units(self, unit_classes: Sequence[str]):
+        taken = set()
+        idx = 0
+        for name, mod in self.module.named_modules():
+            if type(mod).__name__ in unit_classes:
+                params = [
+                    (f"{name}.{pn}" if name else pn, p)
+                    for pn, p in mod.named_parameters(recurse=True)
+                    if id(p) not in taken
+                ]
+                if not params:
+                    continue
+                for _, p in params:
+                    taken.add(id(p))
+                self.units.append(
+                    FSDPUnit(
+                        name or f"unit{idx}", mod, params, self.world_size,
+                        self.rank, self.device, self.param_dtype,
+                    )
+                )
+                idx += 1
+        # root unit: everything else
+        rest = [
+            (pn, p)
+            for pn, p in self.module.named_parameters()
+            if id(p) not in taken
+        ]
+        if rest:
+            self.units.append(
+                FSDPUnit(
+                    "__root__", self.module, rest, self.world_size, self.rank,
+                    self.device, self.param_dtype,
+                )
+            )
+        self._unit_of_module: Dict[int, FSDPUnit] = {
+            id(u.module): u for u in self.units if u.name != "__root__"
+        }
+        self._root_unit = next((u for u in self.units if u.name == "__root__"), None)
+
+    # ------------------------------------------------------------------
+    def _install_hooks(self):
+        if self.world_size == 1:
+            # no comm; grads still flow into flat buffers lazily
+            for u in self.units:
+                self._install_grad_hooks(u)
+            return
+        for u in self.units:
+            if u.name == "__root__":
+                continue
+            u.module.register_forward_pre_hook(self._make_fwd_pre(u))
+            if self.reshard_after_forward:
+                u.module.register_forward_hook(self._make_fwd_post(u))
+            u.module.register_full_backward_pre_hook(self._make_bwd_pre(u))
+            self._install_grad_hooks(u)
+        if self._root_unit is not None:
+            self._install_grad_hooks(self._root_unit)
+
+    def _install_grad_hooks(self, unit: FSDPUnit):
+        for p in unit.params:
+            p.register_post_accumulate_grad_hook(self._make_grad_hook(unit))
+
+    # ---------------- forward path ----------------
+    def _unshard(self, unit: FSDPUnit, async_on_stream: bool = True):
+        if unit._is_unsharded or self.world_size == 1:
+            return
+        unit.full = torch.empty(
+            unit.flat_numel, dtype=unit.param_dtype, device=self.device
+        )
+        if self._on_gpu and async_on_stream:
+            self._ag_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._ag_stream):
+                dist.all_gather_into_tensor(unit.full, unit.shard, group=self.ag_pg)
+                unit._ag_event = torch.cuda.Event()
+                unit._ag_event.record(self._ag_stream)
+            unit.full.record_stream(self._ag_stream)
+        else:
+            _all_gather_flat(unit.full, unit.shard, self.ag_pg)
+            unit._ag_event = None
+        unit._attach_param_views()
+        unit._is_unsharded = True
+
+    def _wait_unshard(self, unit: FSDPUnit):
+        if unit._ag_event is not None:
+            torch.cuda.current_stream().wait_event(unit._ag_event)
+            unit._ag_event = None
+
+    def _reshard(self, unit: FSDPUnit):
+        if self.world_size == 1 or not unit._is_unsharded:
+            return
+        unit._detach_param_views()
+        if self._on_gpu:
+            unit.full.record_stream(torch.cuda.current_stream())
+        unit.full = None
+        unit._is_unsharded = False
+
+    def _make_fwd_pre(self, unit: FSDPUnit):
+        def hook(mod, args):
+            if not unit._is_unsharded:
+                self._unshard(unit)
+            self._wait_unshard(unit)
+            if unit not in [self.units[i] for i in self._exec_order]:
+                self._exec_order.append(self.units.index(unit))
+            # prefetch the next unit in recorded order
+            if self.prefetch:
+                nxt = self._next_unit_after(unit, forward=True)
+                if nxt is not None and not nxt._is_unsharded:
+                    self._unshard(nxt)
+            return None
+
+        return hook
+
+    def _make_fwd_post(self, unit: FSDPUnit):
+        def hook(mod, args, out):
+            if not self._in_backward and not (self.training and torch.is_grad_enabled() and False):
+                self._reshard(unit)
+            return None
+
+        return hook
+
+    def _make_bwd_pre(self, unit: FSDPUnit):
+        def hook(mod, grad_output):
+            self._in_backward = True
+            if not unit._is_unsharded:
+                self._unshard(unit)
+            self._wait_unshard(unit)
+            unit.attach_grad_views()
+            if self.prefetch:
+                nxt = self._next_unit_after(unit, forward=False)
+                if nxt is not None and not nxt._is_unsharded:
+                    self._unshard(nxt)
+            return None
+
+        return hook
+
+    def _next_unit_after(self, unit: FSDPUnit, forward: bool) -> Optional[FSDPUnit]:
+        try:
+            pos = self._exec_order.index(self.units.index(unit))
+        except ValueError:
+            return None
+        if forward:
+            if pos + 1 < len(self._exec_order):
+                return self.units[self._exec_order[pos + 1]]
+        else:
+            if pos - 1 >= 0:
+                return self.units[self._exec_order[pos - 1]]
+        return None
+
+    # ---------------- gradient path ----------------
+    def _make_grad_hook(self, unit: FSDPUnit):
+        def hook(param):
+            unit._grads_ready += 1
+            if unit._grads_ready < len(unit.params):
+                return
+            unit._grads_ready = 0
+            self._finish_unit_grads(unit)
+
+        return hook
+
+    def _finish_unit_grads(self, unit: FSDPUnit):
+        # all grads of this unit accumulated into unit.grad_full
+        for p in unit.params:
+            p.grad = None
+        if self.world_size == 1:
+            unit.grad_shard = unit.grad_full
+            return
+        if unit.grad_shard is None:
+            unit.grad_shard = torch.empty(
+                unit.shard_numel, dtype=unit.param_dtype, device=self.device
+            )
+            self._grad_shard_fresh = True
+        # reduce-scatter on the RS stream
+        gf = unit.grad_full
+        if self._on_gpu:
+            self._rs_stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._rs_stream):
+                dist.reduce_scatter_tensor(
+                    unit.grad_shard, gf, op=dist.ReduceOp.AVG if hasattr(dist.ReduceOp, "AVG") else dist.ReduceOp.SUM,
+                    group=self.rs_pg,
+                )
+            gf.record_stream(self._rs_stream)
+            unit.grad_shard.record_stream(self._rs_stream)
+        else:
+            dist.all_reduce(gf, group=self.rs_pg)
+            gf.div_(self.world_size)
+            unit.grad_shard.copy_(gf[unit.shard_off : unit.shard_off + unit.shard_numel])
+        unit.grad_full = None
+        # reshard params after backward
+        self._reshard(unit)
+
+    def finish_grad_sync(self):
+        """Wait for all grad reduce-scatters (call after loss.backward())."""
+        self._in_backward = False
+        self._exec_order_done = True
+        if self._on_gpu and self.world_size > 1:
+            torch.cuda.current_stream().wait_stream(self._rs_stream)
+        # re-scale: RCCL without AVG support -> grads were summed
+        if self.world_size > 1 and not hasattr(dist.ReduceOp, "AVG"):
+            for u in self.units:
+                if u.grad_shard is not None:
+                    u.grad_shard.div_(self.world_size)
+
+    def zero_grad_buffers(self):
+        for u in self.units:
+            u.grad_full = None
+            u.grad_shard = None
+            u._grads_ready = 0
+            for p in u.params:
+                p.grad = None
+
+    # ---------------- public API ----------------
+    def forward(self, *args, **kwargs):
+        # root unit (embeddings/head/final-norm) stays unsharded for the
+        # whole step: its params bracket the graph
+        if self._root_unit is not None:
+            if not self._root_unit._is_unsharded:
+                self._unshard(self._root_unit)
+                self._wait_unshard(self._root_unit)
+            if self.training and torch.is_grad_enabled():
+                self._root_unit.attach_grad_views()
+        if self.world_size == 1:
+            if self.training and torch.is_grad_enabled():
+                for u in self.units:
+                    u.attach_grad_views()
+        elif self.prefetch and self._exec_order:
+            # pre-issue the first unit's all-gather before stepping in
+            first = self.units[self._exec_order[0]]
+            if not first._is_unsharded:
+                self._unshard(first)
+        return self.module(*args, **kwargs)
+
+    @torch.no_grad()
+    def grad_norm_sq(self) -> torch.Tensor:
+        """Global L2 norm^2 of gradients (fused kernel per shard +
+        allreduce over DP)."""
+        from ..ops import l2norm_sq
+
+        total = None
+        for u in self.units:
+            if u.grad_shard is None:
+                continue
+            s = l2norm_sq(u.grad_shard)
+            total = s if total is None else total + s
+        if total is None:
+            total = torch.zeros((), device=self.device)
+        if self.world_size > 1:
+            dist.all_reduce(total, group=self.pg)
+        return total
+
+    def sharded_state_dict(self) -> Dict[str, DTensor]:
+        """Per-parameter DTensors in RaggedShard layout over the flat unit
+        buffers — the communication-free checkpoint format (reference
+        vescale/dtensor/vescale_utils/checkpoint.py)."""
+        out: Dict[str, DTensor] = {}
+        assert self.mesh is not None, "sharded_state_dict requires a DeviceMesh"
+        for u in self.units:
+            for fqn, shape, off, n in u.param_infos:
+                # view of this param's slice within this rank's shard
+                s0, s1 = u.shard_off, u.shard_off + u.shard_numel
+                lo, hi = max(off, s0), min(off + n, s1)
+                local = (
+                    u.shard[lo - s0 : hi - s0]
+                    if hi > lo
+                    else u.shard.new_empty(0)
+                )
+                units = self._param_units(off, n, u)
+                placement = RaggedShard((0,), units)
+                tm = TensorMeta(torch.Size((n,)), (1,), u.param_dtype)
+                spec = DTensorSpec(self.mesh, (placement,), tm)
+                out[fqn] = DTensor(local, spec, requires_grad=False)
+        return out
+
+    def _param_units(self, off: int, n: int, u: FSDPUnit) -> Tuple[int, ...]:
+        """Element counts of this param on each rank (RaggedShard units)."""
+        counts = []
+        for r in range(u.world_size):
+            s0 = r * u.shard_numel
+            s1 = s0 + u.shard_numel
+            counts.append(max(0, min(off + n, s1) - max(off, s0)))
+        return tuple(counts)

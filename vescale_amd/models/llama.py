@@ -1,0 +1,182 @@
+"""Llama-3 model family — MI355X-native implementation.
+
+The flagship bench model (BASELINE.json: Llama-3-8B FSDP tokens/sec).
+Forward runs on plain tensors (the FSDP engine manages sharding at the
+parameter level); the hot non-GEMM ops are our CDNA4 HIP kernels
+(RMSNorm / RoPE / SwiGLU / fused CE), GEMMs go to hipBLASLt via torch.
+Parity role: reference examples' HF Llama2 / open_llama configs
+(legacy/examples/llama2_4D_finetune/, open_llama_4D_benchmark/).
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops import build_rope_table, fused_cross_entropy, rmsnorm, rope_apply, swiglu
+
+
+@dataclass
+class LlamaConfig:
+    dim: int = 4096
+    n_layers: int = 32
+    n_heads: int = 32
+    n_kv_heads: int = 8
+    ffn_dim: int = 14336
+    vocab_size: int = 128256
+    max_seq_len: int = 8192
+    rope_theta: float = 500000.0
+    norm_eps: float = 1e-5
+    tie_embeddings: bool = False
+
+    @property
+    def head_dim(self) -> int:
+        return self.dim // self.n_heads
+
+    def num_params(self) -> int:
+        d, f, v = self.dim, self.ffn_dim, self.vocab_size
+        per_layer = (
+            d * d  # wq
+            + 2 * d * (self.n_kv_heads * self.head_dim)  # wk, wv
+            + d * d  # wo
+            + 3 * d * f  # w1, w3, w2
+            + 2 * d  # norms
+        )
+        emb = v * d * (1 if self.tie_embeddings else 2)
+        return self.n_layers * per_layer + emb + d
+
+
+def llama3_8b() -> LlamaConfig:
+    return LlamaConfig()
+
+
+def llama3_70b() -> LlamaConfig:
+    return LlamaConfig(
+        dim=8192, n_layers=80, n_heads=64, n_kv_heads=8, ffn_dim=28672,
+        vocab_size=128256,
+    )
+
+
+def llama_tiny(vocab: int = 512, seq: int = 128) -> LlamaConfig:
+    return LlamaConfig(
+        dim=64, n_layers=2, n_heads=4, n_kv_heads=2, ffn_dim=128,
+        vocab_size=vocab, max_seq_len=seq, rope_theta=10000.0,
+    )
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim: int, eps: float):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        return rmsnorm(x, self.weight, self.eps)
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        d, hd = cfg.dim, cfg.head_dim
+        kv = cfg.n_kv_heads * hd
+        self.wq = nn.Linear(d, d, bias=False)
+        self.wk = nn.Linear(d, kv, bias=False)
+        self.wv = nn.Linear(d, kv, bias=False)
+        self.wo = nn.Linear(d, d, bias=False)
+
+    def forward(self, x: torch.Tensor, rope_table: torch.Tensor) -> torch.Tensor:
+        B, S, _ = x.shape
+        cfg = self.cfg
+        q = self.wq(x).view(B, S, cfg.n_heads, cfg.head_dim)
+        k = self.wk(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
+        v = self.wv(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
+        q = rope_apply(q, rope_table)
+        k = rope_apply(k, rope_table)
+        q = q.transpose(1, 2)  # [B, H, S, D]
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
+        o = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+        o = o.transpose(1, 2).reshape(B, S, cfg.dim)
+        return self.wo(o)
+
+
+class FeedForward(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        # w1 (gate) and w3 (up) fused into one GEMM
+        self.w13 = nn.Linear(cfg.dim, 2 * cfg.ffn_dim, bias=False)
+        self.w2 = nn.Linear(cfg.ffn_dim, cfg.dim, bias=False)
+        self.ffn_dim = cfg.ffn_dim
+
+    def forward(self, x):
+        gu = self.w13(x)
+        gate, up = gu.split(self.ffn_dim, dim=-1)
+        return self.w2(swiglu(gate.contiguous(), up.contiguous()))
+
+
+class TransformerBlock(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.attn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.attn = Attention(cfg)
+        self.ffn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.ffn = FeedForward(cfg)
+
+    def forward(self, x, rope_table):
+        x = x + self.attn(self.attn_norm(x), rope_table)
+        x = x + self.ffn(self.ffn_norm(x))
+        return x
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.tok_embeddings = nn.Embedding(cfg.vocab_size, cfg.dim)
+        self.layers = nn.ModuleList(TransformerBlock(cfg) for _ in range(cfg.n_layers))
+        self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.output = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        if cfg.tie_embeddings:
+            self.output.weight = self.tok_embeddings.weight
+        self.register_buffer(
+            "rope_table",
+            build_rope_table(cfg.max_seq_len, cfg.head_dim, cfg.rope_theta),
+            persistent=False,
+        )
+        self.activation_checkpointing = False
+
+    def init_weights(self, std: float = 0.02):
+        for name, p in self.named_parameters():
+            if p.ndim >= 2:
+                nn.init.normal_(p, mean=0.0, std=std)
+            elif "weight" in name:  # norms
+                nn.init.ones_(p)
+            else:
+                nn.init.zeros_(p)
+
+    def forward(
+        self,
+        tokens: torch.Tensor,
+        targets: Optional[torch.Tensor] = None,
+        ignore_index: int = -100,
+    ):
+        h = self.tok_embeddings(tokens)
+        tab = self.rope_table
+        for layer in self.layers:
+            if self.activation_checkpointing and self.training:
+                h = torch.utils.checkpoint.checkpoint(layer, h, tab, use_reentrant=False)
+            else:
+                h = layer(h, tab)
+        h = self.norm(h)
+        logits = self.output(h)
+        if targets is None:
+            return logits
+        loss = fused_cross_entropy(
+            logits.reshape(-1, self.cfg.vocab_size), targets.reshape(-1), ignore_index
+        )
+        return loss

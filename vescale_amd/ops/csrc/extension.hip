@@ -1,0 +1,291 @@
+// vescale_amd C++/HIP extension — single TU: kernels + torch bindings.
+//
+// Built in-tree for gfx950 only (no multi-backend dispatch, no CUDA shims).
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "rmsnorm.hip"
+#include "rope.hip"
+#include "swiglu.hip"
+#include "adamw.hip"
+#include "cross_entropy.hip"
+#include "philox_random.hip"
+
+#include <vector>
+
+namespace {
+
+inline hipStream_t cur_stream() {
+  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+inline void check_bf16_contig(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_cuda(), name, " must be on device");
+}
+
+// ------------------------------ RMSNorm ------------------------------
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  int hidden = (int)x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  int64_t rows = x.numel() / hidden;
+  auto out = at::empty_like(x);
+  auto rrms = at::empty({rows}, x.options().dtype(at::kFloat));
+  int grid = (int)std::min<int64_t>(rows, 2048);
+  hipLaunchKernelGGL(rmsnorm_fwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)x.data_ptr(),
+                     (const unsigned short*)w.data_ptr(),
+                     (unsigned short*)out.data_ptr(), rrms.data_ptr<float>(),
+                     rows, hidden, (float)eps);
+  return {out, rrms};
+}
+
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                    at::Tensor rrms) {
+  check_bf16_contig(dy, "dy");
+  check_bf16_contig(x, "x");
+  int hidden = (int)x.size(-1);
+  TORCH_CHECK(hidden <= 16384, "rmsnorm_bwd supports hidden <= 16384");
+  int64_t rows = x.numel() / hidden;
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({hidden}, x.options().dtype(at::kFloat));
+  int grid = (int)std::min<int64_t>(rows, 1024);
+  hipLaunchKernelGGL(rmsnorm_bwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)dy.data_ptr(),
+                     (const unsigned short*)x.data_ptr(),
+                     (const unsigned short*)w.data_ptr(),
+                     rrms.data_ptr<float>(), (unsigned short*)dx.data_ptr(),
+                     dw.data_ptr<float>(), rows, hidden);
+  return {dx, dw};
+}
+
+// ------------------------------ RoPE ------------------------------
+at::Tensor rope(at::Tensor x, at::Tensor table, int64_t pos_offset,
+                bool backward, bool inplace) {
+  // x: [B, S, H, D] bf16, table: [S_max, D/2, 2] f32
+  check_bf16_contig(x, "x");
+  TORCH_CHECK(table.scalar_type() == at::kFloat && table.is_contiguous());
+  TORCH_CHECK(x.dim() == 4, "rope expects [B,S,H,D]");
+  int B = (int)x.size(0), S = (int)x.size(1), H = (int)x.size(2), D = (int)x.size(3);
+  TORCH_CHECK(D % 2 == 0);
+  auto out = inplace ? x : at::empty_like(x);
+  int64_t n_tokens = (int64_t)B * S * H;
+  int64_t pairs = n_tokens * (D / 2);
+  int grid = grid_for(pairs / 2, 256);
+  hipLaunchKernelGGL(rope_fwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)x.data_ptr(),
+                     (unsigned short*)out.data_ptr(), table.data_ptr<float>(),
+                     n_tokens, H, S, D, (int)pos_offset, backward ? 1 : 0);
+  return out;
+}
+
+// ------------------------------ SwiGLU ------------------------------
+at::Tensor swiglu_fwd(at::Tensor gate, at::Tensor up) {
+  check_bf16_contig(gate, "gate");
+  check_bf16_contig(up, "up");
+  auto out = at::empty_like(gate);
+  int64_t n = gate.numel();
+  int grid = grid_for(n / 8, 256);
+  hipLaunchKernelGGL(swiglu_fwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)gate.data_ptr(),
+                     (const unsigned short*)up.data_ptr(),
+                     (unsigned short*)out.data_ptr(), n);
+  return out;
+}
+
+std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor gate, at::Tensor up) {
+  auto dgate = at::empty_like(gate);
+  auto dup = at::empty_like(up);
+  int64_t n = gate.numel();
+  int grid = grid_for(n / 8, 256);
+  hipLaunchKernelGGL(swiglu_bwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)dy.data_ptr(),
+                     (const unsigned short*)gate.data_ptr(),
+                     (const unsigned short*)up.data_ptr(),
+                     (unsigned short*)dgate.data_ptr(),
+                     (unsigned short*)dup.data_ptr(), n);
+  return {dgate, dup};
+}
+
+// ------------------------------ Cross entropy ------------------------------
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor target,
+                               int64_t ignore_index) {
+  check_bf16_contig(logits, "logits");
+  TORCH_CHECK(target.scalar_type() == at::kLong && target.is_contiguous());
+  int vocab = (int)logits.size(-1);
+  TORCH_CHECK(vocab % 8 == 0, "vocab must be a multiple of 8");
+  int64_t rows = logits.numel() / vocab;
+  auto loss = at::empty({rows}, logits.options().dtype(at::kFloat));
+  auto lse = at::empty({rows}, logits.options().dtype(at::kFloat));
+  int grid = (int)std::min<int64_t>(rows, 2048);
+  hipLaunchKernelGGL(ce_fwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)logits.data_ptr(),
+                     target.data_ptr<int64_t>(), loss.data_ptr<float>(),
+                     lse.data_ptr<float>(), rows, vocab, ignore_index);
+  return {loss, lse};
+}
+
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
+                  at::Tensor dloss, int64_t ignore_index, bool inplace) {
+  int vocab = (int)logits.size(-1);
+  int64_t rows = logits.numel() / vocab;
+  auto dlogits = inplace ? logits : at::empty_like(logits);
+  int grid = (int)std::min<int64_t>(rows, 2048);
+  hipLaunchKernelGGL(ce_bwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+                     (const unsigned short*)logits.data_ptr(),
+                     (unsigned short*)dlogits.data_ptr(),
+                     target.data_ptr<int64_t>(), lse.data_ptr<float>(),
+                     dloss.data_ptr<float>(), rows, vocab, ignore_index);
+  return dlogits;
+}
+
+// ------------------------------ AdamW ------------------------------
+void adamw_step(at::Tensor param, c10::optional<at::Tensor> master,
+                at::Tensor grad, at::Tensor m, at::Tensor v, double lr,
+                double beta1, double beta2, double eps, double weight_decay,
+                int64_t step, double grad_scale) {
+  check_bf16_contig(param, "param");
+  int64_t n = param.numel();
+  float bc1 = 1.f - powf((float)beta1, (float)step);
+  float bc2 = 1.f - powf((float)beta2, (float)step);
+  const unsigned short* gb = nullptr;
+  const float* gf = nullptr;
+  if (grad.scalar_type() == at::kBFloat16) gb = (const unsigned short*)grad.data_ptr();
+  else gf = grad.data_ptr<float>();
+  float* mp = master.has_value() ? master->data_ptr<float>() : nullptr;
+  int grid = grid_for(n / 4, 256);
+  hipLaunchKernelGGL(adamw_flat_bf16, dim3(grid), dim3(256), 0, cur_stream(),
+                     (unsigned short*)param.data_ptr(), mp, gb, gf,
+                     m.data_ptr<float>(), v.data_ptr<float>(), n, (float)lr,
+                     (float)beta1, (float)beta2, (float)eps,
+                     (float)weight_decay, bc1, bc2, (float)grad_scale);
+}
+
+at::Tensor l2norm_sq(at::Tensor x) {
+  auto out = at::zeros({1}, x.options().dtype(at::kFloat));
+  int64_t n = x.numel();
+  const unsigned short* xb = nullptr;
+  const float* xf = nullptr;
+  if (x.scalar_type() == at::kBFloat16) xb = (const unsigned short*)x.data_ptr();
+  else xf = x.data_ptr<float>();
+  int grid = grid_for(n / 8, 256);
+  hipLaunchKernelGGL(l2norm_sq_flat, dim3(grid), dim3(256), 0, cur_stream(),
+                     xb, xf, out.data_ptr<float>(), n);
+  return out;
+}
+
+void scale_(at::Tensor x, c10::optional<at::Tensor> scale_t, double scale_c) {
+  int64_t n = x.numel();
+  unsigned short* xb = nullptr;
+  float* xf = nullptr;
+  if (x.scalar_type() == at::kBFloat16) xb = (unsigned short*)x.data_ptr();
+  else xf = x.data_ptr<float>();
+  const float* sp = scale_t.has_value() ? scale_t->data_ptr<float>() : nullptr;
+  int grid = grid_for(n / 8, 256);
+  hipLaunchKernelGGL(scale_flat, dim3(grid), dim3(256), 0, cur_stream(), xb,
+                     xf, sp, (float)scale_c, n);
+}
+
+// ------------------------------ philox random ------------------------------
+ShardDesc make_desc(const std::vector<int64_t>& gshape,
+                    const std::vector<int64_t>& lshape,
+                    const std::vector<int64_t>& offset, int64_t flat_offset,
+                    bool is_flat) {
+  ShardDesc d{};
+  d.ndim = (int)gshape.size();
+  TORCH_CHECK(d.ndim <= MAXD);
+  for (int i = 0; i < d.ndim; ++i) {
+    d.gshape[i] = gshape[i];
+    d.lshape[i] = lshape[i];
+    d.offset[i] = offset[i];
+  }
+  d.flat_offset = flat_offset;
+  d.is_flat = is_flat ? 1 : 0;
+  return d;
+}
+
+void philox_uniform_(at::Tensor out, std::vector<int64_t> gshape,
+                     std::vector<int64_t> lshape, std::vector<int64_t> offset,
+                     int64_t flat_offset, bool is_flat, int64_t seed,
+                     int64_t philox_offset, double lo, double hi) {
+  auto d = make_desc(gshape, lshape, offset, flat_offset, is_flat);
+  int64_t n = out.numel();
+  int grid = grid_for(n, 256);
+  if (out.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(philox_uniform_bf16, dim3(grid), dim3(256), 0,
+                       cur_stream(), (unsigned short*)out.data_ptr(), d, n,
+                       (uint64_t)seed, (uint64_t)philox_offset, (float)lo,
+                       (float)hi);
+  } else {
+    hipLaunchKernelGGL(philox_uniform_f32, dim3(grid), dim3(256), 0,
+                       cur_stream(), out.data_ptr<float>(), d, n,
+                       (uint64_t)seed, (uint64_t)philox_offset, (float)lo,
+                       (float)hi);
+  }
+}
+
+void philox_normal_(at::Tensor out, std::vector<int64_t> gshape,
+                    std::vector<int64_t> lshape, std::vector<int64_t> offset,
+                    int64_t flat_offset, bool is_flat, int64_t seed,
+                    int64_t philox_offset, double mean, double std) {
+  auto d = make_desc(gshape, lshape, offset, flat_offset, is_flat);
+  int64_t n = out.numel();
+  int grid = grid_for(n, 256);
+  if (out.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(philox_normal_bf16, dim3(grid), dim3(256), 0,
+                       cur_stream(), (unsigned short*)out.data_ptr(), d, n,
+                       (uint64_t)seed, (uint64_t)philox_offset, (float)mean,
+                       (float)std);
+  } else {
+    hipLaunchKernelGGL(philox_normal_f32, dim3(grid), dim3(256), 0,
+                       cur_stream(), out.data_ptr<float>(), d, n,
+                       (uint64_t)seed, (uint64_t)philox_offset, (float)mean,
+                       (float)std);
+  }
+}
+
+std::vector<at::Tensor> philox_dropout(at::Tensor x, std::vector<int64_t> gshape,
+                                       std::vector<int64_t> lshape,
+                                       std::vector<int64_t> offset,
+                                       int64_t flat_offset, bool is_flat,
+                                       int64_t seed, int64_t philox_offset,
+                                       double p, bool need_mask) {
+  check_bf16_contig(x, "x");
+  auto d = make_desc(gshape, lshape, offset, flat_offset, is_flat);
+  auto out = at::empty_like(x);
+  at::Tensor mask;
+  unsigned char* mp = nullptr;
+  if (need_mask) {
+    mask = at::empty(x.sizes(), x.options().dtype(at::kByte));
+    mp = mask.data_ptr<unsigned char>();
+  }
+  int64_t n = x.numel();
+  int grid = grid_for(n, 256);
+  hipLaunchKernelGGL(philox_dropout_bf16, dim3(grid), dim3(256), 0,
+                     cur_stream(), (const unsigned short*)x.data_ptr(),
+                     (unsigned short*)out.data_ptr(), mp, d, n, (uint64_t)seed,
+                     (uint64_t)philox_offset, (float)p);
+  if (need_mask) return {out, mask};
+  return {out};
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rope", &rope);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
+  m.def("adamw_step", &adamw_step);
+  m.def("l2norm_sq", &l2norm_sq);
+  m.def("scale_", &scale_);
+  m.def("philox_uniform_", &philox_uniform_);
+  m.def("philox_normal_", &philox_normal_);
+  m.def("philox_dropout", &philox_dropout);
+}

@@ -1,0 +1,130 @@
+// RMSNorm forward/backward — CDNA4, memory-bound (target: HBM BW ceiling).
+//
+// Replaces the reference's layer-norm call sites for Llama (SURVEY.md §2.7
+// "layer_norm fwd/bwd ... RMSNorm for Llama in new build").  bf16 I/O,
+// fp32 accumulation, bf16x8 vectorized loads (guide Guideline 13: scalar
+// bf16 loads cost ~2x), one 256-thread block per row for H<=8192, fp32
+// rrms stashed for backward.
+#include "common.h"
+
+#define BLOCK 256
+
+// x: [N, H] bf16;  w: [H] bf16;  out: [N, H] bf16;  rrms: [N] f32
+extern "C" __global__ void __launch_bounds__(BLOCK)
+rmsnorm_fwd_bf16(const unsigned short* __restrict__ x,
+                 const unsigned short* __restrict__ w,
+                 unsigned short* __restrict__ out,
+                 float* __restrict__ rrms_out,
+                 int64_t n_rows, int hidden, float eps) {
+  __shared__ float lds[BLOCK / WAVE];
+  const int vec = 8;
+  const int per_row_iters = (hidden / vec + BLOCK - 1) / BLOCK;
+  for (int64_t row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    const unsigned short* xr = x + row * hidden;
+    unsigned short* orow = out + row * hidden;
+    float ss = 0.f;
+    // pass 1: sum of squares
+    for (int it = 0; it < per_row_iters; ++it) {
+      int i = (it * BLOCK + threadIdx.x) * vec;
+      if (i < hidden) {
+        short8v v = *reinterpret_cast<const short8v*>(xr + i);
+#pragma unroll
+        for (int j = 0; j < vec; ++j) {
+          float f = bf16_to_f32((unsigned short)v[j]);
+          ss += f * f;
+        }
+      }
+    }
+    float total = block_reduce_sum<BLOCK>(ss, lds);
+    float rrms = rsqrtf(total / (float)hidden + eps);
+    if (threadIdx.x == 0 && rrms_out) rrms_out[row] = rrms;
+    // pass 2: normalize + scale (x row is in L1/L2 now)
+    for (int it = 0; it < per_row_iters; ++it) {
+      int i = (it * BLOCK + threadIdx.x) * vec;
+      if (i < hidden) {
+        short8v v = *reinterpret_cast<const short8v*>(xr + i);
+        short8v wv = *reinterpret_cast<const short8v*>(w + i);
+        short8v o;
+#pragma unroll
+        for (int j = 0; j < vec; ++j) {
+          float f = bf16_to_f32((unsigned short)v[j]);
+          float g = bf16_to_f32((unsigned short)wv[j]);
+          o[j] = (short)f32_to_bf16(f * rrms * g);
+        }
+        *reinterpret_cast<short8v*>(orow + i) = o;
+      }
+    }
+  }
+}
+
+// backward:
+//   dx = rrms * w * dy - x * rrms^3/H * sum_j(dy_j * w_j * x_j)
+//   dw = sum_rows(dy * x * rrms)
+// dw accumulated per-block in registers over this block's rows, then one
+// atomicAdd per column into fp32 dw buffer (caller zero-inits).
+extern "C" __global__ void __launch_bounds__(BLOCK)
+rmsnorm_bwd_bf16(const unsigned short* __restrict__ dy,
+                 const unsigned short* __restrict__ x,
+                 const unsigned short* __restrict__ w,
+                 const float* __restrict__ rrms_in,
+                 unsigned short* __restrict__ dx,
+                 float* __restrict__ dw,  // [H] fp32, zero-init
+                 int64_t n_rows, int hidden) {
+  __shared__ float lds[BLOCK / WAVE];
+  const int vec = 8;
+  const int cols_per_thread = (hidden / vec + BLOCK - 1) / BLOCK * vec;
+  // register accumulator for this thread's columns (max 64 cols => H<=16384)
+  float dw_acc[64];
+#pragma unroll
+  for (int j = 0; j < 64; ++j) dw_acc[j] = 0.f;
+
+  const int per_row_iters = (hidden / vec + BLOCK - 1) / BLOCK;
+  for (int64_t row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    const unsigned short* dyr = dy + row * hidden;
+    const unsigned short* xr = x + row * hidden;
+    unsigned short* dxr = dx + row * hidden;
+    float rrms = rrms_in[row];
+    float dot = 0.f;
+    for (int it = 0; it < per_row_iters; ++it) {
+      int i = (it * BLOCK + threadIdx.x) * vec;
+      if (i < hidden) {
+        short8v dv = *reinterpret_cast<const short8v*>(dyr + i);
+        short8v xv = *reinterpret_cast<const short8v*>(xr + i);
+        short8v wv = *reinterpret_cast<const short8v*>(w + i);
+#pragma unroll
+        for (int j = 0; j < vec; ++j) {
+          dot += bf16_to_f32((unsigned short)dv[j]) * bf16_to_f32((unsigned short)wv[j]) *
+                 bf16_to_f32((unsigned short)xv[j]);
+        }
+      }
+    }
+    float tdot = block_reduce_sum<BLOCK>(dot, lds);
+    float coef = tdot * rrms * rrms * rrms / (float)hidden;
+    for (int it = 0; it < per_row_iters; ++it) {
+      int i = (it * BLOCK + threadIdx.x) * vec;
+      if (i < hidden) {
+        short8v dv = *reinterpret_cast<const short8v*>(dyr + i);
+        short8v xv = *reinterpret_cast<const short8v*>(xr + i);
+        short8v wv = *reinterpret_cast<const short8v*>(w + i);
+        short8v o;
+#pragma unroll
+        for (int j = 0; j < vec; ++j) {
+          float dyf = bf16_to_f32((unsigned short)dv[j]);
+          float xf = bf16_to_f32((unsigned short)xv[j]);
+          float wf = bf16_to_f32((unsigned short)wv[j]);
+          o[j] = (short)f32_to_bf16(rrms * wf * dyf - xf * coef);
+          dw_acc[it * vec + j] += dyf * xf * rrms;
+        }
+        *reinterpret_cast<short8v*>(dxr + i) = o;
+      }
+    }
+  }
+  // flush dw
+  for (int it = 0; it < per_row_iters; ++it) {
+    int i = (it * BLOCK + threadIdx.x) * vec;
+    if (i < hidden) {
+#pragma unroll
+      for (int j = 0; j < vec; ++j) atomicAdd(dw + i + j, dw_acc[it * vec + j]);
+    }
+  }
+}

@@ -1,0 +1,274 @@
+"""Autograd wrappers for the CDNA4 HIP kernels, with pure-torch reference
+implementations used (a) on CPU boxes for tests, (b) as the fp32 numerics
+oracle the GPU tests compare against (SURVEY.md §4 test strategy)."""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+
+def _ext():
+    from . import _C, require_ext
+
+    return require_ext()
+
+
+def _use_hip(*tensors) -> bool:
+    from . import has_ext
+
+    on_gpu = all(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
+    if not on_gpu:
+        return False
+    if not has_ext():
+        # loud failure: GPU present but no extension
+        from . import require_ext
+
+        require_ext()
+    return True
+
+
+# ---------------------------------------------------------------------------
+# RMSNorm
+# ---------------------------------------------------------------------------
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, eps: float):
+        if _use_hip(x, weight):
+            x2 = x.contiguous()
+            out, rrms = _ext().rmsnorm_fwd(x2, weight.contiguous(), eps)
+            ctx.save_for_backward(x2, weight, rrms)
+            ctx.eps = eps
+            ctx.hip = True
+            return out
+        # reference (fp32 math)
+        xf = x.float()
+        rrms = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+        out = (xf * rrms * weight.float()).to(x.dtype)
+        ctx.save_for_backward(x, weight, rrms.squeeze(-1))
+        ctx.eps = eps
+        ctx.hip = False
+        return out
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        x, w, rrms = ctx.saved_tensors
+        if ctx.hip:
+            dx, dw = _ext().rmsnorm_bwd(dy.contiguous(), x, w, rrms)
+            return dx, dw.to(w.dtype), None
+        xf = x.float()
+        dyf = dy.float()
+        wf = w.float()
+        r = rrms.unsqueeze(-1)
+        H = x.shape[-1]
+        dot = (dyf * wf * xf).sum(-1, keepdim=True)
+        dx = (r * wf * dyf - xf * dot * r.pow(3) / H).to(x.dtype)
+        dw = (dyf * xf * r).reshape(-1, H).sum(0).to(w.dtype)
+        return dx, dw, None
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    return _RMSNorm.apply(x, weight, eps)
+
+
+# ---------------------------------------------------------------------------
+# RoPE
+# ---------------------------------------------------------------------------
+def build_rope_table(
+    seq_len: int, head_dim: int, theta: float = 500000.0, device="cpu"
+) -> torch.Tensor:
+    """[S, D/2, 2] fp32 (cos, sin) — host-precomputed (guide App. B: no
+    on-device trig)."""
+    rot = head_dim // 2
+    # explicit cpu device so meta-device init contexts don't capture these
+    inv = 1.0 / (theta ** (torch.arange(0, rot, dtype=torch.float64, device="cpu") / rot))
+    pos = torch.arange(seq_len, dtype=torch.float64, device="cpu")
+    ang = torch.outer(pos, inv)
+    tab = torch.stack([ang.cos(), ang.sin()], dim=-1).to(torch.float32)
+    return tab.to(device)
+
+
+class _RoPE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, table: torch.Tensor, pos_offset: int):
+        ctx.pos_offset = pos_offset
+        if _use_hip(x, table):
+            ctx.hip = True
+            ctx.save_for_backward(table)
+            return _ext().rope(x.contiguous(), table, pos_offset, False, False)
+        ctx.hip = False
+        ctx.save_for_backward(table)
+        return _rope_ref(x, table, pos_offset, False)
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        (table,) = ctx.saved_tensors
+        if ctx.hip:
+            return _ext().rope(dy.contiguous(), table, ctx.pos_offset, True, False), None, None
+        return _rope_ref(dy, table, ctx.pos_offset, True), None, None
+
+
+def _rope_ref(x, table, pos_offset, backward):
+    B, S, H, D = x.shape
+    rot = D // 2
+    tb = table[pos_offset : pos_offset + S]  # [S, rot, 2]
+    cos = tb[..., 0].unsqueeze(0).unsqueeze(2)  # [1,S,1,rot]
+    sin = tb[..., 1].unsqueeze(0).unsqueeze(2)
+    if backward:
+        sin = -sin
+    xf = x.float()
+    x0, x1 = xf[..., :rot], xf[..., rot:]
+    o0 = x0 * cos - x1 * sin
+    o1 = x1 * cos + x0 * sin
+    return torch.cat([o0, o1], dim=-1).to(x.dtype)
+
+
+def rope_apply(x: torch.Tensor, table: torch.Tensor, pos_offset: int = 0) -> torch.Tensor:
+    """x: [B, S, H, D]."""
+    return _RoPE.apply(x, table, pos_offset)
+
+
+# ---------------------------------------------------------------------------
+# SwiGLU
+# ---------------------------------------------------------------------------
+class _SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate: torch.Tensor, up: torch.Tensor):
+        if _use_hip(gate, up):
+            g = gate.contiguous()
+            u = up.contiguous()
+            ctx.save_for_backward(g, u)
+            ctx.hip = True
+            return _ext().swiglu_fwd(g, u)
+        ctx.save_for_backward(gate, up)
+        ctx.hip = False
+        return (torch.nn.functional.silu(gate.float()) * up.float()).to(gate.dtype)
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        gate, up = ctx.saved_tensors
+        if ctx.hip:
+            dgate, dup = _ext().swiglu_bwd(dy.contiguous(), gate, up)
+            return dgate, dup
+        g = gate.float()
+        u = up.float()
+        d = dy.float()
+        sig = torch.sigmoid(g)
+        dgate = (d * u * sig * (1 + g * (1 - sig))).to(gate.dtype)
+        dup = (d * g * sig).to(up.dtype)
+        return dgate, dup
+
+
+def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    return _SwiGLU.apply(gate, up)
+
+
+# ---------------------------------------------------------------------------
+# Fused cross-entropy (mean over non-ignored tokens)
+# ---------------------------------------------------------------------------
+class _FusedCE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits: torch.Tensor, target: torch.Tensor, ignore_index: int):
+        n_tok = (target != ignore_index).sum().clamp(min=1)
+        if _use_hip(logits):
+            lg = logits.contiguous()
+            loss, lse = _ext().ce_fwd(lg, target.contiguous(), ignore_index)
+            ctx.save_for_backward(lg, target, lse, n_tok)
+            ctx.ignore_index = ignore_index
+            ctx.hip = True
+            return loss.sum() / n_tok.to(loss.dtype)
+        xf = logits.float()
+        lse = torch.logsumexp(xf, dim=-1)
+        valid = target != ignore_index
+        tgt = target.clamp(min=0)
+        xt = xf.gather(-1, tgt.unsqueeze(-1)).squeeze(-1)
+        loss = torch.where(valid, lse - xt, torch.zeros_like(lse))
+        ctx.save_for_backward(logits, target, lse, n_tok)
+        ctx.ignore_index = ignore_index
+        ctx.hip = False
+        return loss.sum() / n_tok.to(loss.dtype)
+
+    @staticmethod
+    def backward(ctx, dloss: torch.Tensor):
+        logits, target, lse, n_tok = ctx.saved_tensors
+        scale = (dloss / n_tok.to(dloss.dtype)).float()
+        if ctx.hip:
+            drow = scale.expand(lse.shape[0]).contiguous()
+            dlogits = _ext().ce_bwd(
+                logits, target, lse, drow, ctx.ignore_index, False
+            )
+            return dlogits, None, None
+        xf = logits.float()
+        p = torch.exp(xf - lse.unsqueeze(-1))
+        valid = (target != ctx.ignore_index).unsqueeze(-1)
+        tgt = target.clamp(min=0)
+        p.scatter_add_(
+            -1, tgt.unsqueeze(-1), -torch.ones_like(tgt, dtype=p.dtype).unsqueeze(-1)
+        )
+        d = (p * scale * valid).to(logits.dtype)
+        return d, None, None
+
+
+def fused_cross_entropy(
+    logits: torch.Tensor, target: torch.Tensor, ignore_index: int = -100
+) -> torch.Tensor:
+    """logits [N, V] (any leading dims flattened by caller), target [N]."""
+    return _FusedCE.apply(logits, target, ignore_index)
+
+
+# ---------------------------------------------------------------------------
+# flat AdamW + grad utilities (no autograd)
+# ---------------------------------------------------------------------------
+@torch.no_grad()
+def adamw_step_flat(
+    param_bf16: torch.Tensor,
+    master_f32: Optional[torch.Tensor],
+    grad: torch.Tensor,
+    m: torch.Tensor,
+    v: torch.Tensor,
+    *,
+    lr: float,
+    beta1: float = 0.9,
+    beta2: float = 0.95,
+    eps: float = 1e-8,
+    weight_decay: float = 0.0,
+    step: int = 1,
+    grad_scale: float = 1.0,
+):
+    if _use_hip(param_bf16, grad, m, v):
+        _ext().adamw_step(
+            param_bf16, master_f32, grad, m, v, lr, beta1, beta2, eps,
+            weight_decay, step, grad_scale,
+        )
+        return
+    g = grad.float() * grad_scale
+    p = master_f32 if master_f32 is not None else param_bf16.float()
+    p.mul_(1 - lr * weight_decay)
+    m.mul_(beta1).add_(g, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+    bc1 = 1 - beta1**step
+    bc2 = 1 - beta2**step
+    denom = (v / bc2).sqrt().add_(eps)
+    p.addcdiv_(m, denom, value=-lr / bc1)
+    param_bf16.copy_(p.to(param_bf16.dtype))
+
+
+@torch.no_grad()
+def l2norm_sq(x: torch.Tensor) -> torch.Tensor:
+    if _use_hip(x):
+        return _ext().l2norm_sq(x).squeeze(0)
+    return x.float().pow(2).sum()
+
+
+@torch.no_grad()
+def scale_flat_(x: torch.Tensor, scale) -> None:
+    if _use_hip(x):
+        if isinstance(scale, torch.Tensor):
+            _ext().scale_(x, scale.float().contiguous(), 1.0)
+        else:
+            _ext().scale_(x, None, float(scale))
+        return
+    if isinstance(scale, torch.Tensor):
+        x.mul_(scale.to(x.dtype))
+    else:
+        x.mul_(scale)
