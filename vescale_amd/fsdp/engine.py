@@ -188,11 +188,11 @@ class FSDP(nn.Module):
 
         module.to(dev)
         if param_dtype is not None:
-            module.to(dtype=param_dtype)
-            # keep fp32 buffers (rope table) in fp32
-            for b in module.buffers():
-                if b.dtype == param_dtype and b.is_floating_point():
-                    pass
+            # convert parameters only — buffers (e.g. fp32 RoPE tables)
+            # keep their dtype
+            for p in module.parameters():
+                if p.is_floating_point():
+                    p.data = p.data.to(param_dtype)
         self.units: List[FSDPUnit] = []
         self._build_units(unit_classes)
         self._install_hooks()
