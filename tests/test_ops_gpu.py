@@ -81,6 +81,45 @@ def test_swiglu():
 
 
 @requires_gpu
+def test_swiglu_packed():
+    from vescale_amd.ops import swiglu_packed
+
+    torch.manual_seed(21)
+    gu = torch.randn(64, 2048, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    out = swiglu_packed(gu)
+    dy = torch.randn_like(out)
+    out.backward(dy)
+    F = 1024
+    gf = gu.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.silu(gf[..., :F]) * gf[..., F:]
+    ref.backward(dy.float())
+    assert torch.allclose(out.float(), ref.detach(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(gu.grad.float(), gf.grad, atol=3e-2, rtol=3e-2)
+
+
+@requires_gpu
+def test_adamw_clip_fused():
+    from vescale_amd.ops import adamw_step_flat
+
+    torch.manual_seed(22)
+    n = 4096
+    p = torch.randn(n, device="cuda").bfloat16()
+    master = p.float().clone()
+    g = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    m = torch.zeros(n, device="cuda")
+    v = torch.zeros(n, device="cuda")
+    clip = torch.tensor([0.5], device="cuda")
+    adamw_step_flat(p, master, g, m, v, lr=1e-3, step=1, clip_scale=clip)
+    # reference with pre-scaled grad
+    p2 = torch.randn(0)
+    m2 = torch.zeros(n, device="cuda")
+    v2 = torch.zeros(n, device="cuda")
+    gf = g.float() * 0.5
+    m2.add_(gf, alpha=0.1)
+    assert torch.allclose(m, m2, atol=1e-6)
+
+
+@requires_gpu
 def test_fused_cross_entropy():
     from vescale_amd.ops import fused_cross_entropy
 

@@ -61,8 +61,6 @@ class FlatAdamW:
             if u.grad_shard is None:
                 continue
             st = self.state[u.name]
-            if clip_scale_t is not None:
-                scale_flat_(u.grad_shard, clip_scale_t)
             adamw_step_flat(
                 u.shard,
                 st.get("master"),
@@ -76,6 +74,7 @@ class FlatAdamW:
                 weight_decay=self.weight_decay,
                 step=self.step_count,
                 grad_scale=scale,
+                clip_scale=clip_scale_t,  # fused into the kernel (one read)
             )
         self.zero_grad()
 

@@ -17,7 +17,13 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops import build_rope_table, fused_cross_entropy, rmsnorm, rope_apply, swiglu
+from ..ops import (
+    build_rope_table,
+    fused_cross_entropy,
+    rmsnorm,
+    rope_qkv,
+    swiglu_packed,
+)
 
 
 @dataclass
@@ -40,10 +46,9 @@ class LlamaConfig:
     def num_params(self) -> int:
         d, f, v = self.dim, self.ffn_dim, self.vocab_size
         per_layer = (
-            d * d  # wq
-            + 2 * d * (self.n_kv_heads * self.head_dim)  # wk, wv
+            d * (d + 2 * self.n_kv_heads * self.head_dim)  # wqkv (fused)
             + d * d  # wo
-            + 3 * d * f  # w1, w3, w2
+            + 3 * d * f  # w13 (fused gate+up) + w2
             + 2 * d  # norms
         )
         emb = v * d * (1 if self.tie_embeddings else 2)
@@ -84,19 +89,17 @@ class Attention(nn.Module):
         self.cfg = cfg
         d, hd = cfg.dim, cfg.head_dim
         kv = cfg.n_kv_heads * hd
-        self.wq = nn.Linear(d, d, bias=False)
-        self.wk = nn.Linear(d, kv, bias=False)
-        self.wv = nn.Linear(d, kv, bias=False)
+        # fused qkv projection: one GEMM instead of three
+        self.wqkv = nn.Linear(d, d + 2 * kv, bias=False)
         self.wo = nn.Linear(d, d, bias=False)
 
     def forward(self, x: torch.Tensor, rope_table: torch.Tensor) -> torch.Tensor:
         B, S, _ = x.shape
         cfg = self.cfg
-        q = self.wq(x).view(B, S, cfg.n_heads, cfg.head_dim)
-        k = self.wk(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
-        v = self.wv(x).view(B, S, cfg.n_kv_heads, cfg.head_dim)
-        q = rope_apply(q, rope_table)
-        k = rope_apply(k, rope_table)
+        qkv = self.wqkv(x)
+        q, k, v = rope_qkv(
+            qkv, rope_table, cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
+        )
         q = q.transpose(1, 2)  # [B, H, S, D]
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)
@@ -114,9 +117,7 @@ class FeedForward(nn.Module):
         self.ffn_dim = cfg.ffn_dim
 
     def forward(self, x):
-        gu = self.w13(x)
-        gate, up = gu.split(self.ffn_dim, dim=-1)
-        return self.w2(swiglu(gate.contiguous(), up.contiguous()))
+        return self.w2(swiglu_packed(self.w13(x)))
 
 
 class TransformerBlock(nn.Module):

@@ -40,8 +40,10 @@ from .functional import (  # noqa: E402,F401
     fused_cross_entropy,
     rmsnorm,
     rope_apply,
+    rope_qkv,
     build_rope_table,
     swiglu,
+    swiglu_packed,
     adamw_step_flat,
     l2norm_sq,
     scale_flat_,

@@ -2,12 +2,18 @@
 //
 // The FSDP engine keeps params/grads/opt-state as FLAT per-rank shards
 // (RaggedShard layout), so the optimizer is a single fused kernel per
-// buffer — no multi-tensor-apply bookkeeping needed (SURVEY.md §2.7
-// "_fused_adamw_ ... over ragged local shards").  bf16 params + fp32
-// master weights + fp32 m/v; decoupled weight decay; bias correction.
+// buffer (SURVEY.md §2.7 "_fused_adamw_ ... over ragged local shards").
+// Buffers are 256-element aligned (engine ALIGN), so the kernel runs a
+// fully-vectorized main body: short8 bf16 loads, float4 state loads
+// (Guideline 13).  Traffic/param: r(g2 + m4 + v4 + p4) + w(m4 + v4 + p4 +
+// pb2) = 28 B -> roofline ~36 ms for 8B params at 6.3 TB/s.
+// The grad-clip scale is FUSED via clip_ptr (one global read) instead of a
+// separate full-sweep scale pass.
 #include "common.h"
 
 #define BLOCK 256
+
+typedef float float4vv __attribute__((ext_vector_type(4)));
 
 extern "C" __global__ void __launch_bounds__(BLOCK)
 adamw_flat_bf16(unsigned short* __restrict__ param_bf16,
@@ -18,29 +24,82 @@ adamw_flat_bf16(unsigned short* __restrict__ param_bf16,
                 float* __restrict__ v,
                 int64_t n, float lr, float beta1, float beta2, float eps,
                 float weight_decay, float bc1, float bc2,  // 1-beta^t
-                float grad_scale) {
-  const int vec = 4;
+                float grad_scale, const float* __restrict__ clip_ptr) {
+  const float step_size = lr / bc1;
+  const float gs = grad_scale * (clip_ptr ? *clip_ptr : 1.0f);
+  const float inv_bc2 = 1.0f / bc2;
+  const float wd = 1.0f - lr * weight_decay;
+  const int vec = 8;
   int64_t i0 = (int64_t)(blockIdx.x * BLOCK + threadIdx.x) * vec;
   int64_t stride = (int64_t)gridDim.x * BLOCK * vec;
-  const float step_size = lr / bc1;
-  for (int64_t i = i0; i < n; i += stride) {
+  int64_t nv = n / vec * vec;
+  for (int64_t i = i0; i < nv; i += stride) {
+    float g[8];
+    if (grad_f32) {
+      float4vv g0 = *reinterpret_cast<const float4vv*>(grad_f32 + i);
+      float4vv g1 = *reinterpret_cast<const float4vv*>(grad_f32 + i + 4);
 #pragma unroll
-    for (int j = 0; j < vec; ++j) {
-      int64_t k = i + j;
-      if (k >= n) break;
-      float g = grad_f32 ? grad_f32[k] : bf16_to_f32(grad_bf16[k]);
-      g *= grad_scale;
-      float p = master ? master[k] : bf16_to_f32(param_bf16[k]);
-      p -= lr * weight_decay * p;
-      float mk = beta1 * m[k] + (1.f - beta1) * g;
-      float vk = beta2 * v[k] + (1.f - beta2) * g * g;
-      m[k] = mk;
-      v[k] = vk;
-      float denom = sqrtf(vk / bc2) + eps;
-      p -= step_size * mk / denom;
-      if (master) master[k] = p;
-      param_bf16[k] = f32_to_bf16(p);
+      for (int j = 0; j < 4; ++j) { g[j] = g0[j]; g[j + 4] = g1[j]; }
+    } else {
+      short8v gv = *reinterpret_cast<const short8v*>(grad_bf16 + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) g[j] = bf16_to_f32((unsigned short)gv[j]);
     }
+    float4vv m0 = *reinterpret_cast<const float4vv*>(m + i);
+    float4vv m1 = *reinterpret_cast<const float4vv*>(m + i + 4);
+    float4vv v0 = *reinterpret_cast<const float4vv*>(v + i);
+    float4vv v1 = *reinterpret_cast<const float4vv*>(v + i + 4);
+    float p[8];
+    if (master) {
+      float4vv p0 = *reinterpret_cast<const float4vv*>(master + i);
+      float4vv p1 = *reinterpret_cast<const float4vv*>(master + i + 4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) { p[j] = p0[j]; p[j + 4] = p1[j]; }
+    } else {
+      short8v pv = *reinterpret_cast<const short8v*>(param_bf16 + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) p[j] = bf16_to_f32((unsigned short)pv[j]);
+    }
+    float mm[8], vv[8];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) { mm[j] = m0[j]; mm[j + 4] = m1[j]; vv[j] = v0[j]; vv[j + 4] = v1[j]; }
+    short8v pout;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gj = g[j] * gs;
+      float pj = p[j] * wd;
+      float mj = beta1 * mm[j] + (1.f - beta1) * gj;
+      float vj = beta2 * vv[j] + (1.f - beta2) * gj * gj;
+      pj -= step_size * mj / (sqrtf(vj * inv_bc2) + eps);
+      mm[j] = mj; vv[j] = vj; p[j] = pj;
+      pout[j] = (short)f32_to_bf16(pj);
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) { m0[j] = mm[j]; m1[j] = mm[j + 4]; v0[j] = vv[j]; v1[j] = vv[j + 4]; }
+    *reinterpret_cast<float4vv*>(m + i) = m0;
+    *reinterpret_cast<float4vv*>(m + i + 4) = m1;
+    *reinterpret_cast<float4vv*>(v + i) = v0;
+    *reinterpret_cast<float4vv*>(v + i + 4) = v1;
+    if (master) {
+      float4vv p0, p1;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) { p0[j] = p[j]; p1[j] = p[j + 4]; }
+      *reinterpret_cast<float4vv*>(master + i) = p0;
+      *reinterpret_cast<float4vv*>(master + i + 4) = p1;
+    }
+    *reinterpret_cast<short8v*>(param_bf16 + i) = pout;
+  }
+  // scalar tail (generic callers; engine buffers are 256-aligned)
+  for (int64_t k = nv + blockIdx.x * BLOCK + threadIdx.x; k < n;
+       k += (int64_t)gridDim.x * BLOCK) {
+    float gj = (grad_f32 ? grad_f32[k] : bf16_to_f32(grad_bf16[k])) * gs;
+    float pj = (master ? master[k] : bf16_to_f32(param_bf16[k])) * wd;
+    float mj = beta1 * m[k] + (1.f - beta1) * gj;
+    float vj = beta2 * v[k] + (1.f - beta2) * gj * gj;
+    pj -= step_size * mj / (sqrtf(vj * inv_bc2) + eps);
+    m[k] = mj; v[k] = vj;
+    if (master) master[k] = pj;
+    param_bf16[k] = f32_to_bf16(pj);
   }
 }
 
@@ -54,15 +113,27 @@ l2norm_sq_flat(const unsigned short* __restrict__ x_bf16,
   const int vec = 8;
   int64_t i0 = (int64_t)(blockIdx.x * BLOCK + threadIdx.x) * vec;
   int64_t stride = (int64_t)gridDim.x * BLOCK * vec;
+  int64_t nv = n / vec * vec;
   float ss = 0.f;
-  for (int64_t i = i0; i < n; i += stride) {
+  for (int64_t i = i0; i < nv; i += stride) {
+    if (x_f32) {
+      float4vv a = *reinterpret_cast<const float4vv*>(x_f32 + i);
+      float4vv b = *reinterpret_cast<const float4vv*>(x_f32 + i + 4);
 #pragma unroll
-    for (int j = 0; j < vec; ++j) {
-      int64_t k = i + j;
-      if (k >= n) break;
-      float f = x_f32 ? x_f32[k] : bf16_to_f32(x_bf16[k]);
-      ss += f * f;
+      for (int j = 0; j < 4; ++j) ss += a[j] * a[j] + b[j] * b[j];
+    } else {
+      short8v v = *reinterpret_cast<const short8v*>(x_bf16 + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf16_to_f32((unsigned short)v[j]);
+        ss += f * f;
+      }
     }
+  }
+  for (int64_t k = nv + blockIdx.x * BLOCK + threadIdx.x; k < n;
+       k += (int64_t)gridDim.x * BLOCK) {
+    float f = x_f32 ? x_f32[k] : bf16_to_f32(x_bf16[k]);
+    ss += f * f;
   }
   float total = block_reduce_sum<BLOCK>(ss, lds);
   if (threadIdx.x == 0) atomicAdd(out, total);
@@ -76,13 +147,26 @@ scale_flat(unsigned short* __restrict__ x_bf16, float* __restrict__ x_f32,
   float s = scale_ptr ? *scale_ptr : scale_const;
   int64_t i0 = (int64_t)(blockIdx.x * BLOCK + threadIdx.x) * vec;
   int64_t stride = (int64_t)gridDim.x * BLOCK * vec;
-  for (int64_t i = i0; i < n; i += stride) {
+  int64_t nv = n / vec * vec;
+  for (int64_t i = i0; i < nv; i += stride) {
+    if (x_f32) {
+      float4vv a = *reinterpret_cast<const float4vv*>(x_f32 + i);
+      float4vv b = *reinterpret_cast<const float4vv*>(x_f32 + i + 4);
 #pragma unroll
-    for (int j = 0; j < vec; ++j) {
-      int64_t k = i + j;
-      if (k >= n) break;
-      if (x_f32) x_f32[k] *= s;
-      else x_bf16[k] = f32_to_bf16(bf16_to_f32(x_bf16[k]) * s);
+      for (int j = 0; j < 4; ++j) { a[j] *= s; b[j] *= s; }
+      *reinterpret_cast<float4vv*>(x_f32 + i) = a;
+      *reinterpret_cast<float4vv*>(x_f32 + i + 4) = b;
+    } else {
+      short8v v = *reinterpret_cast<const short8v*>(x_bf16 + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        v[j] = (short)f32_to_bf16(bf16_to_f32((unsigned short)v[j]) * s);
+      *reinterpret_cast<short8v*>(x_bf16 + i) = v;
     }
+  }
+  for (int64_t k = nv + blockIdx.x * BLOCK + threadIdx.x; k < n;
+       k += (int64_t)gridDim.x * BLOCK) {
+    if (x_f32) x_f32[k] *= s;
+    else x_bf16[k] = f32_to_bf16(bf16_to_f32(x_bf16[k]) * s);
   }
 }

@@ -6,6 +6,7 @@
 
 #include "rmsnorm.hip"
 #include "rope.hip"
+#include "rope_qkv.hip"
 #include "swiglu.hip"
 #include "adamw.hip"
 #include "cross_entropy.hip"
@@ -82,6 +83,44 @@ at::Tensor rope(at::Tensor x, at::Tensor table, int64_t pos_offset,
   return out;
 }
 
+std::vector<at::Tensor> rope_qkv_fwd(at::Tensor qkv, at::Tensor table,
+                                     int64_t B, int64_t S, int64_t Hq,
+                                     int64_t Hkv, int64_t D,
+                                     int64_t pos_offset) {
+  check_bf16_contig(qkv, "qkv");
+  int64_t T = B * S;
+  auto opt = qkv.options();
+  auto q = at::empty({B, S, Hq, D}, opt);
+  auto k = at::empty({B, S, Hkv, D}, opt);
+  auto v = at::empty({B, S, Hkv, D}, opt);
+  int64_t total = T * (Hq + 2 * Hkv) * (D / 2);
+  int grid = grid_for(total, 256);
+  hipLaunchKernelGGL(rope_qkv_fwd_bf16, dim3(grid), dim3(256), 0,
+                     cur_stream(), (const unsigned short*)qkv.data_ptr(),
+                     (unsigned short*)q.data_ptr(),
+                     (unsigned short*)k.data_ptr(),
+                     (unsigned short*)v.data_ptr(), table.data_ptr<float>(),
+                     T, (int)S, (int)Hq, (int)Hkv, (int)D, (int)pos_offset);
+  return {q, k, v};
+}
+
+at::Tensor rope_qkv_bwd(at::Tensor dq, at::Tensor dk, at::Tensor dv,
+                        at::Tensor table, int64_t B, int64_t S, int64_t Hq,
+                        int64_t Hkv, int64_t D, int64_t pos_offset) {
+  int64_t T = B * S;
+  auto dqkv = at::empty({B, S, (Hq + 2 * Hkv) * D}, dq.options());
+  int64_t total = T * (Hq + 2 * Hkv) * (D / 2);
+  int grid = grid_for(total, 256);
+  hipLaunchKernelGGL(rope_qkv_bwd_bf16, dim3(grid), dim3(256), 0,
+                     cur_stream(),
+                     (const unsigned short*)dq.contiguous().data_ptr(),
+                     (const unsigned short*)dk.contiguous().data_ptr(),
+                     (const unsigned short*)dv.contiguous().data_ptr(),
+                     (unsigned short*)dqkv.data_ptr(), table.data_ptr<float>(),
+                     T, (int)S, (int)Hq, (int)Hkv, (int)D, (int)pos_offset);
+  return dqkv;
+}
+
 // ------------------------------ SwiGLU ------------------------------
 at::Tensor swiglu_fwd(at::Tensor gate, at::Tensor up) {
   check_bf16_contig(gate, "gate");
@@ -94,6 +133,35 @@ at::Tensor swiglu_fwd(at::Tensor gate, at::Tensor up) {
                      (const unsigned short*)up.data_ptr(),
                      (unsigned short*)out.data_ptr(), n);
   return out;
+}
+
+at::Tensor swiglu_packed_fwd(at::Tensor gu) {
+  check_bf16_contig(gu, "gu");
+  int F2 = (int)gu.size(-1);
+  TORCH_CHECK(F2 % 16 == 0, "packed ffn dim must be multiple of 16");
+  int F = F2 / 2;
+  int64_t rows = gu.numel() / F2;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = F;
+  auto out = at::empty(sizes, gu.options());
+  int grid = grid_for(rows * (F / 8), 256);
+  hipLaunchKernelGGL(swiglu_packed_fwd_bf16, dim3(grid), dim3(256), 0,
+                     cur_stream(), (const unsigned short*)gu.data_ptr(),
+                     (unsigned short*)out.data_ptr(), rows, F);
+  return out;
+}
+
+at::Tensor swiglu_packed_bwd(at::Tensor dy, at::Tensor gu) {
+  int F2 = (int)gu.size(-1);
+  int F = F2 / 2;
+  int64_t rows = gu.numel() / F2;
+  auto dgu = at::empty_like(gu);
+  int grid = grid_for(rows * (F / 8), 256);
+  hipLaunchKernelGGL(swiglu_packed_bwd_bf16, dim3(grid), dim3(256), 0,
+                     cur_stream(), (const unsigned short*)dy.contiguous().data_ptr(),
+                     (const unsigned short*)gu.data_ptr(),
+                     (unsigned short*)dgu.data_ptr(), rows, F);
+  return dgu;
 }
 
 std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor gate, at::Tensor up) {
@@ -146,7 +214,8 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
 void adamw_step(at::Tensor param, c10::optional<at::Tensor> master,
                 at::Tensor grad, at::Tensor m, at::Tensor v, double lr,
                 double beta1, double beta2, double eps, double weight_decay,
-                int64_t step, double grad_scale) {
+                int64_t step, double grad_scale,
+                c10::optional<at::Tensor> clip_scale) {
   check_bf16_contig(param, "param");
   int64_t n = param.numel();
   float bc1 = 1.f - powf((float)beta1, (float)step);
@@ -156,12 +225,13 @@ void adamw_step(at::Tensor param, c10::optional<at::Tensor> master,
   if (grad.scalar_type() == at::kBFloat16) gb = (const unsigned short*)grad.data_ptr();
   else gf = grad.data_ptr<float>();
   float* mp = master.has_value() ? master->data_ptr<float>() : nullptr;
-  int grid = grid_for(n / 4, 256);
+  const float* cp = clip_scale.has_value() ? clip_scale->data_ptr<float>() : nullptr;
+  int grid = grid_for(n / 8, 256);
   hipLaunchKernelGGL(adamw_flat_bf16, dim3(grid), dim3(256), 0, cur_stream(),
                      (unsigned short*)param.data_ptr(), mp, gb, gf,
                      m.data_ptr<float>(), v.data_ptr<float>(), n, (float)lr,
                      (float)beta1, (float)beta2, (float)eps,
-                     (float)weight_decay, bc1, bc2, (float)grad_scale);
+                     (float)weight_decay, bc1, bc2, (float)grad_scale, cp);
 }
 
 at::Tensor l2norm_sq(at::Tensor x) {
@@ -278,8 +348,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("rope", &rope);
+  m.def("rope_qkv_fwd", &rope_qkv_fwd);
+  m.def("rope_qkv_bwd", &rope_qkv_bwd);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("swiglu_packed_fwd", &swiglu_packed_fwd);
+  m.def("swiglu_packed_bwd", &swiglu_packed_bwd);
   m.def("ce_fwd", &ce_fwd);
   m.def("ce_bwd", &ce_bwd);
   m.def("adamw_step", &adamw_step);

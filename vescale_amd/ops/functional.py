@@ -128,6 +128,52 @@ def rope_apply(x: torch.Tensor, table: torch.Tensor, pos_offset: int = 0) -> tor
     return _RoPE.apply(x, table, pos_offset)
 
 
+class _RoPEQKV(torch.autograd.Function):
+    """Fused qkv-split + RoPE (packed wqkv GEMM output -> q/k/v)."""
+
+    @staticmethod
+    def forward(ctx, qkv: torch.Tensor, table: torch.Tensor, n_heads: int,
+                n_kv_heads: int, head_dim: int, pos_offset: int):
+        B, S, _ = qkv.shape
+        ctx.dims = (B, S, n_heads, n_kv_heads, head_dim, pos_offset)
+        ctx.save_for_backward(table)
+        if _use_hip(qkv, table):
+            ctx.hip = True
+            q, k, v = _ext().rope_qkv_fwd(
+                qkv.contiguous(), table, B, S, n_heads, n_kv_heads, head_dim,
+                pos_offset,
+            )
+            return q, k, v
+        ctx.hip = False
+        d = n_heads * head_dim
+        kv = n_kv_heads * head_dim
+        q = qkv[..., :d].reshape(B, S, n_heads, head_dim)
+        k = qkv[..., d : d + kv].reshape(B, S, n_kv_heads, head_dim)
+        v = qkv[..., d + kv :].reshape(B, S, n_kv_heads, head_dim).contiguous()
+        q = _rope_ref(q, table, pos_offset, False)
+        k = _rope_ref(k, table, pos_offset, False)
+        return q, k, v
+
+    @staticmethod
+    def backward(ctx, dq, dk, dv):
+        (table,) = ctx.saved_tensors
+        B, S, Hq, Hkv, D, pos = ctx.dims
+        if ctx.hip:
+            dqkv = _ext().rope_qkv_bwd(dq, dk, dv, table, B, S, Hq, Hkv, D, pos)
+            return dqkv, None, None, None, None, None
+        dqr = _rope_ref(dq, table, pos, True).reshape(B, S, Hq * D)
+        dkr = _rope_ref(dk, table, pos, True).reshape(B, S, Hkv * D)
+        dvr = dv.reshape(B, S, Hkv * D)
+        return torch.cat([dqr, dkr, dvr], dim=-1), None, None, None, None, None
+
+
+def rope_qkv(qkv: torch.Tensor, table: torch.Tensor, n_heads: int,
+             n_kv_heads: int, head_dim: int, pos_offset: int = 0):
+    """qkv: [B, S, (Hq+2*Hkv)*D] packed -> (q, k, v) in [B, S, H, D], RoPE
+    applied to q and k."""
+    return _RoPEQKV.apply(qkv, table, n_heads, n_kv_heads, head_dim, pos_offset)
+
+
 # ---------------------------------------------------------------------------
 # SwiGLU
 # ---------------------------------------------------------------------------
@@ -161,6 +207,41 @@ class _SwiGLU(torch.autograd.Function):
 
 def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
     return _SwiGLU.apply(gate, up)
+
+
+class _SwiGLUPacked(torch.autograd.Function):
+    """gu = [.., 2F] (gate||up) -> [.., F]; avoids split+cat round trips."""
+
+    @staticmethod
+    def forward(ctx, gu: torch.Tensor):
+        if _use_hip(gu):
+            g = gu.contiguous()
+            ctx.save_for_backward(g)
+            ctx.hip = True
+            return _ext().swiglu_packed_fwd(g)
+        ctx.save_for_backward(gu)
+        ctx.hip = False
+        F = gu.shape[-1] // 2
+        gate, up = gu[..., :F], gu[..., F:]
+        return (torch.nn.functional.silu(gate.float()) * up.float()).to(gu.dtype)
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        (gu,) = ctx.saved_tensors
+        if ctx.hip:
+            return _ext().swiglu_packed_bwd(dy, gu)
+        F = gu.shape[-1] // 2
+        g = gu[..., :F].float()
+        u = gu[..., F:].float()
+        d = dy.float()
+        sig = torch.sigmoid(g)
+        dgate = d * u * sig * (1 + g * (1 - sig))
+        dup = d * g * sig
+        return torch.cat([dgate, dup], dim=-1).to(gu.dtype)
+
+
+def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
+    return _SwiGLUPacked.apply(gu)
 
 
 # ---------------------------------------------------------------------------
@@ -234,14 +315,17 @@ def adamw_step_flat(
     weight_decay: float = 0.0,
     step: int = 1,
     grad_scale: float = 1.0,
+    clip_scale: Optional[torch.Tensor] = None,
 ):
     if _use_hip(param_bf16, grad, m, v):
         _ext().adamw_step(
             param_bf16, master_f32, grad, m, v, lr, beta1, beta2, eps,
-            weight_decay, step, grad_scale,
+            weight_decay, step, grad_scale, clip_scale,
         )
         return
     g = grad.float() * grad_scale
+    if clip_scale is not None:
+        g = g * clip_scale.item()
     p = master_f32 if master_f32 is not None else param_bf16.float()
     p.mul_(1 - lr * weight_decay)
     m.mul_(beta1).add_(g, alpha=1 - beta1)
