@@ -1,0 +1,87 @@
+"""Forward hooks converting module inputs/outputs per the plan.
+
+Parity: legacy/vescale/dmodule/_hook.py:76-272 (PreHookInput /
+PostHookOutput).  Each hook converts positional tensor args:
+  plain tensor -> DTensor.from_local(t, mesh, placements)
+  DTensor      -> redistribute(placements)   [=> RCCL comm]
+A None placement entry leaves the arg untouched.
+"""
+from __future__ import annotations
+
+import re
+from typing import Any, Dict, List, Optional, Sequence
+
+import torch
+import torch.nn as nn
+
+from ..dtensor import DeviceMesh, DTensor
+
+
+def _convert(x, placements, mesh: DeviceMesh):
+    if placements is None or not isinstance(x, torch.Tensor):
+        return x
+    if isinstance(x, DTensor):
+        if tuple(x.placements) == tuple(placements):
+            return x
+        return x.redistribute(placements=list(placements))
+    return DTensor.from_local(x, mesh, list(placements))
+
+
+def _convert_seq(args, plan_list, mesh):
+    out = []
+    ti = 0
+    for a in args:
+        if isinstance(a, torch.Tensor):
+            pl = plan_list[ti] if plan_list is not None and ti < len(plan_list) else None
+            out.append(_convert(a, pl, mesh))
+            ti += 1
+        else:
+            out.append(a)
+    return tuple(out)
+
+
+def install_forward_hooks(root: nn.Module, mesh: DeviceMesh, fwd_plan: Dict[str, Any]):
+    input_plans: Dict[str, Any] = {}
+    output_plans: Dict[str, Any] = {}
+    weight_plans: Dict[str, Any] = {}
+    for k, v in fwd_plan.items():
+        if k.endswith(".input"):
+            input_plans[k[: -len(".input")]] = v
+        elif k.endswith(".output"):
+            output_plans[k[: -len(".output")]] = v
+        elif k.endswith(".weight_placement"):
+            weight_plans[k[: -len(".weight_placement")]] = v
+
+    def match(table, fqn):
+        for pattern, v in table.items():
+            if re.fullmatch(pattern, fqn):
+                return v
+        return None
+
+    for mod_name, mod in root.named_modules():
+        fqn = mod_name if mod_name else ""
+        ip = match(input_plans, fqn) if fqn or "" in input_plans else match(input_plans, fqn)
+        op = match(output_plans, fqn)
+        if ip is not None:
+            mod.register_forward_pre_hook(_make_pre_hook(ip, mesh))
+        if op is not None:
+            mod.register_forward_hook(_make_post_hook(op, mesh))
+
+
+def _make_pre_hook(plan_list, mesh):
+    def hook(mod, args):
+        return _convert_seq(args, plan_list, mesh)
+
+    return hook
+
+
+def _make_post_hook(plan_list, mesh):
+    def hook(mod, args, output):
+        if isinstance(output, torch.Tensor):
+            pl = plan_list[0] if plan_list else None
+            return _convert(output, pl, mesh)
+        if isinstance(output, (tuple, list)):
+            return type(output)(_convert_seq(output, plan_list, mesh))
+        return output
+
+    return hook

@@ -84,14 +84,30 @@ def _handler_embedding_dense_backward(dispatcher, op, args, kwargs):
     # the VocabParallelEmbedding module path — generic path reduces first.
     out_pl = []
     need_rep = []
+    g_ndim = grad_output._spec.ndim
+    coord = mesh.get_coordinate()
     for md, p in enumerate(gspec.placements):
         if isinstance(p, Partial):
             need_rep.append(md)
             out_pl.append(Replicate())
-        elif isinstance(p, Shard) and p.dim == local_g.ndim - 1:
+        elif isinstance(p, Shard) and p.dim == g_ndim - 1:
             out_pl.append(Shard(1))
         elif isinstance(p, (Shard, InterleavedShard)):
             out_pl.append(Partial("sum"))
+            # indices must be sliced to match the grad's batch/seq shard
+            if (
+                not isinstance(indices, DTensor)
+                and isinstance(p, Shard)
+                and p.dim < local_i.ndim
+                and local_i.shape[p.dim] == gspec.shape[p.dim]
+            ):
+                from ..placement_types import Shard as _S
+
+                total = local_i.shape[p.dim]
+                w = mesh.size(md)
+                off = _S.chunk_offset(total, w, coord[md])
+                sz = _S.chunk_size(total, w, coord[md])
+                local_i = local_i.narrow(p.dim, off, sz)
         else:
             out_pl.append(Replicate())
     if need_rep:

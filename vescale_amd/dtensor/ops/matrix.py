@@ -16,10 +16,30 @@ import torch
 from .._collective_utils import allgather_cost, allreduce_cost
 from .._dtensor_spec import DTensorSpec
 from .._op_schema import OpSchema, OutputSharding
-from ..placement_types import Partial, Placement, Replicate, Shard
+from ..placement_types import InterleavedShard, Partial, Placement, Replicate, Shard
 from .common import out_spec
 
 aten = torch.ops.aten
+
+
+def _is_cands(specs, b_idx: int, out_ndim: int, bias_idx=None):
+    """Extra candidates for InterleavedShard weights (packed-QKV colwise
+    TP): b placed IS(1, y) -> out IS(last, y) with zero movement."""
+    out = []
+    b = specs[b_idx]
+    for md in range(b.mesh.ndim):
+        p = b.placements[md]
+        if isinstance(p, InterleavedShard) and p.dim == b.ndim - 1:
+            cand_in = []
+            for i, s in enumerate(specs):
+                if i == b_idx:
+                    cand_in.append(p)
+                elif bias_idx is not None and i == bias_idx:
+                    cand_in.append(InterleavedShard(s.ndim - 1, p.interleaved_size))
+                else:
+                    cand_in.append(Replicate())
+            out.append((tuple(cand_in), InterleavedShard(out_ndim - 1, p.interleaved_size)))
+    return out
 
 
 def _move_cost(cur: Placement, want: Placement, nbytes_gb: float, w: int) -> float:
@@ -30,7 +50,9 @@ def _move_cost(cur: Placement, want: Placement, nbytes_gb: float, w: int) -> flo
     if isinstance(cur, Partial):
         return allreduce_cost(nbytes_gb, w)
     if isinstance(cur, Replicate):
-        return 0.0  # local slice
+        # local slice — cheap, but never FREE: an all-replicate op must not
+        # spontaneously shard its output (placement-preserving tie-break)
+        return 1e-3
     return allgather_cost(nbytes_gb, w) * 1.5
 
 
@@ -73,7 +95,7 @@ def mm_rule(schema: OpSchema) -> OutputSharding:
         ((R, Shard(1)), Shard(1)),
         ((Shard(1), Shard(0)), Partial("sum")),
         ((R, R), R),
-    ]
+    ] + _is_cands([a, b], 1, 2)
     targets, outs = _pick(mesh, [a, b], cands)
     osp = out_spec(mesh, outs, (M, N), a.dtype)
     return OutputSharding(osp, [tuple(t) for t in targets])
@@ -99,6 +121,7 @@ def addmm_rule(schema: OpSchema) -> OutputSharding:
         return p
 
     cands = [((fix_bias(ci[0]), ci[1], ci[2]), co) for ci, co in cands]
+    cands += _is_cands([bias, a, b], 2, 2, bias_idx=0)
     targets, outs = _pick(mesh, [bias, a, b], cands)
     osp = out_spec(mesh, outs, (M, N), a.dtype)
     return OutputSharding(osp, [tuple(t) for t in targets])

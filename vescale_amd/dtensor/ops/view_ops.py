@@ -252,7 +252,32 @@ def cat_rule(schema: OpSchema) -> OutputSharding:
     d = schema.args_schema[1] if len(schema.args_schema) > 1 else 0
     d = _norm_dim(d, specs[0].ndim)
     mesh = specs[0].mesh
-    # all inputs align to the first's placements; cat dim must not be sharded
+    # special case: cat of equal-size Shard(d) chunks ON the cat dim is the
+    # inverse of the interleaved split -> output InterleavedShard(d, n)
+    # (packed-QKV backward path; no communication)
+    def shard_on_d(sp, md):
+        p = sp.placements[md]
+        return isinstance(p, Shard) and not isinstance(p, InterleavedShard) and p.dim == d
+
+    same_size = all(tuple(sp.shape) == tuple(specs[0].shape) for sp in specs)
+    il_base: List[Placement] = []
+    ok_il = same_size and len(specs) > 1
+    if ok_il:
+        for md in range(mesh.ndim):
+            if all(shard_on_d(sp, md) for sp in specs):
+                il_base.append(InterleavedShard(d, len(specs)))
+            elif all(sp.placements[md] == specs[0].placements[md] and not sp.placements[md].is_partial() and not (isinstance(sp.placements[md], (Shard, InterleavedShard)) and sp.placements[md].dim == d) for sp in specs):
+                il_base.append(specs[0].placements[md])
+            else:
+                ok_il = False
+                break
+    if ok_il and any(isinstance(p, InterleavedShard) and p.dim == d for p in il_base):
+        shape = list(specs[0].shape)
+        shape[d] = sum(sp.shape[d] for sp in specs)
+        return OutputSharding(out_spec(mesh, il_base, shape, specs[0].dtype), None)
+
+    # general case: all inputs align to the first's placements; cat dim must
+    # not be sharded
     base = []
     for p in specs[0].placements:
         if (isinstance(p, (Shard, InterleavedShard)) and p.dim == d) or isinstance(p, RaggedShard) or isinstance(p, Partial):
@@ -332,22 +357,221 @@ def unbind_rule(schema: OpSchema) -> OutputSharding:
     return OutputSharding([osp] * s.shape[d], None)
 
 
+def _handler_view(dispatcher, op, args, kwargs):
+    """view/reshape/_unsafe_view with LOCALIZED shape args (the reference's
+    _adjust_shape_and_stride_args, vescale/dtensor/_sharding_prop.py:228):
+    the target shape the user wrote is GLOBAL; the local op needs each
+    sharded output dim divided by its shard count."""
+    from ..dtensor import DTensor
+
+    x = args[0]
+    if not isinstance(x, DTensor):
+        return op(*args, **kwargs)
+    spec = x._spec
+    mesh = spec.mesh
+    target = list(args[1])
+    numel = 1
+    for s in spec.shape:
+        numel *= s
+    known, neg = 1, -1
+    for i, t in enumerate(target):
+        if t == -1:
+            neg = i
+        else:
+            known *= t
+    if neg >= 0:
+        target[neg] = numel // max(1, known)
+
+    mapping = _view_dim_map(tuple(spec.shape), tuple(target))
+
+    out_placements: List[Placement] = []
+    bad_dims = []
+    for md, p in enumerate(spec.placements):
+        if isinstance(p, InterleavedShard):
+            nd = mapping.get(p.dim)
+            if nd is None or target[nd] % (p.interleaved_size * mesh.size(md)) != 0 or spec.shape[p.dim] != target[nd]:
+                bad_dims.append(md)
+                out_placements.append(Replicate())
+            else:
+                out_placements.append(InterleavedShard(nd, p.interleaved_size))
+        elif isinstance(p, Shard):
+            nd = mapping.get(p.dim)
+            g_in = spec.shape[p.dim]
+            l_in = x._local_tensor.shape[p.dim]
+            # local out size = target[nd] * l_in / g_in must be integral
+            if nd is None or g_in == 0 or (target[nd] * l_in) % g_in != 0:
+                bad_dims.append(md)
+                out_placements.append(Replicate())
+            else:
+                out_placements.append(Shard(nd))
+        elif isinstance(p, RaggedShard):
+            bad_dims.append(md)
+            out_placements.append(Replicate())
+        else:
+            out_placements.append(p)
+
+    if bad_dims:
+        placements = [
+            Replicate() if md in bad_dims else p
+            for md, p in enumerate(spec.placements)
+        ]
+        x = x.redistribute(placements=placements)
+        spec = x._spec
+
+    # localize target shape via the input's actual local sizes
+    inv_mapping = {}
+    for md, p in enumerate(spec.placements):
+        if isinstance(p, (Shard, InterleavedShard)):
+            nd = mapping.get(p.dim)
+            if nd is not None:
+                inv_mapping[nd] = p.dim
+    local_target = list(target)
+    for md, p in enumerate(out_placements):
+        if isinstance(p, (Shard, InterleavedShard)):
+            d_in = inv_mapping.get(p.dim)
+            if d_in is None:
+                continue
+            g_in = spec.shape[d_in]
+            l_in = x._local_tensor.shape[d_in]
+            local_target[p.dim] = target[p.dim] * l_in // g_in
+    local = op(x._local_tensor, local_target, *args[2:], **kwargs)
+    osp = out_spec(mesh, out_placements, target, spec.dtype)
+    return DTensor(local, osp, requires_grad=local.requires_grad)
+
+
+def _handler_expand(dispatcher, op, args, kwargs):
+    from ..dtensor import DTensor
+
+    x = args[0]
+    if not isinstance(x, DTensor):
+        return op(*args, **kwargs)
+    spec = x._spec
+    mesh = spec.mesh
+    target = list(args[1])
+    offset = len(target) - spec.ndim
+    for i, t in enumerate(target):
+        if t == -1:
+            target[i] = spec.shape[i - offset]
+    out_placements = []
+    bad = []
+    for md, p in enumerate(spec.placements):
+        if isinstance(p, Shard):
+            out_placements.append(Shard(p.dim + offset))
+        elif isinstance(p, (InterleavedShard, RaggedShard)):
+            bad.append(md)
+            out_placements.append(Replicate())
+        else:
+            out_placements.append(p)
+    if bad:
+        x = x.redistribute(placements=[
+            Replicate() if md in bad else p for md, p in enumerate(spec.placements)
+        ])
+    local_target = list(target)
+    for md, p in enumerate(out_placements):
+        if isinstance(p, Shard):
+            local_target[p.dim] = x._local_tensor.shape[p.dim - offset]
+    local = op(x._local_tensor, local_target, *args[2:], **kwargs)
+    osp = out_spec(mesh, out_placements, target, spec.dtype)
+    return DTensor(local, osp, requires_grad=local.requires_grad)
+
+
+def _handler_split(dispatcher, op, args, kwargs):
+    """split on a DTensor, InterleavedShard-aware (the packed-QKV TP path:
+    splitting an IS(d, y) tensor into its y sections needs the LOCAL split
+    size rewritten to local_dim/y and yields Shard(d) sections)."""
+    from ..dtensor import DTensor
+    from ..redistribute import redistribute_local_tensor
+
+    x = args[0]
+    if not isinstance(x, DTensor):
+        return op(*args, **kwargs)
+    spec = x._spec
+    mesh = spec.mesh
+    split = args[1]
+    d = args[2] if len(args) > 2 else kwargs.get("dim", 0)
+    d = _norm_dim(d, spec.ndim)
+    sizes = (
+        [split] * ((spec.shape[d] + split - 1) // split)
+        if isinstance(split, int)
+        else list(split)
+    )
+    if isinstance(split, int):
+        sizes = [min(split, spec.shape[d] - i * split) for i in range(len(sizes))]
+    n_sec = len(sizes)
+
+    # find an IS placement on dim d with matching section structure
+    is_md = None
+    for md, p in enumerate(spec.placements):
+        if isinstance(p, InterleavedShard) and p.dim == d and p.interleaved_size == n_sec and all(s == sizes[0] for s in sizes):
+            is_md = md
+            break
+    local = x._local_tensor
+    if is_md is not None:
+        # other mesh dims must not shard d
+        bad = [
+            md for md, p in enumerate(spec.placements)
+            if md != is_md and ((isinstance(p, (Shard, InterleavedShard)) and p.dim == d) or isinstance(p, RaggedShard))
+        ]
+        if not bad:
+            local_sz = local.shape[d] // n_sec
+            chunks = torch.split(local, local_sz, dim=d)
+            outs = []
+            for i, c in enumerate(chunks):
+                pl = tuple(
+                    Shard(d) if md == is_md else p
+                    for md, p in enumerate(spec.placements)
+                )
+                shape = list(spec.shape)
+                shape[d] = sizes[i]
+                outs.append(
+                    DTensor(
+                        c,
+                        out_spec(mesh, pl, shape, spec.dtype),
+                        requires_grad=c.requires_grad,
+                    )
+                )
+            return outs
+
+    # general path: dim d must be unsharded; redistribute if needed
+    needs = [
+        md for md, p in enumerate(spec.placements)
+        if (isinstance(p, (Shard, InterleavedShard)) and p.dim == d) or isinstance(p, RaggedShard)
+    ]
+    placements = list(spec.placements)
+    if needs:
+        for md in needs:
+            placements[md] = Replicate()
+        x = x.redistribute(placements=placements)
+        spec = x._spec
+        local = x._local_tensor
+    chunks = torch.split(local, split if isinstance(split, int) else sizes, dim=d)
+    outs = []
+    for i, c in enumerate(chunks):
+        shape = list(spec.shape)
+        shape[d] = sizes[i]
+        outs.append(
+            DTensor(c, out_spec(mesh, spec.placements, shape, spec.dtype),
+                    requires_grad=c.requires_grad)
+        )
+    return outs
+
+
 def register(dispatcher):
     for op in (aten.detach, aten.alias, aten.clone, aten.contiguous, aten._unsafe_view_copy if hasattr(aten, "_unsafe_view_copy") else aten.alias):
         dispatcher.register_rule(op, same_as_input_rule)
     dispatcher.register_rule(aten.t.default, transpose_rule)
     dispatcher.register_rule(aten.transpose.int, transpose_rule)
     dispatcher.register_rule(aten.permute.default, permute_rule)
-    dispatcher.register_rule(aten.view.default, view_rule)
-    dispatcher.register_rule(aten._unsafe_view.default, view_rule)
-    dispatcher.register_rule(aten.reshape.default, view_rule)
-    dispatcher.register_rule(aten.expand.default, expand_rule)
+    dispatcher.register_handler(aten.view.default, _handler_view)
+    dispatcher.register_handler(aten._unsafe_view.default, _handler_view)
+    dispatcher.register_handler(aten.reshape.default, _handler_view)
+    dispatcher.register_handler(aten.expand.default, _handler_expand)
     dispatcher.register_rule(aten.unsqueeze.default, unsqueeze_rule)
     dispatcher.register_rule(aten.squeeze, squeeze_rule)
     dispatcher.register_rule(aten.slice.Tensor, slice_rule)
     dispatcher.register_rule(aten.select.int, select_rule)
     dispatcher.register_rule(aten.cat.default, cat_rule)
     dispatcher.register_rule(aten.stack.default, stack_rule)
-    dispatcher.register_rule(aten.split.Tensor, split_rule)
-    dispatcher.register_rule(aten.split_with_sizes.default, split_rule)
+    dispatcher.register_handler(aten.split.Tensor, _handler_split)
+    dispatcher.register_handler(aten.split_with_sizes.default, _handler_split)
     dispatcher.register_rule(aten.unbind.int, unbind_rule)
