@@ -1,0 +1,123 @@
+"""DistributedDataParallel — flat-buffer DP gradient engine.
+
+Parity: legacy/vescale/ddp/distributed_data_parallel.py:20-300 — grads
+accumulate into per-dtype flat GradBuffer views (`param.main_grad`);
+bucket-complete triggers an async all_reduce (or reduce_scatter when a
+DistributedOptimizer owns the shards); TP/SP Partial grads are allreduced
+on their own mesh dim first.  Composes with DModule (DTensor params) and
+plain modules.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Union
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..dtensor import DeviceMesh, DTensor, Partial
+from .grad_buffer import GradBuffer
+
+
+class DistributedDataParallel(nn.Module):
+    def __init__(
+        self,
+        module: nn.Module,
+        data_pg_or_device_mesh: Union[dist.ProcessGroup, DeviceMesh, None],
+        *,
+        accumulate_allreduce_grads_in_fp32: bool = True,
+        overlap_grad_reduce: bool = True,
+        use_distributed_optimizer: bool = False,
+        bucket_size: int = 40_000_000,
+        disable_bucketing: bool = False,
+    ):
+        super().__init__()
+        self.module = module
+        if isinstance(data_pg_or_device_mesh, DeviceMesh):
+            self.dp_group = data_pg_or_device_mesh.get_group(0)
+        else:
+            self.dp_group = data_pg_or_device_mesh
+        self.dp_world = (
+            dist.get_world_size(self.dp_group)
+            if self.dp_group is not None and dist.is_initialized()
+            else 1
+        )
+        self.overlap_grad_reduce = overlap_grad_reduce
+        self.use_distributed_optimizer = use_distributed_optimizer
+        self.accumulate_allreduce_grads_in_fp32 = accumulate_allreduce_grads_in_fp32
+        if disable_bucketing:
+            bucket_size = 2**62
+
+        params = [p for p in module.parameters() if p.requires_grad]
+        device = _param_device(params[0]) if params else torch.device("cpu")
+
+        by_dtype: Dict[torch.dtype, List[nn.Parameter]] = {}
+        for p in params:
+            dt = torch.float32 if accumulate_allreduce_grads_in_fp32 else _param_dtype(p)
+            by_dtype.setdefault(dt, []).append(p)
+
+        self.grad_buffers: Dict[torch.dtype, GradBuffer] = {}
+        for dt, ps in by_dtype.items():
+            self.grad_buffers[dt] = GradBuffer(
+                dt, ps, self.dp_group, bucket_size, use_distributed_optimizer,
+                overlap_grad_reduce, device,
+            )
+
+        # attach main_grad views + hooks
+        self._grad_buffer_of: Dict[nn.Parameter, GradBuffer] = {}
+        for dt, gb in self.grad_buffers.items():
+            for p in gb.param_index_map:
+                self._grad_buffer_of[p] = gb
+        for p in params:
+            p.main_grad = self._grad_buffer_of[p].get_main_grad_view(p)
+            p.register_post_accumulate_grad_hook(self._make_param_hook(p))
+
+    # ------------------------------------------------------------------
+    def _make_param_hook(self, param):
+        def hook(p):
+            g = p.grad
+            if g is None:
+                return
+            local = g._local_tensor if isinstance(g, DTensor) else g
+            # TP/SP partial grads: reduce on their own mesh dim(s) first
+            # (reference grad_buffer.py:97 all_reduce_partial_grad)
+            if isinstance(g, DTensor):
+                for md, pl in enumerate(g._spec.placements):
+                    if isinstance(pl, Partial):
+                        dist.all_reduce(local, group=g._spec.mesh.get_group(md))
+            p.main_grad.add_(local)
+            p.grad = None
+            self._grad_buffer_of[p].register_grad_ready(p)
+
+        return hook
+
+    # ------------------------------------------------------------------
+    def forward(self, *args, **kwargs):
+        return self.module(*args, **kwargs)
+
+    def zero_grad_buffer(self):
+        for gb in self.grad_buffers.values():
+            gb.reset()
+        for p in self.module.parameters():
+            p.grad = None
+
+    def finish_grad_sync(self):
+        for gb in self.grad_buffers.values():
+            gb.finish_grad_sync()
+
+    def state_dict(self, *args, **kwargs):
+        return self.module.state_dict(*args, **kwargs)
+
+    def load_state_dict(self, *args, **kwargs):
+        return self.module.load_state_dict(*args, **kwargs)
+
+
+def _param_device(p):
+    d = p.data
+    if hasattr(d, "_local_tensor"):
+        return d._local_tensor.device
+    return d.device
+
+
+def _param_dtype(p):
+    return p.data.dtype
