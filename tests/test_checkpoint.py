@@ -1,0 +1,147 @@
+"""Checkpoint tests: DCP-format save/load with resharding across layouts
+(mirrors legacy/test/checkpoint/*)."""
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.nn as nn
+
+from tests.common import spawn
+
+from vescale_amd.dtensor import (
+    DTensor,
+    RaggedShard,
+    Replicate,
+    Shard,
+    distribute_tensor,
+    init_device_mesh,
+)
+
+
+def test_break_ragged_box():
+    from vescale_amd.checkpoint import break_ragged_box
+
+    # full tensor
+    assert break_ragged_box((4, 6), 0, 24) == [((0, 0), (4, 6))]
+    # middle range spanning partial rows
+    boxes = break_ragged_box((4, 6), 3, 20)
+    total = sum(b[1][0] * b[1][1] for b in boxes)
+    assert total == 17
+    # verify coverage matches flat range exactly
+    covered = set()
+    for (ro, co), (rs, cs) in boxes:
+        for r in range(ro, ro + rs):
+            for c in range(co, co + cs):
+                covered.add(r * 6 + c)
+    assert covered == set(range(3, 20))
+    # 3-D
+    boxes = break_ragged_box((3, 4, 5), 7, 53)
+    covered = set()
+    for (o0, o1, o2), (s0, s1, s2) in boxes:
+        for a in range(o0, o0 + s0):
+            for b in range(o1, o1 + s1):
+                for c in range(o2, o2 + s2):
+                    covered.add(a * 20 + b * 5 + c)
+    assert covered == set(range(7, 53))
+
+
+class Net(nn.Module):
+    def __init__(self, d=8):
+        super().__init__()
+        self.fc1 = nn.Linear(d, 4 * d, bias=False)
+        self.fc2 = nn.Linear(4 * d, d, bias=False)
+
+
+def _t_save_sharded(rank, ws, path):
+    import vescale_amd.checkpoint as ckpt
+
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("TP",))
+    torch.manual_seed(11)
+    net = Net()
+    sd = {
+        "fc1.weight": distribute_tensor(net.fc1.weight.detach(), mesh, [Shard(0)]),
+        "fc2.weight": distribute_tensor(net.fc2.weight.detach(), mesh, [Shard(1)]),
+    }
+    ckpt.save(path, {"model": sd})
+
+
+def _t_load_resharded(rank, ws, path):
+    import vescale_amd.checkpoint as ckpt
+
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("TP",))
+    torch.manual_seed(11)
+    ref = Net()
+    # load into the OPPOSITE sharding layout
+    sd = {
+        "fc1.weight": distribute_tensor(torch.zeros(32, 8), mesh, [Shard(1)]),
+        "fc2.weight": distribute_tensor(torch.zeros(8, 32), mesh, [Shard(0)]),
+    }
+    ckpt.load(path, {"model": sd})
+    assert torch.allclose(sd["fc1.weight"].full_tensor(), ref.fc1.weight.detach())
+    assert torch.allclose(sd["fc2.weight"].full_tensor(), ref.fc2.weight.detach())
+
+
+def test_save_load_reshard_tp():
+    with tempfile.TemporaryDirectory() as td:
+        spawn(2, _t_save_sharded, td)
+        spawn(2, _t_load_resharded, td)
+
+
+def _t_save_ragged(rank, ws, path):
+    import vescale_amd.checkpoint as ckpt
+
+    mesh = init_device_mesh("cpu", (ws,))
+    g = torch.arange(48, dtype=torch.float32).reshape(6, 8)
+    d = distribute_tensor(g, mesh, [RaggedShard((0, 1), (19, 29))])
+    ckpt.save(path, {"model": {"w": d}})
+
+
+def _t_load_ragged_full(rank, ws, path):
+    import vescale_amd.checkpoint as ckpt
+
+    mesh = init_device_mesh("cpu", (ws,))
+    tgt = {"w": distribute_tensor(torch.zeros(6, 8), mesh, [Replicate()])}
+    ckpt.load(path, {"model": tgt})
+    expect = torch.arange(48, dtype=torch.float32).reshape(6, 8)
+    assert torch.allclose(tgt["w"].full_tensor(), expect)
+
+
+def test_ragged_save_then_replicate_load():
+    """Communication-free ragged save; load under a different layout."""
+    with tempfile.TemporaryDirectory() as td:
+        spawn(2, _t_save_ragged, td)
+        spawn(1, _t_load_ragged_full, td)
+
+
+def _t_fsdp_roundtrip(rank, ws, path, save_ws):
+    import vescale_amd.checkpoint as ckpt
+    from vescale_amd.fsdp import FSDP
+    from vescale_amd.models.llama import LlamaModel, llama_tiny
+
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("DP",))
+    torch.manual_seed(42)
+    cfg = llama_tiny()
+    model = LlamaModel(cfg)
+    model.init_weights()
+    eng = FSDP(model, mesh, param_dtype=torch.float32, device=torch.device("cpu"))
+    if ws == save_ws:
+        ckpt.save(path, {"model": eng})
+    else:
+        # perturb, then load: must restore the saved values
+        for u in eng.units:
+            u.shard.add_(1.0)
+        ckpt.load(path, {"model": eng})
+        torch.manual_seed(42)
+        ref = LlamaModel(cfg)
+        ref.init_weights()
+        sd = eng.sharded_state_dict()
+        for k, v in ref.named_parameters():
+            got = sd[k].full_tensor().reshape(v.shape)
+            assert torch.allclose(got, v.detach(), atol=1e-6), k
+
+
+def test_fsdp_checkpoint_reshard_ws2_to_ws1():
+    with tempfile.TemporaryDirectory() as td:
+        spawn(2, _t_fsdp_roundtrip, td, 2)
+        spawn(1, _t_fsdp_roundtrip, td, 2)
