@@ -1,0 +1,108 @@
+"""Pipeline-parallel tests (CPU/gloo ws=4): loss/grad parity vs single
+device (mirrors legacy/test/parallel/pipeline/e2e/test_pp_accuracy_alignment.py)
+for 1F1B, GPipe, interleaved, and zero-bubble schedules."""
+import json
+import os
+import tempfile
+
+import pytest
+import torch
+import torch.nn as nn
+
+from tests.common import spawn
+
+from vescale_amd.plan import (
+    PipelineParallelPlan,
+    PipelineScheduleType,
+    PipelineSplitMethodType,
+)
+
+
+def _make_modules(seed=17, n=8, d=16):
+    torch.manual_seed(seed)
+    return [nn.Sequential(nn.Linear(d, d), nn.Tanh()) for _ in range(n)]
+
+
+def _loss_fn(out, tgt):
+    return (out - tgt).pow(2).mean()
+
+
+def _single_device_ref(n_mb=4, bs=8, d=16):
+    mods = _make_modules()
+    model = nn.Sequential(*mods)
+    torch.manual_seed(23)
+    x = torch.randn(bs, d)
+    y = torch.randn(bs, d)
+    total = 0.0
+    for xm, ym in zip(torch.chunk(x, n_mb), torch.chunk(y, n_mb)):
+        loss = _loss_fn(model(xm), ym) / n_mb
+        loss.backward()
+        total += float(loss) * n_mb
+    grads = {k: p.grad.clone() for k, p in model.named_parameters()}
+    return total, grads
+
+
+def _t_pp(rank, ws, sched, virtual_chunks, out_path):
+    from vescale_amd.engine import PipeEngine
+    from vescale_amd.pipe.pipe_stage import construct_pipeline_stage
+
+    n_mb, bs, d = 4, 8, 16
+    mods = _make_modules()
+    plan = PipelineParallelPlan(
+        num_stages=ws,
+        virtual_chunks=virtual_chunks,
+        schedule_type=PipelineScheduleType(sched),
+        split_method=PipelineSplitMethodType.UNIFORM,
+    )
+    stage = construct_pipeline_stage(mods, plan, rank)
+    engine = PipeEngine(stage, plan, loss_fn=_loss_fn, device=torch.device("cpu"))
+    torch.manual_seed(23)
+    x = torch.randn(bs, d)
+    y = torch.randn(bs, d)
+    loss = engine.forward_backward((x, y), n_mb)
+    ref_loss, ref_grads = _single_device_ref(n_mb, bs, d)
+    if rank == ws - 1:
+        assert loss is not None
+        assert abs(float(loss) - ref_loss) < 1e-6, (float(loss), ref_loss)
+    # grad parity for this stage's params: map chunk modules back to the
+    # global module list by identity of shapes + values is fragile; instead
+    # compare against a fresh single-device backward on the same module
+    # objects (they ARE the same objects, split in place)
+    for name, p in stage.named_parameters():
+        assert p.grad is not None, name
+    # compare numerically: rebuild reference with same seed, walk in order
+    ref_mods = _make_modules()
+    ref_model = nn.Sequential(*ref_mods)
+    torch.manual_seed(23)
+    xr = torch.randn(bs, d)
+    yr = torch.randn(bs, d)
+    for xm, ym in zip(torch.chunk(xr, n_mb), torch.chunk(yr, n_mb)):
+        (_loss_fn(ref_model(xm), ym) / n_mb).backward()
+    # figure out which global modules this stage holds: UNIFORM split of 8
+    # equal modules over (ws*virtual_chunks) parts
+    n_parts = ws * virtual_chunks
+    per = 8 // n_parts
+    for ck in range(virtual_chunks):
+        part_idx = ck * ws + rank
+        gmods = ref_mods[part_idx * per : (part_idx + 1) * per]
+        stage_chunk = stage.chunks[ck]
+        sp = list(stage_chunk.parameters())
+        rp = [p for m in gmods for p in m.parameters()]
+        assert len(sp) == len(rp)
+        for a, b in zip(sp, rp):
+            assert torch.allclose(a.grad, b.grad, atol=1e-6), (
+                sched, ck, (a.grad - b.grad).abs().max(),
+            )
+
+
+@pytest.mark.parametrize("sched,vc", [("1f1b", 1), ("gpipe", 1), ("interleaved_1f1b", 2), ("zero_bubble_v", 1)])
+def test_pp_accuracy_alignment(sched, vc):
+    spawn(4, _t_pp, sched, vc, None)
+
+
+def test_uniform_split_balance():
+    from vescale_amd.pipe.pipe_stage import uniform_split
+
+    mods = _make_modules(n=8)
+    parts = uniform_split(mods, 4)
+    assert [len(p) for p in parts] == [2, 2, 2, 2]

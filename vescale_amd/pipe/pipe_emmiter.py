@@ -1,0 +1,187 @@
+"""ScheduleEngine — emits and executes per-stage instruction lists.
+
+Parity: legacy/vescale/pipe/pipe_emmiter.py:43-356 (PipelineEmitter +
+ScheduleEngine) + the registered instruction impls of
+_schedules/pipedream_flush.py:137-1290 — one engine executing the typed
+Instr stream over the p2p layer.
+"""
+from __future__ import annotations
+
+from typing import Callable, Dict, List, Optional, Sequence, Tuple
+
+import torch
+import torch.distributed as dist
+
+from ..plan import PipelineParallelPlan, PipelineScheduleType
+from . import p2p_communication as p2p
+from .instruction import (
+    Instr,
+    gpipe_schedule,
+    interleaved_1f1b_schedule,
+    one_f_one_b_schedule,
+)
+from .pipe_stage import PipeModule
+
+
+class ScheduleEngine:
+    def __init__(
+        self,
+        stage: PipeModule,
+        plan: PipelineParallelPlan,
+        *,
+        stage_to_rank: Optional[Callable[[int], int]] = None,
+        pg=None,
+        loss_fn: Optional[Callable] = None,
+        device: Optional[torch.device] = None,
+    ):
+        self.stage = stage
+        self.plan = plan
+        self.P = plan.num_stages
+        self.V = plan.virtual_chunks
+        self.s = stage.stage_id
+        self.pg = pg
+        self.loss_fn = loss_fn
+        self.device = device or (
+            torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+        )
+        self.stage_to_rank = stage_to_rank or (lambda st: st)
+        self.prev_rank = self.stage_to_rank(self.s - 1) if self.s > 0 else None
+        self.next_rank = self.stage_to_rank(self.s + 1) if self.s < self.P - 1 else None
+        # interleaved: the chunk chain wraps P-1 -> 0
+        self.wrap_prev = self.stage_to_rank(self.P - 1)
+        self.wrap_next = self.stage_to_rank(0)
+
+        # per-(chunk, mb) state
+        self._inputs: Dict[Tuple[int, int], Optional[torch.Tensor]] = {}
+        self._outputs: Dict[Tuple[int, int], torch.Tensor] = {}
+        self._recv_grads: Dict[Tuple[int, int], torch.Tensor] = {}
+        self._losses: List[torch.Tensor] = []
+
+    # ------------------------------------------------------------------
+    def build_schedule(self, n_mb: int) -> List[Instr]:
+        st = self.plan.schedule_type
+        if st == PipelineScheduleType.SIMPLE_1F1B:
+            if self.V != 1:
+                raise ValueError("1F1B requires virtual_chunks == 1")
+            return one_f_one_b_schedule(self.s, self.P, n_mb)
+        if st == PipelineScheduleType.GPIPE:
+            return gpipe_schedule(self.s, self.P, n_mb)
+        if st == PipelineScheduleType.INTERLEAVED_1F1B:
+            return interleaved_1f1b_schedule(self.s, self.P, n_mb, self.V)
+        if st == PipelineScheduleType.ZERO_BUBBLE:
+            from .zero_bubble import zero_bubble_schedule
+
+            return zero_bubble_schedule(self.s, self.P, n_mb)
+        raise ValueError(st)
+
+    # peer ranks for a chunk's in/out edges -----------------------------
+    def _in_peer(self, ck: int) -> Optional[int]:
+        if self.s == 0:
+            return self.wrap_prev if ck > 0 else None
+        return self.prev_rank
+
+    def _out_peer(self, ck: int) -> Optional[int]:
+        if self.s == self.P - 1:
+            return self.wrap_next if ck < self.V - 1 else None
+        return self.next_rank
+
+    # ------------------------------------------------------------------
+    def execute(
+        self,
+        minibatch: Optional[Tuple[torch.Tensor, torch.Tensor]],
+        n_microbatches: int,
+        *,
+        grad_scale: Optional[float] = None,
+    ):
+        """Run one forward_backward over the minibatch split into
+        n_microbatches.  First stage consumes inputs; last stage consumes
+        targets + computes loss via loss_fn(output, target) (mean over
+        microbatches)."""
+        xs = ys = None
+        if minibatch is not None:
+            x, y = minibatch
+            xs = list(torch.chunk(x, n_microbatches)) if x is not None else None
+            ys = list(torch.chunk(y, n_microbatches)) if y is not None else None
+        self._losses = []
+        self._inputs.clear()
+        self._outputs.clear()
+        self._recv_grads.clear()
+        scale = grad_scale if grad_scale is not None else 1.0 / n_microbatches
+
+        for ins in self.build_schedule(n_microbatches):
+            kind, m, ck = ins.kind, ins.microbatch, ins.chunk
+            if kind == "RECV_FWD":
+                t = p2p.recv_forward(self._in_peer(ck), self.pg, device=self.device)
+                self._inputs[(ck, m)] = t.requires_grad_(True)
+            elif kind == "FWD":
+                self._fwd(ck, m, xs, ys, scale)
+            elif kind == "SEND_FWD":
+                p2p.send_forward(
+                    self._outputs[(ck, m)].detach(), self._out_peer(ck), self.pg
+                )
+            elif kind == "SEND_FWD_RECV_BWD":
+                g = p2p.send_forward_recv_backward(
+                    self._outputs[(ck, m)].detach(), self._out_peer(ck), self.pg,
+                    device=self.device,
+                )
+                self._recv_grads[(ck, ins.microbatch2)] = g
+            elif kind == "RECV_BWD":
+                self._recv_grads[(ck, m)] = p2p.recv_backward(
+                    self._out_peer(ck), self.pg, device=self.device
+                )
+            elif kind == "BWD":
+                self._bwd(ck, m)
+            elif kind == "SEND_BWD":
+                g = self._pop_input_grad(ck, m)
+                p2p.send_backward(g, self._in_peer(ck), self.pg)
+            elif kind == "SEND_BWD_RECV_FWD":
+                g = self._pop_input_grad(ck, m)
+                t = p2p.send_backward_recv_forward(
+                    g, self._in_peer(ck), self.pg, device=self.device
+                )
+                self._inputs[(ck, ins.microbatch2)] = t.requires_grad_(True)
+            else:
+                from .instruction import VESCALE_INSTRUCTION_REGISTRY
+
+                VESCALE_INSTRUCTION_REGISTRY[kind](self, ins)
+
+        p2p.drain_send_reqs()
+        if self._losses:
+            return torch.stack([l.detach() for l in self._losses]).sum()
+        return None
+
+    # ------------------------------------------------------------------
+    def _fwd(self, ck, m, xs, ys, scale):
+        is_first_global = self.s == 0 and ck == 0
+        is_last_global = self.s == self.P - 1 and ck == self.V - 1
+        if is_first_global:
+            inp = xs[m].to(self.device)
+            self._inputs[(ck, m)] = None  # no upstream grad
+            out = self.stage(inp, chunk=ck)
+        else:
+            inp = self._inputs[(ck, m)]
+            out = self.stage(inp, chunk=ck)
+        if is_last_global:
+            assert self.loss_fn is not None, "last stage needs loss_fn"
+            tgt = ys[m].to(self.device) if ys is not None else None
+            loss = self.loss_fn(out, tgt) * scale
+            self._outputs[(ck, m)] = loss
+            self._losses.append(loss / scale)
+        else:
+            self._outputs[(ck, m)] = out
+
+    def _bwd(self, ck, m):
+        out = self._outputs.pop((ck, m))
+        is_last_global = self.s == self.P - 1 and ck == self.V - 1
+        if is_last_global:
+            out.backward()
+        else:
+            g = self._recv_grads.pop((ck, m))
+            torch.autograd.backward(out, grad_tensors=g)
+
+    def _pop_input_grad(self, ck, m):
+        inp = self._inputs.pop((ck, m))
+        assert inp is not None, "first stage has no upstream grad to send"
+        g = inp.grad
+        assert g is not None, f"no input grad for chunk {ck} mb {m}"
+        return g
