@@ -44,13 +44,20 @@ def install_forward_hooks(root: nn.Module, mesh: DeviceMesh, fwd_plan: Dict[str,
     input_plans: Dict[str, Any] = {}
     output_plans: Dict[str, Any] = {}
     weight_plans: Dict[str, Any] = {}
+    def strip_suffix(k: str, suffix: str):
+        """Remove '.suffix' or the regex-escaped '\\.suffix'."""
+        base = k[: -len("." + suffix) ]
+        if base.endswith("\\"):
+            base = base[:-1]
+        return base
+
     for k, v in fwd_plan.items():
         if k.endswith(".input"):
-            input_plans[k[: -len(".input")]] = v
+            input_plans[strip_suffix(k, "input")] = v
         elif k.endswith(".output"):
-            output_plans[k[: -len(".output")]] = v
+            output_plans[strip_suffix(k, "output")] = v
         elif k.endswith(".weight_placement"):
-            weight_plans[k[: -len(".weight_placement")]] = v
+            weight_plans[strip_suffix(k, "weight_placement")] = v
 
     def match(table, fqn):
         for pattern, v in table.items():

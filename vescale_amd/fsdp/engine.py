@@ -35,6 +35,7 @@ from ..dtensor.device_mesh import DeviceMesh
 from ..dtensor.dtensor import DTensor
 from ..dtensor.placement_types import RaggedShard, TensorMeta
 from ..dtensor._dtensor_spec import DTensorSpec
+from ..ndtimeline import ndtimeit, predefined as ndm
 
 logger = logging.getLogger(__name__)
 
@@ -265,6 +266,7 @@ class FSDP(nn.Module):
     def _unshard(self, unit: FSDPUnit, async_on_stream: bool = True):
         if unit._is_unsharded or self.world_size == 1:
             return
+        _t = ndtimeit(ndm.UNSHARD_AG); _t.__enter__()
         unit.full = torch.empty(
             unit.flat_numel, dtype=unit.param_dtype, device=self.device
         )
@@ -280,6 +282,7 @@ class FSDP(nn.Module):
             unit._ag_event = None
         unit._attach_param_views()
         unit._is_unsharded = True
+        _t.__exit__(None, None, None)
 
     def _wait_unshard(self, unit: FSDPUnit):
         if unit._ag_event is not None:
@@ -360,6 +363,7 @@ class FSDP(nn.Module):
 
     def _finish_unit_grads(self, unit: FSDPUnit):
         # all grads of this unit accumulated into unit.grad_full
+        _t = ndtimeit(ndm.GRAD_RS); _t.__enter__()
         for p in unit.params:
             p.grad = None
         if self.world_size == 1:
@@ -388,6 +392,7 @@ class FSDP(nn.Module):
         unit.grad_full = None
         # reshard params after backward
         self._reshard(unit)
+        _t.__exit__(None, None, None)
 
     def finish_grad_sync(self):
         """Wait for all grad reduce-scatters (call after loss.backward())."""

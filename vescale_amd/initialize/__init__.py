@@ -1,0 +1,15 @@
+from .deferred_init import (
+    deferred_init,
+    is_deferred,
+    materialize_dparameter,
+    materialize_dtensor,
+    materialize_module,
+)
+
+__all__ = [
+    "deferred_init",
+    "is_deferred",
+    "materialize_dtensor",
+    "materialize_dparameter",
+    "materialize_module",
+]

@@ -1,0 +1,25 @@
+from .api import flush, init_ndtimers, wait
+from .timer import (
+    DeviceTimer,
+    GlobalReferenceTime,
+    NDMetricLevel,
+    NDTimerManager,
+    ndtimeit,
+    ndtimeit_p2p,
+    ndtimer,
+)
+from . import predefined
+
+__all__ = [
+    "init_ndtimers",
+    "flush",
+    "wait",
+    "DeviceTimer",
+    "NDTimerManager",
+    "GlobalReferenceTime",
+    "NDMetricLevel",
+    "ndtimer",
+    "ndtimeit",
+    "ndtimeit_p2p",
+    "predefined",
+]

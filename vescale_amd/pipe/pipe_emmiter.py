@@ -12,6 +12,7 @@ from typing import Callable, Dict, List, Optional, Sequence, Tuple
 import torch
 import torch.distributed as dist
 
+from ..ndtimeline import ndtimeit, predefined as ndm
 from ..plan import PipelineParallelPlan, PipelineScheduleType
 from . import p2p_communication as p2p
 from .instruction import (
@@ -111,10 +112,12 @@ class ScheduleEngine:
         for ins in self.build_schedule(n_microbatches):
             kind, m, ck = ins.kind, ins.microbatch, ins.chunk
             if kind == "RECV_FWD":
-                t = p2p.recv_forward(self._in_peer(ck), self.pg, device=self.device)
+                with ndtimeit(ndm.RECV_FORWARD):
+                    t = p2p.recv_forward(self._in_peer(ck), self.pg, device=self.device)
                 self._inputs[(ck, m)] = t.requires_grad_(True)
             elif kind == "FWD":
-                self._fwd(ck, m, xs, ys, scale)
+                with ndtimeit(ndm.FORWARD_COMPUTE):
+                    self._fwd(ck, m, xs, ys, scale)
             elif kind == "SEND_FWD":
                 p2p.send_forward(
                     self._outputs[(ck, m)].detach(), self._out_peer(ck), self.pg
@@ -130,7 +133,8 @@ class ScheduleEngine:
                     self._out_peer(ck), self.pg, device=self.device
                 )
             elif kind == "BWD":
-                self._bwd(ck, m)
+                with ndtimeit(ndm.BACKWARD_COMPUTE):
+                    self._bwd(ck, m)
             elif kind == "SEND_BWD":
                 g = self._pop_input_grad(ck, m)
                 p2p.send_backward(g, self._in_peer(ck), self.pg)
