@@ -106,3 +106,22 @@ def test_uniform_split_balance():
     mods = _make_modules(n=8)
     parts = uniform_split(mods, 4)
     assert [len(p) for p in parts] == [2, 2, 2, 2]
+
+
+def test_fx_pipe_parser():
+    """GRAPH_EAGER: fx-trace an MLP stack and split into stage graphs."""
+    import torch.nn as nn
+    from vescale_amd.pipe.pipe_parser import construct_pipeline_split_graph
+
+    torch.manual_seed(0)
+    model = nn.Sequential(*[nn.Linear(8, 8) for _ in range(8)])
+    stages = construct_pipeline_split_graph(model, ["2", "4", "6"], leaf_classes=("Linear",))
+    assert len(stages) == 4
+    x = torch.randn(3, 8)
+    ref = model(x)
+    h = x
+    for st in stages:
+        h = st(h)
+        if isinstance(h, tuple):
+            h = h[0]
+    assert torch.allclose(h, ref, atol=1e-6)
