@@ -1,0 +1,66 @@
+"""Mixtral expert-parallel training example (parity role:
+legacy/examples/mixtral_EP_training/).
+
+Run: python -m torch.distributed.run --nnodes=1 --nproc-per-node 4 \
+  --master-addr 127.0.0.1 examples/mixtral_ep_train.py
+"""
+import argparse
+import os
+
+import sys
+
+sys.path.insert(0, __file__.rsplit("/examples/", 1)[0])
+
+import torch
+import torch.distributed as dist
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--batch", type=int, default=4)
+    ap.add_argument("--seq", type=int, default=512)
+    args = ap.parse_args()
+
+    on_gpu = torch.cuda.is_available()
+    dist.init_process_group("nccl" if on_gpu else "gloo")
+    rank = dist.get_rank()
+    world = dist.get_world_size()
+    device = torch.device("cuda") if on_gpu else torch.device("cpu")
+
+    from vescale_amd.dtensor import init_device_mesh
+    from vescale_amd.models.mixtral import MixtralModel, mixtral_8x7b, mixtral_tiny
+    from vescale_amd.moe import parallelize_experts
+
+    mesh = init_device_mesh(device.type, (world,), mesh_dim_names=("EP",))
+    torch.manual_seed(0)
+    cfg = mixtral_8x7b() if on_gpu else mixtral_tiny()
+    if not on_gpu:
+        args.seq = 32
+    model = MixtralModel(cfg).to(device)
+    model.init_weights()
+    parallelize_experts(model, mesh)
+    # non-expert params are replicated -> plain DDP-style allreduce of their
+    # grads; expert grads already carry all EP ranks' token contributions
+    opt = torch.optim.AdamW([p for p in model.parameters()], lr=3e-4)
+
+    gen = torch.Generator().manual_seed(rank)
+    for step in range(args.steps):
+        x = torch.randint(0, cfg.vocab_size, (args.batch, args.seq), generator=gen).to(device)
+        y = torch.roll(x, -1, dims=1)
+        loss = model(x, y)
+        loss.backward()
+        # sync non-expert grads across EP (they act as DP for dense params)
+        for n, p in model.named_parameters():
+            if p.grad is not None and ".experts." not in n:
+                dist.all_reduce(p.grad, group=mesh.get_group(0))
+                p.grad /= world
+        opt.step()
+        opt.zero_grad()
+        if rank == 0 and step % 5 == 0:
+            print(f"step {step} loss {float(loss):.4f}")
+    dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -11,6 +11,7 @@
 #include "adamw.hip"
 #include "cross_entropy.hip"
 #include "philox_random.hip"
+#include "gemm.hip"
 
 #include <vector>
 
@@ -259,6 +260,29 @@ void scale_(at::Tensor x, c10::optional<at::Tensor> scale_t, double scale_c) {
                      xf, sp, (float)scale_c, n);
 }
 
+// ------------------------------ MFMA GEMM ------------------------------
+at::Tensor gemm_tn(at::Tensor a, at::Tensor b, int64_t variant) {
+  // C[M,N] = a[M,K] @ b[N,K]^T
+  check_bf16_contig(a, "a");
+  check_bf16_contig(b, "b");
+  int64_t M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(b.size(1) == K, "K mismatch");
+  int BM = variant == 1 ? 128 : 256;
+  int BK = variant == 1 ? 64 : 32;
+  TORCH_CHECK(M % BM == 0 && N % 256 == 0 && K % BK == 0,
+              "gemm_tn tile divisibility violated");
+  auto c = at::empty({M, N}, a.options());
+  int grid = (int)((M / BM) * (N / 256));
+  auto kern = gemm_tn_bf16_v0;
+  if (variant == 1) kern = gemm_tn_bf16_v1;
+  if (variant == 2) kern = gemm_tn_bf16_v2;
+  hipLaunchKernelGGL(kern, dim3(grid), dim3(512), 0, cur_stream(),
+                     (const unsigned short*)a.data_ptr(),
+                     (const unsigned short*)b.data_ptr(),
+                     (unsigned short*)c.data_ptr(), (int)M, (int)N, (int)K);
+  return c;
+}
+
 // ------------------------------ philox random ------------------------------
 ShardDesc make_desc(const std::vector<int64_t>& gshape,
                     const std::vector<int64_t>& lshape,
@@ -359,6 +383,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step", &adamw_step);
   m.def("l2norm_sq", &l2norm_sq);
   m.def("scale_", &scale_);
+  m.def("gemm_tn", &gemm_tn, pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("variant") = 0);
   m.def("philox_uniform_", &philox_uniform_);
   m.def("philox_normal_", &philox_normal_);
   m.def("philox_dropout", &philox_dropout);
