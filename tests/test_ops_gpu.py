@@ -234,3 +234,23 @@ def test_philox_dropout():
     assert torch.allclose(kept.float(), torch.full_like(kept.float(), 1.0 / 0.9), atol=1e-2)
     dropped = out[~mask.bool()]
     assert (dropped == 0).all()
+
+
+@requires_gpu
+def test_gemm_tn_mfma():
+    """Hand-written MFMA GEMM numerics vs fp32 reference (all variants)."""
+    C = _ext()
+    torch.manual_seed(30)
+    a = torch.randn(512, 1024, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(512, 1024, device="cuda", dtype=torch.bfloat16)
+    ref = a.float() @ b.float().t()
+    for v in (0, 2, 3):
+        got = C.gemm_tn(a, b, v).float()
+        assert torch.allclose(got, ref, atol=2.0, rtol=2e-2), f"variant {v}"
+    # identity with asymmetric B (transpose-detecting, guide G9)
+    eye = torch.zeros(256, 256, device="cuda", dtype=torch.bfloat16)
+    eye[range(256), range(256)] = 1.0
+    bb = (torch.arange(256, device="cuda").view(-1, 1) * 0.1
+          + torch.arange(256, device="cuda").view(1, -1) * 0.001).bfloat16()
+    got = C.gemm_tn(eye, bb, 0).float()
+    assert torch.allclose(got, bb.float().t(), atol=0.05)

@@ -267,16 +267,20 @@ at::Tensor gemm_tn(at::Tensor a, at::Tensor b, int64_t variant) {
   check_bf16_contig(b, "b");
   int64_t M = a.size(0), K = a.size(1), N = b.size(0);
   TORCH_CHECK(b.size(1) == K, "K mismatch");
-  int BM = variant == 1 ? 128 : 256;
+  int BM = variant == 1 ? 128 : (variant == 3 ? 128 : (variant == 4 ? 512 : 256));
+  int BN = variant == 3 ? 128 : 256;
   int BK = variant == 1 ? 64 : 32;
-  TORCH_CHECK(M % BM == 0 && N % 256 == 0 && K % BK == 0,
+  int threads = variant == 3 ? 256 : (variant == 4 ? 1024 : 512);
+  TORCH_CHECK(M % BM == 0 && N % BN == 0 && K % BK == 0,
               "gemm_tn tile divisibility violated");
   auto c = at::empty({M, N}, a.options());
-  int grid = (int)((M / BM) * (N / 256));
+  int grid = (int)((M / BM) * (N / BN));
   auto kern = gemm_tn_bf16_v0;
   if (variant == 1) kern = gemm_tn_bf16_v1;
   if (variant == 2) kern = gemm_tn_bf16_v2;
-  hipLaunchKernelGGL(kern, dim3(grid), dim3(512), 0, cur_stream(),
+  if (variant == 3) kern = gemm_tn_bf16_v3;
+  if (variant == 4) kern = gemm_tn_bf16_v4;
+  hipLaunchKernelGGL(kern, dim3(grid), dim3(threads), 0, cur_stream(),
                      (const unsigned short*)a.data_ptr(),
                      (const unsigned short*)b.data_ptr(),
                      (unsigned short*)c.data_ptr(), (int)M, (int)N, (int)K);

@@ -7,13 +7,16 @@
 // Design per /opt/skills/guides/cdna_hip_programming.md §5 / §5.5 (T1-T5):
 //   - 512 threads = 8 waves (2M x 4N); mfma_f32_16x16x32_bf16
 //   - NBUF-deep K-tile pipeline: staging issued BEFORE the phase's
-//     ds_read+MFMA (T3 recipe), COUNTED s_waitcnt vmcnt at the K-tile
-//     switch (T4 — never drain to 0 mid-loop)
+//     ds_read+MFMA (T3), COUNTED s_waitcnt vmcnt at the K-tile switch
+//     (T4 — never drain to 0 mid-loop)
 //   - __builtin_amdgcn_global_load_lds width 16, linear LDS dest +
 //     inverse-swizzled global source, swizzled ds_read (rule #21)
-//   - LDS XOR swizzle (T2): 16B-chunk bits XOR row bits
-//   - raw s_barrier (no implicit vmcnt(0) drain), s_setprio around MFMA (T5)
+//   - LDS XOR swizzle (T2); raw s_barrier; s_setprio around MFMA (T5)
 //   - XCD-aware bijective blockIdx swizzle (T1, ERRATA #11 formula)
+//   - ALL addressing hoisted out of the K-loop: LDS read/stage offsets are
+//     per-lane constants; global sources advance by BK per staged tile
+//     (the asm audit showed per-iter 64-bit MADs + bfe swizzle math
+//     competing with MFMA issue)
 #include "common.h"
 
 #define NXCD 8
@@ -26,36 +29,35 @@ DEV int swz_off(int byte_off) {
   if (!SWZ) return byte_off;
   int row = byte_off / ROWB;
   if (ROWB == 64) {
-    // 4 chunks/row: XOR chunk bits (4-5) with row bits 1-2
     return byte_off ^ (((row >> 1) & 3) << 4);
   }
-  // 128B rows, 8 chunks: XOR chunk bits (4-6) with row bits 0-2 (G4 form)
   return byte_off ^ ((row & 7) << 4);
 }
 
-template <int BM, int BN, int BK, int NBUF, bool SWZ>
+template <int BM, int BN, int BK, int NBUF, bool SWZ, int WAVES_M = 2, int WAVES_N = 4>
 __device__ __forceinline__ void
 gemm_tn_kernel(const unsigned short* __restrict__ A,
                const unsigned short* __restrict__ B,
                unsigned short* __restrict__ C, int M, int N, int K) {
-  constexpr int THREADS = 512;
-  constexpr int ROWB = BK * 2;               // bytes per LDS row
-  constexpr int TILE_A = BM * BK;            // elems
+  constexpr int THREADS = WAVES_M * WAVES_N * 64;
+  constexpr int ROWB = BK * 2;
+  constexpr int TILE_A = BM * BK;
   constexpr int TILE_B = BN * BK;
+  constexpr int TILE = TILE_A + TILE_B;
   constexpr int LOADS_A = TILE_A * 2 / (THREADS * 16);
   constexpr int LOADS_B = TILE_B * 2 / (THREADS * 16);
   constexpr int LOADS_PER_TILE = LOADS_A + LOADS_B;
-  constexpr int MFRAG = BM / 2 / 16;         // per-wave M fragments
-  constexpr int NFRAG = BN / 4 / 16;         // per-wave N fragments
+  constexpr int MFRAG = BM / WAVES_M / 16;
+  constexpr int NFRAG = BN / WAVES_N / 16;
   constexpr int KSTEPS = BK / 32;
 
-  __shared__ unsigned short lds[NBUF * (TILE_A + TILE_B)];
+  __shared__ unsigned short lds[NBUF * TILE];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wm = wave >> 2;
-  const int wn = wave & 3;
+  const int wm = wave / WAVES_N;
+  const int wn = wave % WAVES_N;
 
   int nwg = gridDim.x;
   int orig = blockIdx.x;
@@ -68,31 +70,43 @@ gemm_tn_kernel(const unsigned short* __restrict__ A,
 
   const int KT = K / BK;
 
-  auto stage_tile = [&](int kt, int buf) {
-    const int k0 = kt * BK;
-    unsigned short* base_a = lds + buf * (TILE_A + TILE_B);
-    unsigned short* base_b = base_a + TILE_A;
+  // ---- hoisted staging state: per-load LDS byte offset + global ptr ----
+  int st_lds_a[LOADS_A], st_lds_b[LOADS_B];
+  const unsigned short* st_ga[LOADS_A];
+  const unsigned short* st_gb[LOADS_B];
+#pragma unroll
+  for (int j = 0; j < LOADS_A; ++j) {
+    int o = (tid + j * THREADS) * 16;
+    st_lds_a[j] = o;
+    int so = swz_off<ROWB, SWZ>(o);
+    st_ga[j] = A + (int64_t)(bm0 + so / ROWB) * K + (so % ROWB) / 2;
+  }
+#pragma unroll
+  for (int j = 0; j < LOADS_B; ++j) {
+    int o = (tid + j * THREADS) * 16;
+    st_lds_b[j] = o;
+    int so = swz_off<ROWB, SWZ>(o);
+    st_gb[j] = B + (int64_t)(bn0 + so / ROWB) * K + (so % ROWB) / 2;
+  }
+
+  // stage the tile currently pointed at by st_* into LDS buffer `buf`,
+  // then advance the global pointers by BK
+  auto stage_advance = [&](int buf) {
+    char* base_a = (char*)(lds + buf * TILE);
+    char* base_b = (char*)(lds + buf * TILE + TILE_A);
 #pragma unroll
     for (int j = 0; j < LOADS_A; ++j) {
-      int o = (tid + j * THREADS) * 16;
-      int so = swz_off<ROWB, SWZ>(o);
-      int row = so / ROWB;
-      int colb = so % ROWB;
-      const unsigned short* ga = A + (int64_t)(bm0 + row) * K + k0 + colb / 2;
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void*)ga,
-          (__attribute__((address_space(3))) void*)((char*)base_a + o), 16, 0, 0);
+          (const __attribute__((address_space(1))) void*)st_ga[j],
+          (__attribute__((address_space(3))) void*)(base_a + st_lds_a[j]), 16, 0, 0);
+      st_ga[j] += BK;
     }
 #pragma unroll
     for (int j = 0; j < LOADS_B; ++j) {
-      int o = (tid + j * THREADS) * 16;
-      int so = swz_off<ROWB, SWZ>(o);
-      int row = so / ROWB;
-      int colb = so % ROWB;
-      const unsigned short* gb = B + (int64_t)(bn0 + row) * K + k0 + colb / 2;
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void*)gb,
-          (__attribute__((address_space(3))) void*)((char*)base_b + o), 16, 0, 0);
+          (const __attribute__((address_space(1))) void*)st_gb[j],
+          (__attribute__((address_space(3))) void*)(base_b + st_lds_b[j]), 16, 0, 0);
+      st_gb[j] += BK;
     }
   };
 
@@ -102,32 +116,30 @@ gemm_tn_kernel(const unsigned short* __restrict__ A,
 #pragma unroll
     for (int j = 0; j < NFRAG; ++j) acc[i][j] = (floatx4){0.f, 0.f, 0.f, 0.f};
 
-  const int arow0 = wm * (BM / 2) + (lane & 15);
-  const int brow0 = wn * (BN / 4) + (lane & 15);
-  const int kch = lane >> 4;  // 0..3, 8 elems each (32 K per kstep)
-
-  auto lda = [&](int buf, int fm, int ks) -> shortx8 {
-    unsigned short* base = lds + buf * (TILE_A + TILE_B);
-    int off = swz_off<ROWB, SWZ>((arow0 + fm * 16) * ROWB + (ks * 4 + kch) * 16);
-    return *reinterpret_cast<shortx8*>((char*)base + off);
-  };
-  auto ldb = [&](int buf, int fn, int ks) -> shortx8 {
-    unsigned short* base = lds + buf * (TILE_A + TILE_B) + TILE_A;
-    int off = swz_off<ROWB, SWZ>((brow0 + fn * 16) * ROWB + (ks * 4 + kch) * 16);
-    return *reinterpret_cast<shortx8*>((char*)base + off);
-  };
+  // ---- hoisted LDS read offsets (bytes, within the tile) ----
+  const int arow0 = wm * (BM / WAVES_M) + (lane & 15);
+  const int brow0 = wn * (BN / WAVES_N) + (lane & 15);
+  const int kch = lane >> 4;
+  int a_off[MFRAG][KSTEPS], b_off[NFRAG][KSTEPS];
+#pragma unroll
+  for (int fm = 0; fm < MFRAG; ++fm)
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks)
+      a_off[fm][ks] = swz_off<ROWB, SWZ>((arow0 + fm * 16) * ROWB + (ks * 4 + kch) * 16);
+#pragma unroll
+  for (int fn = 0; fn < NFRAG; ++fn)
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks)
+      b_off[fn][ks] = swz_off<ROWB, SWZ>((brow0 + fn * 16) * ROWB + (ks * 4 + kch) * 16) + TILE_A * 2;
 
 #define RAW_BARRIER() asm volatile("s_barrier" ::: "memory")
-  // counted vmcnt: wait until at most n LOADS remain in flight
 #define VMCNT(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
-  constexpr int AHEAD = NBUF - 1;  // tiles staged ahead of compute
+  constexpr int AHEAD = NBUF - 1;
 
-  // prologue: stage tiles 0..AHEAD-1, wait for tile 0
-  for (int t = 0; t < AHEAD && t < KT; ++t) stage_tile(t, t % NBUF);
+  for (int t = 0; t < AHEAD && t < KT; ++t) stage_advance(t % NBUF);
   {
-    int inflight = (AHEAD < KT ? AHEAD : KT) * LOADS_PER_TILE;
-    int want = inflight - LOADS_PER_TILE;  // all but tile 0
-    // immediate-operand dispatch
+    int staged = AHEAD < KT ? AHEAD : KT;
+    int want = (staged - 1) * LOADS_PER_TILE;
     if (want >= 16) { VMCNT(16); }
     else if (want >= 12) { VMCNT(12); }
     else if (want >= 8) { VMCNT(8); }
@@ -136,57 +148,63 @@ gemm_tn_kernel(const unsigned short* __restrict__ A,
   }
   RAW_BARRIER();
 
-  for (int kt = 0; kt < KT; ++kt) {
-    const int buf = kt % NBUF;
-    // stage first (T3: issue loads BEFORE the ds_read+MFMA of this phase)
-    if (kt + AHEAD < KT) stage_tile(kt + AHEAD, (kt + AHEAD) % NBUF);
+  // main loop unrolled by NBUF so the LDS buffer base folds to a constant
+  int kt = 0;
+  while (kt < KT) {
+#pragma unroll
+    for (int ub = 0; ub < NBUF; ++ub) {
+      if (kt >= KT) break;
+      const char* base = (const char*)(lds + ub * TILE);
+      if (kt + AHEAD < KT) stage_advance((ub + AHEAD) % NBUF);
 
 #pragma unroll
-    for (int ks = 0; ks < KSTEPS; ++ks) {
-      shortx8 bfr[NFRAG];
-#pragma unroll
-      for (int fn = 0; fn < NFRAG; ++fn) bfr[fn] = ldb(buf, fn, ks);
-      // phase A: first half of M-frags
-      shortx8 afr[MFRAG / 2];
-#pragma unroll
-      for (int fm = 0; fm < MFRAG / 2; ++fm) afr[fm] = lda(buf, fm, ks);
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int fm = 0; fm < MFRAG / 2; ++fm)
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        shortx8 bfr[NFRAG];
 #pragma unroll
         for (int fn = 0; fn < NFRAG; ++fn)
-          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr[fm], bfr[fn], acc[fm][fn], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
-      // phase B: second half
+          bfr[fn] = *reinterpret_cast<const shortx8*>(base + b_off[fn][ks]);
+        shortx8 afr[MFRAG / 2];
 #pragma unroll
-      for (int fm = 0; fm < MFRAG / 2; ++fm) afr[fm] = lda(buf, fm + MFRAG / 2, ks);
-      __builtin_amdgcn_s_setprio(1);
+        for (int fm = 0; fm < MFRAG / 2; ++fm)
+          afr[fm] = *reinterpret_cast<const shortx8*>(base + a_off[fm][ks]);
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int fm = 0; fm < MFRAG / 2; ++fm)
+        for (int fm = 0; fm < MFRAG / 2; ++fm)
 #pragma unroll
-        for (int fn = 0; fn < NFRAG; ++fn)
-          acc[fm + MFRAG / 2][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr[fm], bfr[fn], acc[fm + MFRAG / 2][fn], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
-    }
+          for (int fn = 0; fn < NFRAG; ++fn)
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[fm], bfr[fn], acc[fm][fn], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+        for (int fm = 0; fm < MFRAG / 2; ++fm)
+          afr[fm] = *reinterpret_cast<const shortx8*>(base + a_off[fm + MFRAG / 2][ks]);
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int fm = 0; fm < MFRAG / 2; ++fm)
+#pragma unroll
+          for (int fn = 0; fn < NFRAG; ++fn)
+            acc[fm + MFRAG / 2][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[fm], bfr[fn], acc[fm + MFRAG / 2][fn], 0, 0, 0);
+        __builtin_amdgcn_s_setprio(0);
+      }
 
-    if (kt + 1 < KT) {
-      // counted wait: tile kt+1 must be landed; younger tiles stay in flight
-      int ahead_staged = (KT - 1 - kt < AHEAD ? KT - 1 - kt : AHEAD);
-      int want = (ahead_staged - 1) * LOADS_PER_TILE;
-      if (want >= 16) { VMCNT(16); }
-      else if (want >= 12) { VMCNT(12); }
-      else if (want >= 8) { VMCNT(8); }
-      else if (want >= 4) { VMCNT(4); }
-      else { VMCNT(0); }
-      RAW_BARRIER();
+      if (kt + 1 < KT) {
+        int ahead_staged = (KT - 1 - kt < AHEAD ? KT - 1 - kt : AHEAD);
+        int want = (ahead_staged - 1) * LOADS_PER_TILE;
+        if (want >= 16) { VMCNT(16); }
+        else if (want >= 12) { VMCNT(12); }
+        else if (want >= 8) { VMCNT(8); }
+        else if (want >= 4) { VMCNT(4); }
+        else { VMCNT(0); }
+        RAW_BARRIER();
+      }
+      ++kt;
     }
   }
 
   // epilogue: D mapping col=lane&15, row=4*(lane>>4)+r (guide §3, m89)
-  const int crow_base = bm0 + wm * (BM / 2) + 4 * (lane >> 4);
-  const int ccol_base = bn0 + wn * (BN / 4) + (lane & 15);
+  const int crow_base = bm0 + wm * (BM / WAVES_M) + 4 * (lane >> 4);
+  const int ccol_base = bn0 + wn * (BN / WAVES_N) + (lane & 15);
 #pragma unroll
   for (int fm = 0; fm < MFRAG; ++fm)
 #pragma unroll
@@ -202,8 +220,6 @@ gemm_tn_kernel(const unsigned short* __restrict__ A,
 #undef VMCNT
 }
 
-// Instantiations: V0 = 256x256x32 4-buf; V1 = 128x256x64 3-buf; V2 = V0
-// without LDS swizzle (A/B probe)
 extern "C" __global__ void __launch_bounds__(512, 1)
 gemm_tn_bf16_v0(const unsigned short* A, const unsigned short* B,
                 unsigned short* C, int M, int N, int K) {
@@ -220,4 +236,19 @@ extern "C" __global__ void __launch_bounds__(512, 1)
 gemm_tn_bf16_v2(const unsigned short* A, const unsigned short* B,
                 unsigned short* C, int M, int N, int K) {
   gemm_tn_kernel<256, 256, 32, 4, false>(A, B, C, M, N, K);
+}
+
+// V3: 128x128 tile, 4 waves, 4-deep pipeline -> 64KB LDS, 2 blocks/CU
+extern "C" __global__ void __launch_bounds__(256, 2)
+gemm_tn_bf16_v3(const unsigned short* A, const unsigned short* B,
+                unsigned short* C, int M, int N, int K) {
+  gemm_tn_kernel<128, 128, 32, 4, true, 2, 2>(A, B, C, M, N, K);
+}
+
+// V4: 512x256 tile, 16 waves (8Mx2N), 3-deep pipeline: higher arithmetic
+// intensity per staged byte (175 vs 131 FLOP/B) for the LLC-BW-bound regime
+extern "C" __global__ void __launch_bounds__(1024, 1)
+gemm_tn_bf16_v4(const unsigned short* A, const unsigned short* B,
+                unsigned short* C, int M, int N, int K) {
+  gemm_tn_kernel<512, 256, 32, 3, true, 8, 2>(A, B, C, M, N, K);
 }
