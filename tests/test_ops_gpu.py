@@ -254,3 +254,28 @@ def test_gemm_tn_mfma():
           + torch.arange(256, device="cuda").view(1, -1) * 0.001).bfloat16()
     got = C.gemm_tn(eye, bb, 0).float()
     assert torch.allclose(got, bb.float().t(), atol=0.05)
+
+
+@requires_gpu
+def test_flash_attention_bwd():
+    """Custom CDNA4 MFMA flash backward vs torch SDPA autograd (causal,
+    GQA, D=128)."""
+    import torch.nn.functional as F
+    from vescale_amd.ops import flash_attention_causal
+
+    torch.manual_seed(7)
+    B, Hq, Hkv, S, D = 2, 8, 2, 512, 128
+    q = torch.randn(B, Hq, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(B, Hkv, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(B, Hkv, S, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    dy = torch.randn(B, Hq, S, D, device="cuda", dtype=torch.bfloat16)
+    out = flash_attention_causal(q, k, v)
+    out.backward(dy)
+    gq, gk, gv = q.grad.clone(), k.grad.clone(), v.grad.clone()
+    q.grad = k.grad = v.grad = None
+    ref = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+    ref.backward(dy)
+    assert torch.allclose(out.float(), ref.float(), atol=5e-2, rtol=5e-2)
+    assert torch.allclose(gq.float(), q.grad.float(), atol=5e-2, rtol=8e-2)
+    assert torch.allclose(gk.float(), k.grad.float(), atol=8e-2, rtol=1e-1)
+    assert torch.allclose(gv.float(), v.grad.float(), atol=8e-2, rtol=1e-1)

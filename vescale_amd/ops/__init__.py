@@ -37,6 +37,7 @@ def require_ext():
 
 
 from .functional import (  # noqa: E402,F401
+    flash_attention_causal,
     fused_cross_entropy,
     rmsnorm,
     rope_apply,

@@ -12,6 +12,7 @@
 #include "cross_entropy.hip"
 #include "philox_random.hip"
 #include "gemm.hip"
+#include "attention_bwd.hip"
 
 #include <vector>
 
@@ -287,6 +288,71 @@ at::Tensor gemm_tn(at::Tensor a, at::Tensor b, int64_t variant) {
   return c;
 }
 
+// ------------------------------ flash-attention backward -------------------
+std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                               at::Tensor o, at::Tensor dout, at::Tensor lse,
+                               double scale) {
+  // q,o,dout: [B,Hq,S,D]; k,v: [B,Hkv,S,D]; lse: [B,Hq,S] fp32; causal
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
+  check_bf16_contig(o, "o");
+  int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2), D = (int)q.size(3);
+  int Hkv = (int)k.size(1);
+  TORCH_CHECK(D == FA_D, "fa_bwd supports head_dim 128");
+  TORCH_CHECK(S % FA_BLK == 0, "seq must be a multiple of 64");
+  TORCH_CHECK(lse.scalar_type() == at::kFloat);
+  auto lse_c = lse.contiguous();
+  auto dout_c = dout.contiguous();
+  auto delta = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  int64_t rows = (int64_t)B * Hq * S;
+  hipLaunchKernelGGL(fa_bwd_preprocess, dim3(grid_for(rows, 1, 4096)), dim3(256),
+                     0, cur_stream(),
+                     (const unsigned short*)dout_c.data_ptr(),
+                     (const unsigned short*)o.data_ptr(),
+                     delta.data_ptr<float>(), rows);
+  auto dq = at::empty_like(q);
+  auto dk_part = at::empty({B, Hq, S, D}, q.options());
+  auto dv_part = at::empty({B, Hq, S, D}, q.options());
+  dim3 gridkv(S / FA_BLK, Hq, B);
+  hipLaunchKernelGGL(fa_bwd_dkdv, gridkv, dim3(FA_THREADS), 0, cur_stream(),
+                     (const unsigned short*)q.data_ptr(),
+                     (const unsigned short*)k.data_ptr(),
+                     (const unsigned short*)v.data_ptr(),
+                     (const unsigned short*)dout_c.data_ptr(),
+                     lse_c.data_ptr<float>(), delta.data_ptr<float>(),
+                     (unsigned short*)dk_part.data_ptr(),
+                     (unsigned short*)dv_part.data_ptr(), B, Hq, Hkv, S,
+                     (float)scale);
+  hipLaunchKernelGGL(fa_bwd_dq, gridkv, dim3(FA_THREADS), 0, cur_stream(),
+                     (const unsigned short*)q.data_ptr(),
+                     (const unsigned short*)k.data_ptr(),
+                     (const unsigned short*)v.data_ptr(),
+                     (const unsigned short*)dout_c.data_ptr(),
+                     lse_c.data_ptr<float>(), delta.data_ptr<float>(),
+                     (unsigned short*)dq.data_ptr(), B, Hq, Hkv, S,
+                     (float)scale);
+  at::Tensor dk, dv;
+  if (Hkv == Hq) {
+    dk = dk_part;
+    dv = dv_part;
+  } else {
+    dk = at::empty({B, Hkv, S, D}, q.options());
+    dv = at::empty({B, Hkv, S, D}, q.options());
+    int64_t SD = (int64_t)S * D;
+    int grid = grid_for((int64_t)B * Hkv * SD / 8, 256);
+    hipLaunchKernelGGL(fa_bwd_reduce_gqa, dim3(grid), dim3(256), 0,
+                       cur_stream(),
+                       (const unsigned short*)dk_part.data_ptr(),
+                       (unsigned short*)dk.data_ptr(), B, Hq, Hkv, SD);
+    hipLaunchKernelGGL(fa_bwd_reduce_gqa, dim3(grid), dim3(256), 0,
+                       cur_stream(),
+                       (const unsigned short*)dv_part.data_ptr(),
+                       (unsigned short*)dv.data_ptr(), B, Hq, Hkv, SD);
+  }
+  return {dq, dk, dv};
+}
+
 // ------------------------------ philox random ------------------------------
 ShardDesc make_desc(const std::vector<int64_t>& gshape,
                     const std::vector<int64_t>& lshape,
@@ -388,6 +454,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("l2norm_sq", &l2norm_sq);
   m.def("scale_", &scale_);
   m.def("gemm_tn", &gemm_tn, pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("variant") = 0);
+  m.def("fa_bwd", &fa_bwd);
   m.def("philox_uniform_", &philox_uniform_);
   m.def("philox_normal_", &philox_normal_);
   m.def("philox_dropout", &philox_dropout);

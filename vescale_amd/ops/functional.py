@@ -298,6 +298,46 @@ def fused_cross_entropy(
 
 
 # ---------------------------------------------------------------------------
+# Flash attention: library forward (returns logsumexp) + our CDNA4 MFMA
+# backward (ops/csrc/attention_bwd.hip) — the stock backward is the profiled
+# bottleneck (~255 TF effective); causal, head_dim 128, GQA.
+# ---------------------------------------------------------------------------
+class _FlashAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        # q,k,v: [B, H, S, D] contiguous bf16
+        out, lse, *_ = torch.ops.aten._scaled_dot_product_flash_attention(
+            q, k, v, 0.0, True, False, scale=scale
+        )
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        dq, dk, dv = _ext().fa_bwd(
+            q, k, v, out.contiguous(), dout, lse.float(), ctx.scale
+        )
+        return dq, dk, dv, None
+
+
+def flash_attention_causal(q, k, v):
+    """Causal flash attention, [B,H,S,D] bf16; GQA via Hkv < Hq.  Falls back
+    to torch SDPA off-GPU / without the extension loudly on GPU."""
+    import math
+
+    scale = 1.0 / math.sqrt(q.shape[-1])
+    if _use_hip(q, k, v) and q.shape[-1] == 128 and q.shape[2] % 64 == 0:
+        return _FlashAttention.apply(
+            q.contiguous(), k.contiguous(), v.contiguous(), scale
+        )
+    return torch.nn.functional.scaled_dot_product_attention(
+        q, k, v, is_causal=True, enable_gqa=q.shape[1] != k.shape[1]
+    )
+
+
+# ---------------------------------------------------------------------------
 # flat AdamW + grad utilities (no autograd)
 # ---------------------------------------------------------------------------
 @torch.no_grad()
