@@ -28,11 +28,13 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--model", type=str, default="llama3_8b",
                     choices=["llama3_8b", "llama3_70b", "llama_tiny"])
-    ap.add_argument("--batch", type=int, default=2, help="per-GPU batch size")
+    ap.add_argument("--batch", type=int, default=4, help="per-GPU batch size")
     ap.add_argument("--seq", type=int, default=8192)
     ap.add_argument("--lr", type=float, default=3e-4)
     ap.add_argument("--activation-checkpointing", action="store_true")
     ap.add_argument("--no-master-weights", action="store_true")
+    ap.add_argument("--tp", type=int, default=1,
+                    help="tensor-parallel degree (BASELINE 2D config: TP x FSDP)")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -60,10 +62,21 @@ def main():
         args.batch, args.seq = 2, 64
 
     torch.manual_seed(1234)
-    mesh = init_device_mesh(device.type, (world_size,), mesh_dim_names=("DP",)) if world_size > 1 else None
+    tp = args.tp
+    assert world_size % tp == 0, "world size must be divisible by --tp"
+    dp = world_size // tp
+    tp_group = None
+    if world_size > 1:
+        if tp > 1:
+            mesh = init_device_mesh(device.type, (dp, tp), mesh_dim_names=("DP", "TP"))
+            tp_group = mesh.get_group(1)
+        else:
+            mesh = init_device_mesh(device.type, (world_size,), mesh_dim_names=("DP",))
+    else:
+        mesh = None
 
     with torch.device("meta"):
-        model = M.LlamaModel(cfg)
+        model = M.LlamaModel(cfg, tp_group=tp_group)
     model = model.to_empty(device=device)
     model.rope_table.copy_(
         M.build_rope_table(cfg.max_seq_len, cfg.head_dim, cfg.rope_theta).to(device)
@@ -73,14 +86,15 @@ def main():
     dtype = torch.bfloat16 if on_gpu else torch.float32
 
     if world_size > 1:
-        eng = FSDP(model, mesh, param_dtype=dtype, device=device)
+        eng = FSDP(model, mesh, mesh_dim=0, param_dtype=dtype, device=device)
     else:
         eng = FSDP(model, None, param_dtype=dtype, device=device)
     opt = FlatAdamW(eng, lr=args.lr, grad_clip=1.0,
                     use_master_weights=not args.no_master_weights)
 
     B, S = args.batch, args.seq
-    gen = torch.Generator(device="cpu").manual_seed(4321 + rank)
+    dp_rank = rank // tp  # TP ranks share the batch
+    gen = torch.Generator(device="cpu").manual_seed(4321 + dp_rank)
     x = torch.randint(0, cfg.vocab_size, (B, S), generator=gen).to(device)
     y = torch.roll(x, -1, dims=1)
 
@@ -111,7 +125,7 @@ def main():
         dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
     sec = float(elapsed.item())
     ms_per_step = sec / args.steps * 1e3
-    tokens_per_step = B * S * n_gpus
+    tokens_per_step = B * S * (n_gpus // tp)
     tok_s = tokens_per_step * args.steps / sec
 
     if rank == 0:
@@ -130,9 +144,9 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": args.model if on_gpu else "llama_tiny(cpu-smoke)",
-                "global_batch": B * n_gpus,
+                "global_batch": B * dp,
                 "seq_len": S,
-                "parallelism": f"fsdp{n_gpus}",
+                "parallelism": f"tp{tp}_fsdp{dp}" if tp > 1 else f"fsdp{n_gpus}",
                 "final_loss": float(loss.detach().float().cpu()),
                 "activation_checkpointing": args.activation_checkpointing,
             },

@@ -56,6 +56,21 @@ def _align(n: int, a: int = ALIGN) -> int:
     return (n + a - 1) // a * a
 
 
+def _duplicate_mesh_dim_group(mesh: DeviceMesh, dim: int):
+    """A SECOND ProcessGroup per mesh-dim row (every rank loops every row —
+    the collective new_group contract)."""
+    perm = [d for d in range(mesh.ndim) if d != dim] + [dim]
+    rows = mesh.mesh.permute(perm).reshape(-1, mesh.mesh.size(dim))
+    mine = None
+    cur = dist.get_rank()
+    for row in rows:
+        ranks = row.tolist()
+        g = dist.new_group(ranks=ranks)
+        if cur in ranks:
+            mine = g
+    return mine
+
+
 class FSDPUnit:
     """One flat-parameter group."""
 
@@ -153,12 +168,14 @@ class FSDP(nn.Module):
         prefetch: bool = True,
         reshard_after_forward: bool = True,
         device: Optional[torch.device] = None,
+        mesh_dim: Optional[int] = None,
     ):
         super().__init__()
         self.module = module
         self.mesh = mesh
+        self.mesh_dim = mesh_dim if mesh_dim is not None else (mesh.ndim - 1 if mesh is not None else 0)
         if mesh is not None:
-            self.pg = mesh.get_group(mesh.ndim - 1) if process_group is None else process_group
+            self.pg = mesh.get_group(self.mesh_dim) if process_group is None else process_group
         else:
             self.pg = process_group
         self.world_size = dist.get_world_size(self.pg) if (self.pg is not None and dist.is_initialized()) else 1
@@ -173,11 +190,16 @@ class FSDP(nn.Module):
         self._on_gpu = dev.type == "cuda"
 
         # separate communicators so AG and RS streams can't reorder against
-        # each other across ranks
+        # each other across ranks.  new_group is COLLECTIVE over the world:
+        # with a mesh we loop every submesh row (all ranks call with every
+        # ranks-list); with a bare process_group we reuse it (stream overlap
+        # then shares one communicator).
         if self.world_size > 1:
-            ranks = dist.get_process_group_ranks(self.pg)
             self.ag_pg = self.pg
-            self.rs_pg = dist.new_group(ranks=ranks) if dist.is_initialized() else self.pg
+            if mesh is not None and dist.is_initialized():
+                self.rs_pg = _duplicate_mesh_dim_group(mesh, self.mesh_dim)
+            else:
+                self.rs_pg = self.pg
         else:
             self.ag_pg = self.rs_pg = None
 

@@ -24,6 +24,7 @@ from ..ops import (
     rope_qkv,
     swiglu_packed,
 )
+from .llama_tp import TPContext, copy_to_tp, reduce_from_tp
 
 
 @dataclass
@@ -84,49 +85,54 @@ class RMSNorm(nn.Module):
 
 
 class Attention(nn.Module):
-    def __init__(self, cfg: LlamaConfig):
+    def __init__(self, cfg: LlamaConfig, tp: TPContext = None):
         super().__init__()
         self.cfg = cfg
+        self.tp = tp or TPContext.single()
+        assert cfg.n_heads % self.tp.world == 0 and cfg.n_kv_heads % self.tp.world == 0
+        self.hq = cfg.n_heads // self.tp.world
+        self.hkv = cfg.n_kv_heads // self.tp.world
         d, hd = cfg.dim, cfg.head_dim
-        kv = cfg.n_kv_heads * hd
-        # fused qkv projection: one GEMM instead of three
-        self.wqkv = nn.Linear(d, d + 2 * kv, bias=False)
-        self.wo = nn.Linear(d, d, bias=False)
+        # fused qkv projection: one GEMM instead of three (TP: local heads)
+        self.wqkv = nn.Linear(d, (self.hq + 2 * self.hkv) * hd, bias=False)
+        self.wo = nn.Linear(self.hq * hd, d, bias=False)
 
     def forward(self, x: torch.Tensor, rope_table: torch.Tensor) -> torch.Tensor:
         B, S, _ = x.shape
         cfg = self.cfg
+        x = copy_to_tp(x, self.tp)
         qkv = self.wqkv(x)
-        q, k, v = rope_qkv(
-            qkv, rope_table, cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
-        )
+        q, k, v = rope_qkv(qkv, rope_table, self.hq, self.hkv, cfg.head_dim)
         q = q.transpose(1, 2)  # [B, H, S, D]
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)
         o = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
-        o = o.transpose(1, 2).reshape(B, S, cfg.dim)
-        return self.wo(o)
+        o = o.transpose(1, 2).reshape(B, S, self.hq * cfg.head_dim)
+        return reduce_from_tp(self.wo(o), self.tp)
 
 
 class FeedForward(nn.Module):
-    def __init__(self, cfg: LlamaConfig):
+    def __init__(self, cfg: LlamaConfig, tp: TPContext = None):
         super().__init__()
-        # w1 (gate) and w3 (up) fused into one GEMM
-        self.w13 = nn.Linear(cfg.dim, 2 * cfg.ffn_dim, bias=False)
-        self.w2 = nn.Linear(cfg.ffn_dim, cfg.dim, bias=False)
-        self.ffn_dim = cfg.ffn_dim
+        self.tp = tp or TPContext.single()
+        assert cfg.ffn_dim % self.tp.world == 0
+        self.ffn_local = cfg.ffn_dim // self.tp.world
+        # w1 (gate) and w3 (up) fused into one GEMM (TP: local ffn slice)
+        self.w13 = nn.Linear(cfg.dim, 2 * self.ffn_local, bias=False)
+        self.w2 = nn.Linear(self.ffn_local, cfg.dim, bias=False)
 
     def forward(self, x):
-        return self.w2(swiglu_packed(self.w13(x)))
+        x = copy_to_tp(x, self.tp)
+        return reduce_from_tp(self.w2(swiglu_packed(self.w13(x))), self.tp)
 
 
 class TransformerBlock(nn.Module):
-    def __init__(self, cfg: LlamaConfig):
+    def __init__(self, cfg: LlamaConfig, tp: TPContext = None):
         super().__init__()
         self.attn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
-        self.attn = Attention(cfg)
+        self.attn = Attention(cfg, tp)
         self.ffn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
-        self.ffn = FeedForward(cfg)
+        self.ffn = FeedForward(cfg, tp)
 
     def forward(self, x, rope_table):
         x = x + self.attn(self.attn_norm(x), rope_table)
@@ -135,11 +141,14 @@ class TransformerBlock(nn.Module):
 
 
 class LlamaModel(nn.Module):
-    def __init__(self, cfg: LlamaConfig):
+    def __init__(self, cfg: LlamaConfig, tp_group=None):
         super().__init__()
         self.cfg = cfg
+        self.tp = TPContext.from_group(tp_group)
         self.tok_embeddings = nn.Embedding(cfg.vocab_size, cfg.dim)
-        self.layers = nn.ModuleList(TransformerBlock(cfg) for _ in range(cfg.n_layers))
+        self.layers = nn.ModuleList(
+            TransformerBlock(cfg, self.tp) for _ in range(cfg.n_layers)
+        )
         self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
         self.output = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
         if cfg.tie_embeddings:
