@@ -113,10 +113,11 @@ fa_bwd_dkdv(const unsigned short* __restrict__ q,
             unsigned short* __restrict__ dk_part,  // [B,Hq,S,D]
             unsigned short* __restrict__ dv_part,  // [B,Hq,S,D]
             int B, int Hq, int Hkv, int S, float scale) {
-  __shared__ unsigned short lq[FA_D * ROWS_N];     // Q^T  [128][72]
-  __shared__ unsigned short ldo[FA_D * ROWS_N];    // dO^T [128][72]
-  __shared__ unsigned short lqr[FA_BLK * ROWS_D];  // Q rows [64][136]
-  __shared__ unsigned short ldor[FA_BLK * ROWS_D]; // dO rows [64][136]
+  // Each q-tile buffer serves two lives: rows [64][136] for the ST/dP^T
+  // GEMMs, then (rewritten from the registers that staged it) transposed
+  // [128][72] for the dV/dK GEMMs.  Sized for the larger layout.
+  __shared__ unsigned short lq[FA_D * ROWS_N];     // Q rows -> Q^T
+  __shared__ unsigned short ldo[FA_D * ROWS_N];    // dO rows -> dO^T
   __shared__ unsigned short lpt[FA_BLK * ROWS_N];  // P^T  [64][72]
   __shared__ unsigned short lds_t[FA_BLK * ROWS_N];// dS^T [64][72]
   __shared__ float lse_s[FA_BLK];
@@ -134,31 +135,19 @@ fa_bwd_dkdv(const unsigned short* __restrict__ q,
   const int64_t kbase = (((int64_t)b * Hkv + hkv) * S) * FA_D;
   const int64_t lbase = ((int64_t)b * Hq + h) * S;
 
-  // K tile rows held in registers for the whole block: wave w owns rows
-  // [16w, 16w+16); lane holds its operand fragments for all 4 k-chunks
-  fa_shortx8 kfrag[4];
+  fa_shortx8 kfrag[4], vfrag[4];
   {
     const unsigned short* kg = k + kbase + (int64_t)(jkv * FA_BLK) * FA_D;
+    const unsigned short* vg = v + kbase + (int64_t)(jkv * FA_BLK) * FA_D;
+    int row = wave * 16 + (lane & 15);
 #pragma unroll
     for (int kc = 0; kc < 4; ++kc) {
-      int row = wave * 16 + (lane & 15);
       int col = kc * 32 + (lane >> 4) * 8;
       kfrag[kc] = *reinterpret_cast<const fa_shortx8*>(kg + row * FA_D + col);
-    }
-  }
-  // V tile fragments (for dP^T = V . dO^T, contraction over d)
-  fa_shortx8 vfrag[4];
-  {
-    const unsigned short* vg = v + kbase + (int64_t)(jkv * FA_BLK) * FA_D;
-#pragma unroll
-    for (int kc = 0; kc < 4; ++kc) {
-      int row = wave * 16 + (lane & 15);
-      int col = kc * 32 + (lane >> 4) * 8;
       vfrag[kc] = *reinterpret_cast<const fa_shortx8*>(vg + row * FA_D + col);
     }
   }
 
-  // accumulators: wave's dK/dV rows [16, 128] = 8 col frags
   fa_floatx4 accK[8], accV[8];
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
@@ -170,45 +159,45 @@ fa_bwd_dkdv(const unsigned short* __restrict__ q,
   for (int iq = jkv; iq < n_qtiles; ++iq) {
     const unsigned short* qg = q + qbase + (int64_t)(iq * FA_BLK) * FA_D;
     const unsigned short* dog = dout + qbase + (int64_t)(iq * FA_BLK) * FA_D;
-    stage_rows_t(qg, lq, tid);
-    stage_rows_t(dog, ldo, tid);
-    stage_rows(qg, lqr, tid);
-    stage_rows(dog, ldor, tid);
+    // stage rows from global ONCE; keep values in registers for the later
+    // in-LDS transpose
+    fa_shortx8 regq[4], regdo[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int e = (tid + j * FA_THREADS) * 8;
+      int r = e / FA_D, c = e % FA_D;
+      regq[j] = *reinterpret_cast<const fa_shortx8*>(qg + r * FA_D + c);
+      regdo[j] = *reinterpret_cast<const fa_shortx8*>(dog + r * FA_D + c);
+      *reinterpret_cast<fa_shortx8*>(lq + r * ROWS_D + c) = regq[j];
+      *reinterpret_cast<fa_shortx8*>(ldo + r * ROWS_D + c) = regdo[j];
+    }
     if (tid < FA_BLK) {
       lse_s[tid] = lse[lbase + iq * FA_BLK + tid];
       dlt_s[tid] = delta[lbase + iq * FA_BLK + tid];
     }
     __syncthreads();
 
-    // ST[kv][q] = K . Q^T  (contraction d), wave rows 16w..16w+16
-    fa_floatx4 st[4];   // 4 q col-frags
+    // ST[kv][q] = K . Q^T; dPT[kv][q] = V . dO^T  (Y operands = rows)
+    fa_floatx4 st[4], dpt[4];
 #pragma unroll
-    for (int f = 0; f < 4; ++f) st[f] = (fa_floatx4){0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-    for (int kc = 0; kc < 4; ++kc) {
-#pragma unroll
-      for (int f = 0; f < 4; ++f) {
-        // Y[j=q][k=d] = Q[q][d]: read Q ROWS
-        fa_shortx8 bq = frag(lqr, ROWS_D, f, kc, lane);
-        st[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfrag[kc], bq, st[f], 0, 0, 0);
-      }
+    for (int f = 0; f < 4; ++f) {
+      st[f] = (fa_floatx4){0.f, 0.f, 0.f, 0.f};
+      dpt[f] = (fa_floatx4){0.f, 0.f, 0.f, 0.f};
     }
-    // dPT[kv][q] = V . dO^T (contraction d) -> operand dO rows
-    fa_floatx4 dpt[4];
-#pragma unroll
-    for (int f = 0; f < 4; ++f) dpt[f] = (fa_floatx4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
     for (int kc = 0; kc < 4; ++kc) {
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        fa_shortx8 bd = frag(ldor, ROWS_D, f, kc, lane);
+        fa_shortx8 bq = frag(lq, ROWS_D, f, kc, lane);
+        st[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfrag[kc], bq, st[f], 0, 0, 0);
+        fa_shortx8 bd = frag(ldo, ROWS_D, f, kc, lane);
         dpt[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag[kc], bd, dpt[f], 0, 0, 0);
       }
     }
 
     // P^T = exp(ST*scale - lse[q]); dS^T = P^T * (dPT - delta[q]) * scale
-    // D-layout: col(q) = lane&15 + 16f, row(kv) = 16*wave + 4*(lane>>4)+r
     const int kvrow = jkv * FA_BLK + wave * 16 + 4 * (lane >> 4);
+    __syncthreads();  // rows layouts fully consumed before the rewrite
 #pragma unroll
     for (int f = 0; f < 4; ++f) {
       int qcol = f * 16 + (lane & 15);
@@ -225,18 +214,28 @@ fa_bwd_dkdv(const unsigned short* __restrict__ q,
         lds_t[row * ROWS_N + qcol] = f32_to_bf16(ds);
       }
     }
+    // rewrite lq/ldo as TRANSPOSED [128][72] from the staging registers
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int e = (tid + j * FA_THREADS) * 8;
+      int r = e / FA_D, c = e % FA_D;
+#pragma unroll
+      for (int x = 0; x < 8; ++x) {
+        lq[(c + x) * ROWS_N + r] = (unsigned short)regq[j][x];
+        ldo[(c + x) * ROWS_N + r] = (unsigned short)regdo[j][x];
+      }
+    }
     __syncthreads();
 
-    // dV += P^T . dO  (contraction q): X = P^T rows, Y[j=d][k=q] = dO^T
-    // dK += dS^T . Q  (contraction q): Y = Q^T
+    // dV += P^T . dO (Y = dO^T); dK += dS^T . Q (Y = Q^T)
 #pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {     // q contraction: 64 = 2 chunks
+    for (int kc = 0; kc < 2; ++kc) {
+      fa_shortx8 xp = frag(lpt, ROWS_N, wave, kc, lane);
+      fa_shortx8 xs = frag(lds_t, ROWS_N, wave, kc, lane);
 #pragma unroll
-      for (int f = 0; f < 8; ++f) {      // d col frags
-        fa_shortx8 xp = frag(lpt, ROWS_N, wave, kc, lane);
+      for (int f = 0; f < 8; ++f) {
         fa_shortx8 yd = frag(ldo, ROWS_N, f, kc, lane);
         accV[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(xp, yd, accV[f], 0, 0, 0);
-        fa_shortx8 xs = frag(lds_t, ROWS_N, wave, kc, lane);
         fa_shortx8 yq = frag(lq, ROWS_N, f, kc, lane);
         accK[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(xs, yq, accK[f], 0, 0, 0);
       }
@@ -244,7 +243,6 @@ fa_bwd_dkdv(const unsigned short* __restrict__ q,
     __syncthreads();
   }
 
-  // write partials: rows kv (D-layout rows), cols d
   unsigned short* dkg = dk_part + qbase + (int64_t)(jkv * FA_BLK) * FA_D;
   unsigned short* dvg = dv_part + qbase + (int64_t)(jkv * FA_BLK) * FA_D;
 #pragma unroll
@@ -271,8 +269,7 @@ fa_bwd_dq(const unsigned short* __restrict__ q,
           const float* __restrict__ delta,
           unsigned short* __restrict__ dq,
           int B, int Hq, int Hkv, int S, float scale) {
-  __shared__ unsigned short lk[FA_BLK * ROWS_D];   // K rows [64][136]
-  __shared__ unsigned short lkt[FA_D * ROWS_N];    // K^T [128][72]
+  __shared__ unsigned short lk[FA_D * ROWS_N];     // K rows -> K^T
   __shared__ unsigned short lv[FA_BLK * ROWS_D];   // V rows [64][136]
   __shared__ unsigned short lds_s[FA_BLK * ROWS_N];// dS [64q][72]
 
@@ -288,28 +285,24 @@ fa_bwd_dq(const unsigned short* __restrict__ q,
   const int64_t kbase = (((int64_t)b * Hkv + hkv) * S) * FA_D;
   const int64_t lbase = ((int64_t)b * Hq + h) * S;
 
-  // Q and dO fragments for this block's q-tile: wave rows 16w..16w+16
   fa_shortx8 qfrag[4], dofrag[4];
   {
     const unsigned short* qg = q + qbase + (int64_t)(iq * FA_BLK) * FA_D;
     const unsigned short* dog = dout + qbase + (int64_t)(iq * FA_BLK) * FA_D;
+    int row = wave * 16 + (lane & 15);
 #pragma unroll
     for (int kc = 0; kc < 4; ++kc) {
-      int row = wave * 16 + (lane & 15);
       int col = kc * 32 + (lane >> 4) * 8;
       qfrag[kc] = *reinterpret_cast<const fa_shortx8*>(qg + row * FA_D + col);
       dofrag[kc] = *reinterpret_cast<const fa_shortx8*>(dog + row * FA_D + col);
     }
   }
   float lse_r[4], dlt_r[4];
-  {
-    // D-layout rows this lane writes: 16w + 4*(lane>>4) + r
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      int row = iq * FA_BLK + wave * 16 + 4 * (lane >> 4) + r;
-      lse_r[r] = lse[lbase + row];
-      dlt_r[r] = delta[lbase + row];
-    }
+  for (int r = 0; r < 4; ++r) {
+    int row = iq * FA_BLK + wave * 16 + 4 * (lane >> 4) + r;
+    lse_r[r] = lse[lbase + row];
+    dlt_r[r] = delta[lbase + row];
   }
 
   fa_floatx4 accQ[8];
@@ -319,16 +312,23 @@ fa_bwd_dq(const unsigned short* __restrict__ q,
   for (int jkv = 0; jkv <= iq; ++jkv) {
     const unsigned short* kg = k + kbase + (int64_t)(jkv * FA_BLK) * FA_D;
     const unsigned short* vg = v + kbase + (int64_t)(jkv * FA_BLK) * FA_D;
-    stage_rows(kg, lk, tid);
-    stage_rows_t(kg, lkt, tid);
-    stage_rows(vg, lv, tid);
+    fa_shortx8 regk[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int e = (tid + j * FA_THREADS) * 8;
+      int r = e / FA_D, c = e % FA_D;
+      regk[j] = *reinterpret_cast<const fa_shortx8*>(kg + r * FA_D + c);
+      *reinterpret_cast<fa_shortx8*>(lk + r * ROWS_D + c) = regk[j];
+      *reinterpret_cast<fa_shortx8*>(lv + r * ROWS_D + c) =
+          *reinterpret_cast<const fa_shortx8*>(vg + r * FA_D + c);
+    }
     __syncthreads();
 
-    // S[q][kv] = Q . K^T (contraction d): X frag = qfrag, Y = K rows
-    fa_floatx4 s[4], dp[4];
+    // S[q][kv] = Q . K^T (Y = K rows); dP[q][kv] = dO . V^T (Y = V rows)
+    fa_floatx4 sacc[4], dp[4];
 #pragma unroll
     for (int f = 0; f < 4; ++f) {
-      s[f] = (fa_floatx4){0.f, 0.f, 0.f, 0.f};
+      sacc[f] = (fa_floatx4){0.f, 0.f, 0.f, 0.f};
       dp[f] = (fa_floatx4){0.f, 0.f, 0.f, 0.f};
     }
 #pragma unroll
@@ -336,13 +336,14 @@ fa_bwd_dq(const unsigned short* __restrict__ q,
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
         fa_shortx8 yk = frag(lk, ROWS_D, f, kc, lane);
-        s[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kc], yk, s[f], 0, 0, 0);
+        sacc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[kc], yk, sacc[f], 0, 0, 0);
         fa_shortx8 yv = frag(lv, ROWS_D, f, kc, lane);
         dp[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dofrag[kc], yv, dp[f], 0, 0, 0);
       }
     }
+    __syncthreads();  // rows consumed; lk may be rewritten transposed
 
-    // dS[q][kv] = P * (dP - delta[q]) * scale;  P = exp(S*scale - lse[q])
+    // dS[q][kv] = P * (dP - delta[q]) * scale
     const int qrow0 = iq * FA_BLK + wave * 16 + 4 * (lane >> 4);
 #pragma unroll
     for (int f = 0; f < 4; ++f) {
@@ -350,21 +351,29 @@ fa_bwd_dq(const unsigned short* __restrict__ q,
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         bool valid = (qrow0 + r) >= kvcol;
-        float p = valid ? __expf(s[f][r] * scale - lse_r[r]) : 0.f;
+        float p = valid ? __expf(sacc[f][r] * scale - lse_r[r]) : 0.f;
         float ds = p * (dp[f][r] - dlt_r[r]) * scale;
         int row = wave * 16 + 4 * (lane >> 4) + r;
         lds_s[row * ROWS_N + f * 16 + (lane & 15)] = f32_to_bf16(ds);
       }
     }
+    // rewrite lk as K^T [128][72] from the staging registers
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int e = (tid + j * FA_THREADS) * 8;
+      int r = e / FA_D, c = e % FA_D;
+#pragma unroll
+      for (int x = 0; x < 8; ++x) lk[(c + x) * ROWS_N + r] = (unsigned short)regk[j][x];
+    }
     __syncthreads();
 
-    // dQ += dS . K (contraction kv): X = dS rows, Y[j=d][k=kv] = K^T
+    // dQ += dS . K (Y = K^T)
 #pragma unroll
     for (int kc = 0; kc < 2; ++kc) {
+      fa_shortx8 xs = frag(lds_s, ROWS_N, wave, kc, lane);
 #pragma unroll
       for (int f = 0; f < 8; ++f) {
-        fa_shortx8 xs = frag(lds_s, ROWS_N, wave, kc, lane);
-        fa_shortx8 yk = frag(lkt, ROWS_N, f, kc, lane);
+        fa_shortx8 yk = frag(lk, ROWS_N, f, kc, lane);
         accQ[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(xs, yk, accQ[f], 0, 0, 0);
       }
     }
