@@ -171,3 +171,86 @@ def _t_dmp(rank, ws):
 
 def test_dmp_auto_plan():
     spawn(2, _t_dmp)
+
+
+# --------------------- cross-mesh + MoE buffers + patches -------------------
+def _t_cross_mesh(rank, ws):
+    import torch.distributed as dist
+    from vescale_amd.dtensor import DTensor, Replicate, Shard, distribute_tensor
+    from vescale_amd.dtensor.cross_mesh import cross_mesh_recv, cross_mesh_send
+    from vescale_amd.dtensor.device_mesh import DeviceMesh
+
+    # rank 0 = "stage 0" mesh, rank 1 = "stage 1" mesh
+    mesh0 = DeviceMesh("cpu", [0], _init_process_groups=False)
+    mesh1 = DeviceMesh("cpu", [1], _init_process_groups=False)
+    if rank == 0:
+        t = torch.arange(12, dtype=torch.float32).reshape(3, 4)
+        d = DTensor.from_local(t, mesh0, [Replicate()])
+        cross_mesh_send(d, dst_rank=1)
+    else:
+        d = cross_mesh_recv(src_rank=0, dst_mesh=mesh1)
+        assert torch.equal(
+            d._local_tensor, torch.arange(12, dtype=torch.float32).reshape(3, 4)
+        )
+        assert d.shape == (3, 4)
+
+
+def test_cross_mesh_send_recv():
+    spawn(2, _t_cross_mesh)
+
+
+def _t_moe_buffer(rank, ws):
+    from vescale_amd.models.mixtral import MixtralModel, mixtral_tiny
+    from vescale_amd.moe._moe_param_buffer import MoEParamBuffer
+    from vescale_amd.moe.moe_optimizer import MoEOptimizer, MoEScheduler
+    from vescale_amd.moe import BasicExpertsAllocator
+
+    torch.manual_seed(0)
+    cfg = mixtral_tiny()
+    model = MixtralModel(cfg)
+    model.init_weights()
+    buf = MoEParamBuffer(model, dp_group=None)
+    assert len(buf.layer_buffers) == cfg.n_layers
+    expert_params = [p for b in buf.layer_buffers.values() for (p, _) in b._views]
+    inner = torch.optim.AdamW(expert_params, lr=1e-3)
+    opt = MoEOptimizer(inner, buf)
+    x = torch.randint(0, cfg.vocab_size, (2, 16))
+    loss = model(x, torch.roll(x, -1, 1))
+    loss.backward()
+    buf.run_reduce_scatter()
+    before = expert_params[0].detach().clone()
+    opt.step()
+    assert not torch.equal(before, expert_params[0])
+    opt.zero_grad()
+    sched = MoEScheduler(model, BasicExpertsAllocator(cfg.n_experts, 1), opt)
+    sched.record_tokens(torch.randint(0, cfg.n_experts, (64,)))
+    assert sched.step() in (False, True)
+
+
+def test_moe_param_buffer_and_optimizer():
+    spawn(1, _t_moe_buffer)
+
+
+def _t_dispatch_patches(rank, ws):
+    from vescale_amd.dtensor import Replicate, distribute_tensor
+    from vescale_amd.dtensor.dispatch import get_dispatcher
+
+    mesh = init_device_mesh("cpu", (1,))
+    d = get_dispatcher()
+    seen = []
+
+    def pre(op, args, kwargs):
+        seen.append(str(op))
+        return None
+
+    d.register_pre_patch(pre)
+    try:
+        t = distribute_tensor(torch.ones(4), mesh, [Replicate()])
+        _ = t + t
+        assert any("add" in s for s in seen)
+    finally:
+        d._pre_patches.clear()
+
+
+def test_dispatch_patches():
+    spawn(1, _t_dispatch_patches)

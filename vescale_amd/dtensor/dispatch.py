@@ -43,6 +43,8 @@ class OpDispatcher:
         self._cache: Dict[Any, OutputSharding] = {}
         self._rng_tracker = None
         self._random_ops = set()
+        self._pre_patches: list = []
+        self._post_patches: list = []
 
     # -- registration ----------------------------------------------------
     def register_rule(self, op, fn):
@@ -61,6 +63,16 @@ class OpDispatcher:
         for o in self._expand(op):
             self._random_ops.add(o)
 
+    # decoupled special-op patching (parity: legacy _dispatch_patch.py:37
+    # DispatchPrePatch / post patch + _dispatch_bypass.py): pre patches may
+    # rewrite (op, args, kwargs) before propagation; post patches may
+    # rewrite the result.
+    def register_pre_patch(self, fn):
+        self._pre_patches.append(fn)
+
+    def register_post_patch(self, fn):
+        self._post_patches.append(fn)
+
     @staticmethod
     def _expand(op):
         if isinstance(op, torch._ops.OpOverloadPacket):
@@ -69,6 +81,18 @@ class OpDispatcher:
 
     # -- main entry ------------------------------------------------------
     def dispatch(self, op, args, kwargs):
+        for pre in self._pre_patches:
+            r = pre(op, args, kwargs)
+            if r is not None:
+                op, args, kwargs = r
+        res = self._dispatch_inner(op, args, kwargs)
+        for post in self._post_patches:
+            r = post(op, args, kwargs, res)
+            if r is not None:
+                res = r
+        return res
+
+    def _dispatch_inner(self, op, args, kwargs):
         from .dtensor import DTensor
 
         h = self._bypass.get(op)
