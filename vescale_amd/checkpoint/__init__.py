@@ -131,7 +131,9 @@ def load(
             process_group=process_group,
         )
         # push back into stateful objects that need it
-        if hasattr(obj, "load_state_dict") and not hasattr(obj, "sharded_state_dict"):
+        if hasattr(obj, "load_sharded_state_dict"):
+            obj.load_sharded_state_dict(sd)
+        elif hasattr(obj, "load_state_dict") and not hasattr(obj, "sharded_state_dict"):
             try:
                 obj.load_state_dict(sd)
             except Exception:

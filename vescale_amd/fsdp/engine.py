@@ -221,6 +221,28 @@ class FSDP(nn.Module):
         self._install_hooks()
         self._exec_order: List[int] = []  # recorded forward order of units
         self._in_backward = False
+        self._warmup_comms()
+
+    def _warmup_comms(self):
+        """Eagerly initialize BOTH RCCL communicators in the same order on
+        every rank (lazy init inside the stream-parallel backward could
+        otherwise interleave differently across ranks)."""
+        if self.world_size <= 1 or not dist.is_initialized():
+            return
+        t = torch.ones(self.world_size, device=self.device)
+        try:
+            dist.all_gather_into_tensor(
+                torch.empty(self.world_size * self.world_size, device=self.device),
+                t, group=self.ag_pg,
+            )
+            out = torch.empty(1, device=self.device)
+            dist.reduce_scatter_tensor(out, t, group=self.rs_pg)
+        except RuntimeError:
+            dist.all_reduce(t, group=self.ag_pg)
+            if self.rs_pg is not self.ag_pg:
+                dist.all_reduce(t, group=self.rs_pg)
+        if self._on_gpu:
+            torch.cuda.synchronize()
 
     # ------------------------------------------------------------------
     def _build_units(self, unit_classes: Sequence[str]):
