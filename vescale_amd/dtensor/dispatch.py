@@ -181,12 +181,27 @@ class OpDispatcher:
         local_args = tuple(fill(a) for a in args_schema)
         local_kwargs = {k: fill(v) for k, v in kwargs_schema.items()}
 
-        # run the local op (random ops inside the RNG tracker region)
+        # run the local op (random ops inside the RNG tracker region; the
+        # ThreadBased tracker EXECUTES them via the sharded-philox HIP
+        # kernels for bitwise single-device parity)
         if op in self._random_ops and self._rng_tracker is not None:
-            with self._rng_tracker._distribute_region(specs[0]):
-                res = op(*local_args, **local_kwargs)
+            res = NotImplemented
+            if hasattr(self._rng_tracker, "exec_random_op"):
+                res = self._rng_tracker.exec_random_op(
+                    op, local_args, local_kwargs, specs[0]
+                )
+            if res is NotImplemented:
+                with self._rng_tracker._distribute_region(specs[0]):
+                    res = op(*local_args, **local_kwargs)
         else:
             res = op(*local_args, **local_kwargs)
+        global _dbg
+        if _dbg is None:
+            from ..debug.debug_log import DebugLogger as _dbg_cls
+
+            _dbg = _dbg_cls
+        if _dbg.enabled():
+            _dbg.log_op(op)
 
         return self._wrap(res, sharding.output_spec, args)
 
@@ -292,6 +307,7 @@ class _ReplicateOut:
 
 _REPLICATE_OUT = _ReplicateOut()
 
+_dbg = None
 _dispatcher: Optional[OpDispatcher] = None
 
 

@@ -88,13 +88,101 @@ class OffsetBasedRNGTracker(_RNGStateTracker):
 
 
 class ThreadBasedRNGTracker(OffsetBasedRNGTracker):
-    """Bitwise single-device parity tracker: our sharded-philox HIP kernels
-    read the spec set here and index philox by *global* element id (see
-    vescale_amd/ops/functional.py sharded_uniform_/sharded_dropout).  For
-    ops that fall back to stock kernels, behaves like OffsetBased."""
+    """Bitwise single-device parity tracker (the reference's patched-kernel
+    semantics): random aten ops on GPU DTensors are
+    EXECUTED by our sharded-philox HIP kernels with the shard's GLOBAL
+    element indexing, so a TP/DP-sharded random fill is bitwise-identical
+    to our own single-GPU fill.  Ops without a philox implementation fall
+    back to OffsetBased semantics."""
 
     def current_spec(self) -> Optional[DTensorSpec]:
         return self._current_spec
+
+    def _desc(self, spec: DTensorSpec):
+        """(gshape, lshape, offset, flat_offset, is_flat) for the philox
+        kernels.  Returns None for layouts the kernels can't index."""
+        from .placement_types import InterleavedShard
+
+        coord = spec.mesh.get_coordinate()
+        if coord is None:
+            return None
+        for p in spec.placements:
+            if isinstance(p, InterleavedShard):
+                return None
+        if spec.has_ragged():
+            md = next(
+                i for i, p in enumerate(spec.placements) if p.is_ragged_shard()
+            )
+            p = spec.placements[md]
+            g = list(spec.shape)
+            numel = 1
+            for s in g:
+                numel *= s
+            off = p.local_offset_numel(tuple(g), coord[md])
+            n = p.local_numel(tuple(g), coord[md])
+            return ([numel], [n], [off], off, True)
+        gshape = list(spec.shape)
+        lshape = list(spec.local_shape(coord))
+        offs = list(spec.local_offsets(coord))
+        return (gshape, lshape, offs, 0, False)
+
+    def exec_random_op(self, op, local_args, local_kwargs, spec: DTensorSpec):
+        """Run a random aten op with global philox indexing; returns the
+        local result or NotImplemented to fall back."""
+        import torch as _t
+
+        from ..ops import has_ext
+
+        aten = _t.ops.aten
+        x = local_args[0]
+        if not (isinstance(x, _t.Tensor) and x.is_cuda and has_ext()):
+            return NotImplemented
+        d = self._desc(spec)
+        if d is None:
+            return NotImplemented
+        from ..ops import _C
+
+        gshape, lshape, offs, foff, flat = d
+        seed = self.seed
+        philox_off = self._offset
+        self._offset += 4
+        pkt = op.overloadpacket
+        if pkt == aten.native_dropout:
+            p = float(local_args[1])
+            train = local_args[2] if len(local_args) > 2 else True
+            if not train or p == 0.0:
+                return NotImplemented
+            if x.dtype != _t.bfloat16:
+                return NotImplemented
+            out, mask = _C.philox_dropout(
+                x.contiguous(), gshape, lshape, offs, foff, flat, seed,
+                philox_off, p, True,
+            )
+            return out, mask.to(_t.bool)
+        if pkt == aten.uniform_:
+            lo = float(local_args[1]) if len(local_args) > 1 else 0.0
+            hi = float(local_args[2]) if len(local_args) > 2 else 1.0
+            if x.dtype not in (_t.bfloat16, _t.float32):
+                return NotImplemented
+            _C.philox_uniform_(x, gshape, lshape, offs, foff, flat, seed, philox_off, lo, hi)
+            return x
+        if pkt == aten.normal_:
+            mean = float(local_args[1]) if len(local_args) > 1 else 0.0
+            std = float(local_args[2]) if len(local_args) > 2 else 1.0
+            if x.dtype not in (_t.bfloat16, _t.float32):
+                return NotImplemented
+            _C.philox_normal_(x, gshape, lshape, offs, foff, flat, seed, philox_off, mean, std)
+            return x
+        if pkt in (aten.rand_like, aten.randn_like):
+            out = _t.empty_like(x)
+            if out.dtype not in (_t.bfloat16, _t.float32):
+                return NotImplemented
+            if pkt == aten.rand_like:
+                _C.philox_uniform_(out, gshape, lshape, offs, foff, flat, seed, philox_off, 0.0, 1.0)
+            else:
+                _C.philox_normal_(out, gshape, lshape, offs, foff, flat, seed, philox_off, 0.0, 1.0)
+            return out
+        return NotImplemented
 
 
 class TensorParallelRNGTracker(_RNGStateTracker):
