@@ -106,3 +106,20 @@ def _t_nanogpt_tp_sp(rank, ws):
 
 def test_nanogpt_tp_sp():
     spawn(2, _t_nanogpt_tp_sp)
+
+
+def _t_factory_mode(rank, ws):
+    from vescale_amd.dmodule._factory import FactoryDispatchMode
+    from vescale_amd.dtensor import DTensor, init_device_mesh
+
+    mesh = init_device_mesh("cpu", (ws,))
+    with FactoryDispatchMode(mesh):
+        z = torch.zeros(4, 4)
+        o = torch.ones(3)
+    assert isinstance(z, DTensor) and isinstance(o, DTensor)
+    assert torch.equal(z.full_tensor(), torch.zeros(4, 4))
+    assert torch.equal(o.full_tensor(), torch.ones(3))
+
+
+def test_factory_mode():
+    spawn(2, _t_factory_mode)

@@ -51,7 +51,7 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
   check_bf16_contig(dy, "dy");
   check_bf16_contig(x, "x");
   int hidden = (int)x.size(-1);
-  TORCH_CHECK(hidden <= 16384, "rmsnorm_bwd supports hidden <= 16384");
+  TORCH_CHECK(hidden <= 8192, "rmsnorm_bwd supports hidden <= 8192");
   int64_t rows = x.numel() / hidden;
   auto dx = at::empty_like(x);
   auto dw = at::zeros({hidden}, x.options().dtype(at::kFloat));

@@ -72,47 +72,60 @@ rmsnorm_bwd_bf16(const unsigned short* __restrict__ dy,
                  int64_t n_rows, int hidden) {
   __shared__ float lds[BLOCK / WAVE];
   const int vec = 8;
-  const int cols_per_thread = (hidden / vec + BLOCK - 1) / BLOCK * vec;
-  // register accumulator for this thread's columns (max 64 cols => H<=16384)
-  float dw_acc[64];
-#pragma unroll
-  for (int j = 0; j < 64; ++j) dw_acc[j] = 0.f;
-
+  // per-thread register cache of the row data: pass 1 reads dy/x/w from
+  // global ONCE; pass 2 consumes the registers (hidden <= 8192 at BLOCK
+  // 256 -> up to 4 iters x 8 elems live per tensor)
   const int per_row_iters = (hidden / vec + BLOCK - 1) / BLOCK;
+  float dw_acc[32];
+#pragma unroll
+  for (int j = 0; j < 32; ++j) dw_acc[j] = 0.f;
+  // w is row-invariant: load once per thread
+  float wreg[32];
+  #pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    int i = (it * BLOCK + threadIdx.x) * vec;
+    if (i < hidden) {
+      short8v wv = *reinterpret_cast<const short8v*>(w + i);
+#pragma unroll
+      for (int j = 0; j < vec; ++j) wreg[it * vec + j] = bf16_to_f32((unsigned short)wv[j]);
+    }
+  }
+
   for (int64_t row = blockIdx.x; row < n_rows; row += gridDim.x) {
     const unsigned short* dyr = dy + row * hidden;
     const unsigned short* xr = x + row * hidden;
     unsigned short* dxr = dx + row * hidden;
     float rrms = rrms_in[row];
     float dot = 0.f;
-    for (int it = 0; it < per_row_iters; ++it) {
+    float dyreg[32], xreg[32];
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
       int i = (it * BLOCK + threadIdx.x) * vec;
       if (i < hidden) {
         short8v dv = *reinterpret_cast<const short8v*>(dyr + i);
         short8v xv = *reinterpret_cast<const short8v*>(xr + i);
-        short8v wv = *reinterpret_cast<const short8v*>(w + i);
 #pragma unroll
         for (int j = 0; j < vec; ++j) {
-          dot += bf16_to_f32((unsigned short)dv[j]) * bf16_to_f32((unsigned short)wv[j]) *
-                 bf16_to_f32((unsigned short)xv[j]);
+          float dyf = bf16_to_f32((unsigned short)dv[j]);
+          float xf = bf16_to_f32((unsigned short)xv[j]);
+          dyreg[it * vec + j] = dyf;
+          xreg[it * vec + j] = xf;
+          dot += dyf * wreg[it * vec + j] * xf;
         }
       }
     }
     float tdot = block_reduce_sum<BLOCK>(dot, lds);
     float coef = tdot * rrms * rrms * rrms / (float)hidden;
-    for (int it = 0; it < per_row_iters; ++it) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
       int i = (it * BLOCK + threadIdx.x) * vec;
       if (i < hidden) {
-        short8v dv = *reinterpret_cast<const short8v*>(dyr + i);
-        short8v xv = *reinterpret_cast<const short8v*>(xr + i);
-        short8v wv = *reinterpret_cast<const short8v*>(w + i);
         short8v o;
 #pragma unroll
         for (int j = 0; j < vec; ++j) {
-          float dyf = bf16_to_f32((unsigned short)dv[j]);
-          float xf = bf16_to_f32((unsigned short)xv[j]);
-          float wf = bf16_to_f32((unsigned short)wv[j]);
-          o[j] = (short)f32_to_bf16(rrms * wf * dyf - xf * coef);
+          float dyf = dyreg[it * vec + j];
+          float xf = xreg[it * vec + j];
+          o[j] = (short)f32_to_bf16(rrms * wreg[it * vec + j] * dyf - xf * coef);
           dw_acc[it * vec + j] += dyf * xf * rrms;
         }
         *reinterpret_cast<short8v*>(dxr + i) = o;
@@ -120,7 +133,8 @@ rmsnorm_bwd_bf16(const unsigned short* __restrict__ dy,
     }
   }
   // flush dw
-  for (int it = 0; it < per_row_iters; ++it) {
+  #pragma unroll
+  for (int it = 0; it < 4; ++it) {
     int i = (it * BLOCK + threadIdx.x) * vec;
     if (i < hidden) {
 #pragma unroll
