@@ -41,6 +41,7 @@ class OpDispatcher:
         self._handlers: Dict[Any, Callable] = {}
         self._bypass: Dict[Any, Callable] = {}
         self._cache: Dict[Any, OutputSharding] = {}
+        self._fast_cache: Dict[Any, OutputSharding] = {}
         self._rng_tracker = None
         self._random_ops = set()
         self._pre_patches: list = []
@@ -102,6 +103,70 @@ class OpDispatcher:
         if h is not None:
             return h(self, op, args, kwargs)
 
+        # ---- fast path: flat positional args, hashable kwargs, warm cache.
+        # Skips schema construction entirely (the eager hot-loop tax the
+        # reference warns about at vescale/dtensor/_dispatch.py:253-258).
+        fast = not kwargs or all(
+            isinstance(v, (int, float, bool, str, type(None), torch.dtype))
+            for v in kwargs.values()
+        )
+        if fast:
+            key = [op]
+            locals_ = []
+            specs = []
+            kp = key.append
+            for a in args:
+                if isinstance(a, DTensor):
+                    sp = a._spec
+                    locals_.append(a._local_tensor)
+                    specs.append(sp)
+                    kp((sp.placements, tuple(sp.tensor_meta.shape), sp.tensor_meta.dtype))
+                elif isinstance(a, torch.Tensor):
+                    fast = False
+                    break
+                elif isinstance(a, (list, tuple)):
+                    fast = False
+                    break
+                else:
+                    kp(a)
+            if fast and specs:
+                if kwargs:
+                    for k in kwargs:
+                        kp((k, kwargs[k]))
+                try:
+                    entry = self._fast_cache.get(tuple(key))
+                except TypeError:
+                    entry = None
+                if entry is not None:
+                    sharding = entry
+                    mesh = specs[0].mesh
+                    if sharding.input_targets is not None:
+                        for i, tgt in enumerate(sharding.input_targets):
+                            if tgt is None or tuple(tgt) == specs[i].placements:
+                                continue
+                            tgt_spec = DTensorSpec(mesh, tuple(tgt), specs[i].tensor_meta)
+                            locals_[i] = redistribute_local_tensor(
+                                locals_[i], specs[i], tgt_spec
+                            )
+                            specs[i] = tgt_spec
+                    it = iter(locals_)
+                    local_args = tuple(
+                        next(it) if isinstance(a, DTensor) else a for a in args
+                    )
+                    if op in self._random_ops and self._rng_tracker is not None:
+                        res = NotImplemented
+                        if hasattr(self._rng_tracker, "exec_random_op"):
+                            res = self._rng_tracker.exec_random_op(
+                                op, local_args, kwargs, specs[0]
+                            )
+                        if res is NotImplemented:
+                            with self._rng_tracker._distribute_region(specs[0]):
+                                res = op(*local_args, **kwargs)
+                    else:
+                        res = op(*local_args, **kwargs)
+                    return self._wrap(res, sharding.output_spec, args)
+        # ---- slow path (also populates the fast cache) ----
+
         # flatten: collect DTensor args in stable order
         specs: List[DTensorSpec] = []
         locals_: List[torch.Tensor] = []
@@ -158,6 +223,29 @@ class OpDispatcher:
             sharding = self._propagate(schema)
             if cache_key is not None:
                 self._cache[cache_key] = sharding
+        # mirror into the fast cache when the signature is flat
+        try:
+            if (not kwargs or all(
+                isinstance(v, (int, float, bool, str, type(None), torch.dtype))
+                for v in kwargs.values()
+            )) and not any(isinstance(a, (list, tuple)) for a in args):
+                fkey = [op]
+                plain_ok = True
+                for a in args:
+                    if isinstance(a, DTensor):
+                        sp0 = a._spec
+                        fkey.append((sp0.placements, tuple(sp0.tensor_meta.shape), sp0.tensor_meta.dtype))
+                    elif isinstance(a, torch.Tensor):
+                        plain_ok = False
+                        break
+                    else:
+                        fkey.append(a)
+                if plain_ok:
+                    for k in kwargs:
+                        fkey.append((k, kwargs[k]))
+                    self._fast_cache[tuple(fkey)] = sharding
+        except TypeError:
+            pass
 
         # redistribute inputs
         if sharding.input_targets is not None:
@@ -292,6 +380,7 @@ class OpDispatcher:
 
     def clear_cache(self):
         self._cache.clear()
+        self._fast_cache.clear()
 
 
 def _fix_dtype(spec: DTensorSpec, res: torch.Tensor) -> DTensorSpec:
