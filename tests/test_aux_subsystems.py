@@ -254,3 +254,29 @@ def _t_dispatch_patches(rank, ws):
 
 def test_dispatch_patches():
     spawn(1, _t_dispatch_patches)
+
+
+def test_sock_streamer():
+    """ndtimeline unix-socket collector round-trip."""
+    import tempfile as tf
+    import time as _time
+
+    from vescale_amd.ndtimeline.sock_streamer import NDtimelineStreamer, SockHandler
+    from vescale_amd.ndtimeline.timer import Span
+
+    with tf.TemporaryDirectory() as td:
+        sock = os.path.join(td, "nd.sock")
+        streamer = NDtimelineStreamer(sock).start()
+        got = []
+        streamer.handlers.append(lambda spans: got.extend(spans))
+        h = SockHandler(sock)
+        h([Span("forward-compute", 1.0, 2.0, 0, 3)])
+        h([Span("grad-reduce-scatter", 5.0, 1.0, 1, 3)])
+        for _ in range(100):
+            if len(got) >= 2:
+                break
+            _time.sleep(0.02)
+        h.close()
+        streamer.stop()
+        assert len(got) == 2
+        assert got[0].metric == "forward-compute" and got[1].rank == 1

@@ -197,3 +197,33 @@ def _t_pp_dp_3d(rank, ws):
 
 def test_pp_dp_zero2_3d():
     spawn(4, _t_pp_dp_3d)
+
+
+def _t_shared_params(rank, ws):
+    """Tied-embedding sync across stages (reference sync_shared_params)."""
+    import torch.distributed as dist
+    from vescale_amd.pipe.pipe_stage import PipeModule, build_shared_module_group
+
+    emb = nn.Linear(4, 4, bias=False)
+    stage = PipeModule([emb], rank, ws)
+    build_shared_module_group(
+        stage, [["weight"]], {"weight": 0 if rank == 0 else 1} if False else {"weight": rank},
+        rank,
+    )
+    # both stages "own" a copy -> group over both, grads averaged by allreduce
+    groups = build_shared_module_group(
+        stage, [["weight"]], {"weight": rank}, rank
+    )
+    # simulate distinct grads, then sync
+    emb.weight.grad = torch.full((4, 4), float(rank + 1))
+    # group contains only this stage (stage_of_fqn maps to own stage) ->
+    # rebuild with a group spanning both stages
+    stage.shared_param_groups = [
+        {"pg": dist.group.WORLD, "param": emb.weight, "fqns": ["weight"]}
+    ]
+    stage.sync_shared_params()
+    assert torch.allclose(emb.weight.grad, torch.full((4, 4), 3.0))
+
+
+def test_pp_shared_params_sync():
+    spawn(2, _t_shared_params)

@@ -59,6 +59,43 @@ class PipeModule(nn.Module):
             p.grad = g
 
 
+def build_shared_module_group(
+    stage: PipeModule,
+    shared_fqns: Sequence[Sequence[str]],
+    stage_of_fqn: dict,
+    my_stage: int,
+    stage_to_rank=None,
+):
+    """Build allreduce groups for TIED parameters living on different
+    stages (reference pipe_stage.py:311 build_shared_module_group — e.g.
+    input/output embeddings).  shared_fqns: groups of parameter fqns that
+    are one logical weight; stage_of_fqn maps fqn -> owning stage.
+    Creates one ProcessGroup per tied group (ALL ranks iterate every group
+    — the collective new_group contract) and records the local param."""
+    import torch.distributed as dist
+
+    stage_to_rank = stage_to_rank or (lambda s: s)
+    groups = []
+    for fqns in shared_fqns:
+        stages = sorted({stage_of_fqn[f] for f in fqns})
+        ranks = [stage_to_rank(s) for s in stages]
+        pg = dist.new_group(ranks=ranks) if dist.is_initialized() and len(ranks) > 1 else None
+        entry = {"pg": pg, "param": None, "fqns": list(fqns)}
+        if my_stage in stages:
+            params = dict(stage.named_parameters())
+            for f in fqns:
+                if stage_of_fqn[f] == my_stage:
+                    # fqn within the stage module namespace
+                    short = f.split(".")[-1]
+                    for n, p in params.items():
+                        if n.endswith(f) or n.endswith(short):
+                            entry["param"] = p
+                            break
+        groups.append(entry)
+    stage.shared_param_groups = groups
+    return groups
+
+
 def uniform_split(modules: Sequence[nn.Module], n_parts: int) -> List[List[nn.Module]]:
     """Split a module list into n contiguous parts balanced by parameter
     count (reference PipelineSplitMethodType.PARAMETERS/UNIFORM)."""
