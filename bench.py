@@ -128,6 +128,18 @@ def main():
     tokens_per_step = B * S * (n_gpus // tp)
     tok_s = tokens_per_step * args.steps / sec
 
+    # MFU per the reference's estimator (open_llama_4D_benchmark/
+    # llama_mfu_calculator.py:22-29 — 3x forward FLOPs, Kaplan-style) with
+    # the MI355X bf16 DENSE peak (2.5 PF/GPU; AMD's 5 PF figure is 2:1-
+    # sparse) in place of the A100/H100 table.
+    n_params = cfg.num_params()
+    emb = cfg.vocab_size * cfg.dim
+    fwd_flops_per_tok = 2 * (n_params - emb) + (
+        2 * 2 * cfg.n_layers * cfg.dim * S * 0.5  # causal attention
+    )
+    total_flops = 3 * fwd_flops_per_tok * tokens_per_step
+    mfu = total_flops / sec * args.steps / (n_gpus * 2.5e15)
+
     if rank == 0:
         out = {
             "metric": "tokens/sec (whole node) Llama-3-8B veScale-FSDP",
@@ -148,6 +160,8 @@ def main():
                 "seq_len": S,
                 "parallelism": f"tp{tp}_fsdp{dp}" if tp > 1 else f"fsdp{n_gpus}",
                 "final_loss": float(loss.detach().float().cpu()),
+                "mfu_est": round(mfu, 4),
+                "mfu_peak_ref": "2.5 PF bf16 dense per MI355X",
                 "activation_checkpointing": args.activation_checkpointing,
             },
         }

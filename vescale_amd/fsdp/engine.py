@@ -450,6 +450,19 @@ class FSDP(nn.Module):
                 if u.grad_shard is not None:
                     u.grad_shard.div_(self.world_size)
 
+    def prefetch_after_step(self):
+        """Issue the root unit's (and first block's) all-gather right after
+        the optimizer step so the next forward doesn't stall on the ~GB
+        embedding/head gather (called by FlatAdamW.step)."""
+        if self.world_size <= 1:
+            return
+        if self._root_unit is not None and not self._root_unit._is_unsharded:
+            self._unshard(self._root_unit)
+        if self._exec_order:
+            first = self.units[self._exec_order[0]]
+            if not first._is_unsharded:
+                self._unshard(first)
+
     def zero_grad_buffers(self):
         for u in self.units:
             u.grad_full = None
@@ -465,7 +478,7 @@ class FSDP(nn.Module):
         if self._root_unit is not None:
             if not self._root_unit._is_unsharded:
                 self._unshard(self._root_unit)
-                self._wait_unshard(self._root_unit)
+            self._wait_unshard(self._root_unit)  # also covers post-step prefetch
             if self.training and torch.is_grad_enabled():
                 self._root_unit.attach_grad_views()
         if self.world_size == 1:

@@ -77,6 +77,8 @@ class FlatAdamW:
                 clip_scale=clip_scale_t,  # fused into the kernel (one read)
             )
         self.zero_grad()
+        # overlap the next forward's first all-gathers with host-side work
+        self.engine.prefetch_after_step()
 
     def zero_grad(self):
         self.engine.zero_grad_buffers()
