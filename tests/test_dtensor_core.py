@@ -293,3 +293,92 @@ def _t_softmax_dropout(rank, ws):
 
 def test_softmax_dropout():
     spawn(2, _t_softmax_dropout)
+
+
+def _t_redistribute_fuzz(rank, ws):
+    """Seeded fuzz over shapes x placement pairs: full_tensor is invariant
+    under any redistribute chain."""
+    import random
+
+    mesh = init_device_mesh("cpu", (ws,))
+    rng = random.Random(1234)
+    placement_pool = [
+        [Replicate()], [Shard(0)], [Shard(1)], [Shard(2)],
+        [InterleavedShard(0, 2)],
+        [RaggedShard((0,), (1, 2))], [RaggedShard((0, 1), (3, 1))],
+    ]
+    for trial in range(30):
+        dims = rng.choice([2, 3])
+        shape = [rng.choice([2, 4, 6, 8, 12]) for _ in range(dims)]
+        # keep divisibility for interleave/ragged candidates
+        shape[0] = rng.choice([4, 8, 12])
+        torch.manual_seed(trial)
+        g = torch.randn(*shape)
+        cands = [p for p in placement_pool if _valid(p[0], shape, ws)]
+        src = rng.choice(cands)
+        mid = rng.choice(cands)
+        dst = rng.choice(cands)
+        d = distribute_tensor(g, mesh, src)
+        d = d.redistribute(placements=mid)
+        d = d.redistribute(placements=dst)
+        assert torch.allclose(d.full_tensor(), g, atol=1e-6), (
+            trial, shape, src, mid, dst,
+        )
+
+
+def _valid(p, shape, ws):
+    if isinstance(p, InterleavedShard):
+        return shape[p.dim] % (p.interleaved_size * ws) == 0
+    if isinstance(p, RaggedShard):
+        if len(p.local_units) != ws:
+            return False
+        flat = 1
+        for d in p.dims:
+            flat *= shape[d]
+        return flat % sum(p.local_units) == 0
+    if isinstance(p, Shard):
+        return p.dim < len(shape)
+    return True
+
+
+def test_redistribute_fuzz():
+    spawn(2, _t_redistribute_fuzz)
+
+
+def _t_uneven_from_local(rank, ws):
+    import torch.distributed as dist
+
+    mesh = init_device_mesh("cpu", (ws,))
+    # rank 0 has 3 rows, rank 1 has 5 rows
+    n = 3 if rank == 0 else 5
+    local = torch.full((n, 4), float(rank))
+    d = DTensor.from_local(local, mesh, [Shard(0)], support_uneven=True)
+    assert d.shape == (8, 4)
+    full = d.full_tensor()
+    assert torch.equal(full[:3], torch.zeros(3, 4))
+    assert torch.equal(full[3:], torch.ones(5, 4))
+
+
+def test_uneven_from_local():
+    spawn(2, _t_uneven_from_local)
+
+
+def _t_explicit_collectives(rank, ws):
+    from vescale_amd.dtensor import vescale_all_gather, vescale_all_reduce
+
+    mesh = init_device_mesh("cpu", (ws,))
+    g = torch.randn(8, 4)
+    torch.manual_seed(3)
+    g = torch.randn(8, 4)
+    d = distribute_tensor(g, mesh, [Shard(0)])
+    r = vescale_all_gather(d)
+    assert all(p.is_replicate() for p in r.placements)
+    assert torch.equal(r.to_local(), g)
+    local = torch.full((2, 2), float(rank + 1))
+    p = DTensor.from_local(local, mesh, [Partial()], shape=torch.Size((2, 2)))
+    s = vescale_all_reduce(p)
+    assert torch.equal(s.to_local(), torch.full((2, 2), 3.0))
+
+
+def test_explicit_collectives():
+    spawn(2, _t_explicit_collectives)

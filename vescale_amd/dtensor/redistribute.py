@@ -213,7 +213,9 @@ def _transition(
             if isinstance(t, Replicate):
                 return out
             return c_split_replicate(out, t, w, coord[mesh_dim])
-        raise NotImplementedError(f"InterleavedShard -> {t}")
+        # any other target: go through Replicate
+        rep = _transition(local, spec, cur_placements, mesh_dim, c, Replicate(), coord)
+        return _transition(rep, spec, cur_placements, mesh_dim, Replicate(), t, coord)
 
     if isinstance(c, RaggedShard):
         if isinstance(t, Replicate):
@@ -232,6 +234,14 @@ def _transition(
         d = c.dim
         logical = _logical_shape_at(spec, others, coord)
         sizes = _shard_sizes_on_dim(logical[d], w)
+        if local.size(d) != sizes[my]:
+            # custom uneven layout (from_local support_uneven): agree on the
+            # TRUE per-rank sizes (reference gather_local_tensor_shape)
+            import torch.distributed as dist
+
+            shapes: list = [None] * w
+            dist.all_gather_object(shapes, int(local.size(d)), group=mesh.get_group(mesh_dim))
+            sizes = [int(s) for s in shapes]
         if isinstance(t, Replicate):
             even = all(s == sizes[0] for s in sizes)
             return cc.mesh_all_gather(
