@@ -43,13 +43,15 @@ def main():
     n_gpus = max(world_size, 1)
 
     on_gpu = torch.cuda.is_available()
-    device = torch.device(f"cuda:{local_rank}") if on_gpu else torch.device("cpu")
+    ndev = torch.cuda.device_count() if on_gpu else 0
+    device = torch.device(f"cuda:{local_rank % max(ndev, 1)}") if on_gpu else torch.device("cpu")
     if on_gpu:
         torch.cuda.set_device(device)
 
     if world_size > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group("nccl" if on_gpu else "gloo")
+        backend = os.environ.get("VESCALE_BENCH_BACKEND", "nccl" if on_gpu else "gloo")
+        dist.init_process_group(backend)
 
     from vescale_amd.dtensor import init_device_mesh
     from vescale_amd.fsdp import FSDP, FlatAdamW

@@ -121,9 +121,10 @@ class DeviceMesh:
 
     def _setup_device(self) -> None:
         if self.device_type == "cuda" and torch.cuda.is_available():
-            # one process per GPU: pin device by LOCAL_RANK (or rank % ngpu)
-            local_rank = int(os.environ.get("LOCAL_RANK", dist.get_rank() % torch.cuda.device_count()))
-            torch.cuda.set_device(local_rank)
+            # one process per GPU: pin device by LOCAL_RANK (modulo device
+            # count so oversubscribed debug runs on 1 GPU still work)
+            local_rank = int(os.environ.get("LOCAL_RANK", dist.get_rank()))
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
 
     def _init_dim_groups(self) -> None:
         if self.mesh.ndim == 1 and self.mesh.numel() == dist.get_world_size():
