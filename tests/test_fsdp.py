@@ -88,3 +88,40 @@ def _t_state_dict(rank, ws):
 
 def test_fsdp_sharded_state_dict():
     spawn(2, _t_state_dict)
+
+
+def _t_grad_accumulation(rank, ws):
+    """Two micro-backwards before step must ACCUMULATE (not overwrite)."""
+    from vescale_amd.dtensor import init_device_mesh
+
+    torch.manual_seed(7)
+    cfg = llama_tiny()
+    model = LlamaModel(cfg)
+    model.init_weights()
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("DP",))
+    eng = FSDP(model, mesh, param_dtype=torch.float32, device=torch.device("cpu"))
+    g = torch.Generator().manual_seed(3)
+    x1 = torch.randint(0, cfg.vocab_size, (2, 32), generator=g)
+    x2 = torch.randint(0, cfg.vocab_size, (2, 32), generator=g)
+
+    # accumulate two backwards
+    eng(x1, torch.roll(x1, -1, 1)).backward()
+    eng(x2, torch.roll(x2, -1, 1)).backward()
+    eng.finish_grad_sync()
+    acc = {u.name: u.grad_shard.clone() for u in eng.units}
+
+    # reference: separate single-backward runs summed
+    eng.zero_grad_buffers()
+    eng(x1, torch.roll(x1, -1, 1)).backward()
+    eng.finish_grad_sync()
+    g1 = {u.name: u.grad_shard.clone() for u in eng.units}
+    eng.zero_grad_buffers()
+    eng(x2, torch.roll(x2, -1, 1)).backward()
+    eng.finish_grad_sync()
+    for u in eng.units:
+        want = g1[u.name] + u.grad_shard
+        assert torch.allclose(acc[u.name], want, atol=1e-5), u.name
+
+
+def test_fsdp_grad_accumulation():
+    spawn(2, _t_grad_accumulation)

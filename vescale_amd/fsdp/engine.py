@@ -220,6 +220,9 @@ class FSDP(nn.Module):
         self._build_units(unit_classes)
         self._install_hooks()
         self._exec_order: List[int] = []  # recorded forward order of units
+        self._exec_order_set = set()
+        for i, u in enumerate(self.units):
+            u._idx = i
         self._in_backward = False
         self._warmup_comms()
 
@@ -347,8 +350,9 @@ class FSDP(nn.Module):
             if not unit._is_unsharded:
                 self._unshard(unit)
             self._wait_unshard(unit)
-            if unit not in [self.units[i] for i in self._exec_order]:
-                self._exec_order.append(self.units.index(unit))
+            if unit._idx not in self._exec_order_set:
+                self._exec_order.append(unit._idx)
+                self._exec_order_set.add(unit._idx)
             # prefetch the next unit in recorded order
             if self.prefetch:
                 nxt = self._next_unit_after(unit, forward=True)
@@ -360,7 +364,7 @@ class FSDP(nn.Module):
 
     def _make_fwd_post(self, unit: FSDPUnit):
         def hook(mod, args, out):
-            if not self._in_backward and not (self.training and torch.is_grad_enabled() and False):
+            if not self._in_backward:
                 self._reshard(unit)
             return None
 
@@ -383,7 +387,7 @@ class FSDP(nn.Module):
 
     def _next_unit_after(self, unit: FSDPUnit, forward: bool) -> Optional[FSDPUnit]:
         try:
-            pos = self._exec_order.index(self.units.index(unit))
+            pos = self._exec_order.index(unit._idx)
         except ValueError:
             return None
         if forward:
@@ -407,36 +411,44 @@ class FSDP(nn.Module):
 
     def _finish_unit_grads(self, unit: FSDPUnit):
         # all grads of this unit accumulated into unit.grad_full
-        _t = ndtimeit(ndm.GRAD_RS); _t.__enter__()
-        for p in unit.params:
-            p.grad = None
-        if self.world_size == 1:
-            unit.grad_shard = unit.grad_full
-            return
-        if unit.grad_shard is None:
-            unit.grad_shard = torch.empty(
-                unit.shard_numel, dtype=unit.param_dtype, device=self.device
+        with ndtimeit(ndm.GRAD_RS):
+            for p in unit.params:
+                p.grad = None
+            if self.world_size == 1:
+                unit.grad_shard = unit.grad_full
+                return
+            accumulate = unit.grad_shard is not None
+            target = (
+                torch.empty(unit.shard_numel, dtype=unit.param_dtype, device=self.device)
+                if accumulate
+                else torch.empty(unit.shard_numel, dtype=unit.param_dtype, device=self.device)
             )
-            self._grad_shard_fresh = True
-        # reduce-scatter on the RS stream
-        gf = unit.grad_full
-        if self._on_gpu:
-            self._rs_stream.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(self._rs_stream):
-                dist.reduce_scatter_tensor(
-                    unit.grad_shard, gf, op=dist.ReduceOp.AVG if hasattr(dist.ReduceOp, "AVG") else dist.ReduceOp.SUM,
-                    group=self.rs_pg,
-                )
-            gf.record_stream(self._rs_stream)
-            unit.grad_shard.record_stream(self._rs_stream)
-        else:
-            dist.all_reduce(gf, group=self.rs_pg)
-            gf.div_(self.world_size)
-            unit.grad_shard.copy_(gf[unit.shard_off : unit.shard_off + unit.shard_numel])
-        unit.grad_full = None
-        # reshard params after backward
-        self._reshard(unit)
-        _t.__exit__(None, None, None)
+            gf = unit.grad_full
+            op = dist.ReduceOp.AVG if hasattr(dist.ReduceOp, "AVG") else dist.ReduceOp.SUM
+            if self._on_gpu:
+                self._rs_stream.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(self._rs_stream):
+                    dist.reduce_scatter_tensor(target, gf, op=op, group=self.rs_pg)
+                    if accumulate:
+                        # grad ACCUMULATION across micro-backwards
+                        unit.grad_shard.add_(target)
+                    else:
+                        unit.grad_shard = target
+                gf.record_stream(self._rs_stream)
+                target.record_stream(self._rs_stream)
+                if accumulate:
+                    unit.grad_shard.record_stream(self._rs_stream)
+            else:
+                dist.all_reduce(gf, group=self.rs_pg)
+                gf.div_(self.world_size)
+                piece = gf[unit.shard_off : unit.shard_off + unit.shard_numel]
+                if accumulate:
+                    unit.grad_shard.add_(piece)
+                else:
+                    unit.grad_shard = piece.clone()
+            unit.grad_full = None
+            # reshard params after backward
+            self._reshard(unit)
 
     def finish_grad_sync(self):
         """Wait for all grad reduce-scatters (call after loss.backward())."""
