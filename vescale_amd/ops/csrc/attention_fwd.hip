@@ -1,0 +1,337 @@
+// Flash-attention FORWARD — CDNA4 MFMA (32x32x16), causal, GQA, D=128.
+//
+// Structure per /opt/skills/guides/cdna_hip_programming.md App. B (the
+// 8-wave 32x32 ladder) using the probe-verified fragment layouts
+// (tools/fa_fwd_check.py runs the layout probes on every check):
+//   - 4 waves x QBLK=32 q-rows each = 128 q rows per block; 4-wave
+//     blocks (not 8) so TWO independent blocks fit per CU and drift out
+//     of phase — one block's MFMA overlaps the other's softmax VALU
+//     (barriers phase-lock waves WITHIN a block only)
+//   - swapped QK^T: ST = mfma(A=K, B=Q) -> D-layout col = q = lane&31, so
+//     the online softmax (running m, l per q) is LANE-LOCAL up to one
+//     permlane32_swap combining the half-wave kv clusters
+//   - P -> PV B-operand relayout in-register: v_cvt_pk_bf16_f32 pairs +
+//     permlane32_swap (guide T12: one swap fills both half-kv words; no
+//     LDS round trip, no divergent branch)
+//   - V staged in a [4 kv][16 d] SUBTILED layout (stride 72 elems) and the
+//     PV A-operand (V^T) read with ds_read_b64_tr_b16 (guide T10): vector
+//     stores stay coalesced/conflict-free and the transpose is free in HW
+//     (the naive scalar V^T staging was a 16-way bank conflict, ~5.7x)
+//   - K/V tiles DOUBLE-BUFFERED in LDS: tile j+1's global loads issue
+//     before tile j's compute, stores land in the other buffer, ONE
+//     barrier per tile (guide T3/T4 phase overlap, structured form)
+//   - O accumulates in D-layout (col = q), so the online rescale by
+//     exp(m_old - m_new) is a lane-local scalar multiply
+//   - outputs: O [B,H,S,D] bf16 and logsumexp [B,H,S] fp32 (ln), matching
+//     the stock flash convention so EITHER backward (ours or the
+//     library's) can consume it.
+// Capability parity: reference vescale relies on library flash attention;
+// this is the MI355X-native forward for the train step's hot op.
+#include "common.h"
+
+#define FF_D 128
+#define FF_QBLK 32             // q rows per wave
+#define FF_WAVES 4
+#define FF_QTILE (FF_QBLK * FF_WAVES)  // 256 q rows per block
+#define FF_KV 64               // kv tile
+#define FF_THREADS (FF_WAVES * 64)
+#define FF_VSUB 72             // V subtile stride in elems ([4][16] + pad 8)
+#define FF_VSZ (128 * FF_VSUB) // (64/4 kv-subs) * (128/16 d-subs) = 128
+
+typedef float ff_floatx4 __attribute__((ext_vector_type(4)));
+typedef float ff_floatx16 __attribute__((ext_vector_type(16)));
+typedef short ff_shortx4 __attribute__((ext_vector_type(4)));
+typedef short ff_shortx8 __attribute__((ext_vector_type(8)));
+typedef int ff_intx4 __attribute__((ext_vector_type(4)));
+typedef __attribute__((address_space(3))) const unsigned short* ff_lds_p;
+
+// partner-half exchange: returns the value this VGPR holds in lane l^32.
+// permlane32_swap(a,b) returns the post-swap pair: r0 = [a:0-31 | b:0-31],
+// r1 = [a:32-63 | b:32-63] (probe-verified: tools/fa_fwd_check.py — lane 0
+// sees partner in r1, lane 32 sees partner in r0).
+DEV int ff_swap_other(int x, int half) {
+  auto r = __builtin_amdgcn_permlane32_swap(x, x, false, false);
+  return half ? r[0] : r[1];
+}
+
+// K-tile LDS swizzle (guide G4): XOR byte-bits 4-6 slot with row bits 0-2
+DEV int ff_kswz(int byte_off) {
+  int row = byte_off >> 8;  // 256B rows ([64][128] bf16)
+  return byte_off ^ ((row & 7) << 4);
+}
+
+// V subtiled address (elems): kv in [0,64), d in [0,128)
+DEV int ff_vaddr(int kv, int d) {
+  return ((kv >> 2) * 8 + (d >> 4)) * FF_VSUB + (kv & 3) * 16 + (d & 15);
+}
+
+template <int MODE>  // 0=full, 1=no-softmax, 2=no-PV(+relayout), 3=no-ST (ablation timing)
+DEV void fa_fwd_t(const unsigned short* __restrict__ q,
+            const unsigned short* __restrict__ k,
+            const unsigned short* __restrict__ v,
+            unsigned short* __restrict__ out,
+            float* __restrict__ lse,
+            int B, int Hq, int Hkv, int S, float scale) {
+  __shared__ unsigned short lk[2][FF_KV * FF_D];  // K, swizzled
+  __shared__ unsigned short lv[2][FF_VSZ];        // V, subtiled [4][16]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int l31 = lane & 31;
+  const int half = lane >> 5;  // 0 | 1
+  const int g16 = lane >> 4;   // 16-lane group, for tr reads
+
+  const int qt = blockIdx.x;       // q macro-tile (256 rows)
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hkv = h / (Hq / Hkv);
+
+  const int64_t qbase = (((int64_t)b * Hq + h) * S) * FF_D;
+  const int64_t kbase = (((int64_t)b * Hkv + hkv) * S) * FF_D;
+  const int q0 = qt * FF_QTILE + wave * FF_QBLK;   // wave's first q row
+  const int my_q = q0 + l31;                        // this lane's q row
+
+  // ---- Q fragments in registers: B-operand B[j=q][k=d] ----
+  // lane holds Q[my_q][ (half*8 + e) + 16*ks ] for ks = 0..7
+  ff_shortx8 qf[8];
+  {
+    const unsigned short* qg = q + qbase + (int64_t)my_q * FF_D;
+#pragma unroll
+    for (int ks = 0; ks < 8; ++ks)
+      qf[ks] = *reinterpret_cast<const ff_shortx8*>(qg + ks * 16 + half * 8);
+  }
+
+  // ---- online-softmax state (per q = l31; replicated across halves) ----
+  float m_run = -INFINITY;
+  float l_run = 0.f;
+  // O accumulator: 4 d-subtiles (32 d each) in D-layout:
+  //   acc[ds] reg r holds O[d = ds*32 + (r&3)+8*(r>>2)+4*half][q=l31]
+  ff_floatx16 oacc[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) oacc[i] = (ff_floatx16)(0.f);
+
+  // BLOCK-UNIFORM kv loop bound (all 8 waves hit the same barriers):
+  // cover kv < (qt+1)*256, i.e. through the block's last causal tile.
+  const int n_kv = (qt + 1) * (FF_QTILE / FF_KV);
+  const int last_kv = min(n_kv, S / FF_KV);
+
+  const unsigned short* kg = k + kbase;
+  const unsigned short* vg = v + kbase;
+  // staging registers: 4 chunks of 8 elems per thread (64*128 / 256 / 8)
+  ff_shortx8 kreg[4], vreg[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int e = (tid + j * FF_THREADS) * 8;
+    kreg[j] = *reinterpret_cast<const ff_shortx8*>(kg + e);
+    vreg[j] = *reinterpret_cast<const ff_shortx8*>(vg + e);
+  }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int e = (tid + j * FF_THREADS) * 8;
+    *reinterpret_cast<ff_shortx8*>((char*)lk[0] + ff_kswz(e * 2)) = kreg[j];
+    *reinterpret_cast<ff_shortx8*>(lv[0] + ff_vaddr(e >> 7, e & 127)) = vreg[j];
+  }
+  __syncthreads();
+
+  for (int jkv = 0; jkv < last_kv; ++jkv) {
+    const int kv0 = jkv * FF_KV;
+    const int buf = jkv & 1;
+    const bool have_next = (jkv + 1 < last_kv);
+    // issue next tile's global loads BEFORE compute (latency overlap)
+    if (have_next) {
+      const unsigned short* kn = kg + (int64_t)(kv0 + FF_KV) * FF_D;
+      const unsigned short* vn = vg + (int64_t)(kv0 + FF_KV) * FF_D;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int e = (tid + j * FF_THREADS) * 8;
+        kreg[j] = *reinterpret_cast<const ff_shortx8*>(kn + e);
+        vreg[j] = *reinterpret_cast<const ff_shortx8*>(vn + e);
+      }
+    }
+
+    // ---- ST = K . Q^T : 2 kv-subtiles of 32, contraction d (8 ksteps) ----
+    ff_floatx16 st[2];
+#pragma unroll
+    for (int s2 = 0; s2 < 2; ++s2) st[s2] = (ff_floatx16)(MODE == 3 ? 1e-3f : 0.f);
+    if constexpr (MODE != 3)
+#pragma unroll
+    for (int s2 = 0; s2 < 2; ++s2) {
+#pragma unroll
+      for (int ks = 0; ks < 8; ++ks) {
+        // A-frag: K rows (s2*32 + l31), d = ks*16 + half*8 .. +8
+        int byte = ((s2 * 32 + l31) * FF_D + ks * 16 + half * 8) * 2;
+        ff_shortx8 kf = *reinterpret_cast<const ff_shortx8*>(
+            (const char*)lk[buf] + ff_kswz(byte));
+        st[s2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[ks], st[s2], 0, 0, 0);
+      }
+    }
+
+    if constexpr (MODE == 1 || MODE == 3) {
+      // ablation: skip softmax; keep state plausible so loop carries on
+      l_run += 1.f; m_run = 0.f;
+    } else {
+    // ---- causal mask + online softmax (per q = l31) ----
+    // lane holds ST[kv][q=l31] at kv = s2*32 + (r&3)+8*(r>>2)+4*half
+    float pmax = -INFINITY;
+#pragma unroll
+    for (int s2 = 0; s2 < 2; ++s2)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int kv = kv0 + s2 * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+        float x = (kv <= my_q) ? st[s2][r] * scale : -INFINITY;
+        st[s2][r] = x;
+        pmax = fmaxf(pmax, x);
+      }
+    // combine halves: each (q) lives in lanes l31 and l31+32
+    pmax = fmaxf(pmax, __int_as_float(ff_swap_other(__float_as_int(pmax), half)));
+    float m_new = fmaxf(m_run, pmax);
+    bool dead = (m_new == -INFINITY);  // fully-masked tile for this lane
+    float corr = (m_run == -INFINITY) ? 0.f : __expf(m_run - (dead ? 0.f : m_new));
+    if (dead) corr = 1.f;
+    float psum = 0.f;
+#pragma unroll
+    for (int s2 = 0; s2 < 2; ++s2)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float p = (dead || st[s2][r] == -INFINITY) ? 0.f : __expf(st[s2][r] - m_new);
+        st[s2][r] = p;
+        psum += p;
+      }
+    if (dead) m_new = m_run;
+    psum += __int_as_float(ff_swap_other(__float_as_int(psum), half));
+    l_run = l_run * corr + psum;
+    m_run = m_new;
+
+    // rescale O accumulator by corr (lane-local: every reg shares q=l31)
+#pragma unroll
+    for (int ds = 0; ds < 4; ++ds)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) oacc[ds][r] *= corr;
+    }  // end softmax (MODE != 1,3)
+
+    if constexpr (MODE == 2) {
+      // ablation: consume st into oacc so ST isn't dead-code-eliminated
+#pragma unroll
+      for (int r = 0; r < 16; ++r) oacc[0][r] += st[0][r] + st[1][r];
+    } else {
+    // ---- P relayout to PV B-operand via cvt_pk + permlane32_swap ----
+    // Derivation (32x32 D-layout, regs r: kv_local = (r&3)+8*(r>>2)+4*half):
+    //   group g (16 kv): own regs g*8+0..3 -> kv g*16 + (0..3)+4*half ("LO")
+    //                    own regs g*8+4..7 -> kv g*16 + (8..11)+4*half ("HI")
+    //   B-frag wants kv = g*16 + half*8 + e:
+    //     half=0: e0-3 = own LO, e4-7 = partner LO (their +4..7)
+    //     half=1: e0-3 = partner HI (their 8..11), e4-7 = own HI (12..15)
+    ff_shortx8 pb[4];  // 4 ksteps of 16 kv (KV=64)
+#pragma unroll
+    for (int s2 = 0; s2 < 2; ++s2) {
+#pragma unroll
+      for (int g = 0; g < 2; ++g) {
+        unsigned lo0, lo1, hi0, hi1;
+        asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(lo0)
+                     : "v"(st[s2][g * 8 + 0]), "v"(st[s2][g * 8 + 1]));
+        asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(lo1)
+                     : "v"(st[s2][g * 8 + 2]), "v"(st[s2][g * 8 + 3]));
+        asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(hi0)
+                     : "v"(st[s2][g * 8 + 4]), "v"(st[s2][g * 8 + 5]));
+        asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(hi1)
+                     : "v"(st[s2][g * 8 + 6]), "v"(st[s2][g * 8 + 7]));
+        int p0 = ff_swap_other((int)lo0, half);  // partner's LO word0
+        int p1 = ff_swap_other((int)lo1, half);  // partner's LO word1
+        int p2 = ff_swap_other((int)hi0, half);  // partner's HI word0
+        int p3 = ff_swap_other((int)hi1, half);  // partner's HI word1
+        ff_intx4 frag;
+        if (half == 0) {
+          frag = (ff_intx4){(int)lo0, (int)lo1, p0, p1};
+        } else {
+          frag = (ff_intx4){p2, p3, (int)hi0, (int)hi1};
+        }
+        pb[s2 * 2 + g] = *reinterpret_cast<ff_shortx8*>(&frag);
+      }
+    }
+
+    // ---- PV: O[d][q] += V^T . P  (contraction kv, 4 ksteps of 16) ----
+    // A-frag (V^T rows d = ds*32 + l31, kv = ks*16 + half*8 + e) via
+    // ds_read_b64_tr_b16 on the subtiled V: per 16-lane group the HW reads
+    // lds[(l&15) + j*16 + (l>>4)*64]; our per-lane base cancels the
+    // (l>>4)*64 term and selects the right [4][16] subtile; two reads
+    // (+0 / +8*FF_VSUB elems) give the 8 kv elems of the fragment.
+#pragma unroll
+    for (int ds = 0; ds < 4; ++ds) {
+      ff_lds_p base = (ff_lds_p)(lv[buf] +
+          ((half * 16 + ds * 2 + (g16 & 1)) * FF_VSUB - g16 * 64));
+      ff_shortx4 t[4][2];
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        asm volatile("ds_read_b64_tr_b16 %0, %2 offset:%c3\n\t"
+                     "ds_read_b64_tr_b16 %1, %2 offset:%c4"
+                     : "=v"(t[ks][0]), "=v"(t[ks][1])
+                     : "v"(base), "i"(ks * 4 * 8 * FF_VSUB * 2),
+                       "i"((ks * 4 + 1) * 8 * FF_VSUB * 2));
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)"
+                   : "+v"(t[0][0]), "+v"(t[0][1]), "+v"(t[1][0]), "+v"(t[1][1]),
+                     "+v"(t[2][0]), "+v"(t[2][1]), "+v"(t[3][0]), "+v"(t[3][1]));
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        ff_shortx8 vf;
+#pragma unroll
+        for (int x = 0; x < 4; ++x) { vf[x] = t[ks][0][x]; vf[4 + x] = t[ks][1][x]; }
+        oacc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, pb[ks], oacc[ds], 0, 0, 0);
+      }
+    }
+
+    }  // end relayout+PV (MODE != 2)
+
+    // store next tile into the other buffer, then one barrier
+    if (have_next) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int e = (tid + j * FF_THREADS) * 8;
+        *reinterpret_cast<ff_shortx8*>((char*)lk[buf ^ 1] + ff_kswz(e * 2)) = kreg[j];
+        *reinterpret_cast<ff_shortx8*>(lv[buf ^ 1] + ff_vaddr(e >> 7, e & 127)) = vreg[j];
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: O[q][d] = acc / l ; lse = m + ln(l) ----
+  float inv_l = (l_run > 0.f) ? 1.0f / l_run : 0.f;
+  unsigned short* og = out + qbase + (int64_t)my_q * FF_D;
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int d = ds * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+      og[d] = f32_to_bf16(oacc[ds][r] * inv_l);
+    }
+  if (half == 0) {
+    lse[((int64_t)b * Hq + h) * S + my_q] = m_run + __logf(l_run);
+  }
+}
+
+// instantiations: the real kernel + ablation variants for profiling
+extern "C" __global__ void __launch_bounds__(FF_THREADS, 2)
+fa_fwd_bf16(const unsigned short* q, const unsigned short* k,
+            const unsigned short* v, unsigned short* out, float* lse,
+            int B, int Hq, int Hkv, int S, float scale) {
+  fa_fwd_t<0>(q, k, v, out, lse, B, Hq, Hkv, S, scale);
+}
+extern "C" __global__ void __launch_bounds__(FF_THREADS, 2)
+fa_fwd_bf16_ab1(const unsigned short* q, const unsigned short* k,
+                const unsigned short* v, unsigned short* out, float* lse,
+                int B, int Hq, int Hkv, int S, float scale) {
+  fa_fwd_t<1>(q, k, v, out, lse, B, Hq, Hkv, S, scale);
+}
+extern "C" __global__ void __launch_bounds__(FF_THREADS, 2)
+fa_fwd_bf16_ab2(const unsigned short* q, const unsigned short* k,
+                const unsigned short* v, unsigned short* out, float* lse,
+                int B, int Hq, int Hkv, int S, float scale) {
+  fa_fwd_t<2>(q, k, v, out, lse, B, Hq, Hkv, S, scale);
+}
+extern "C" __global__ void __launch_bounds__(FF_THREADS, 2)
+fa_fwd_bf16_ab3(const unsigned short* q, const unsigned short* k,
+                const unsigned short* v, unsigned short* out, float* lse,
+                int B, int Hq, int Hkv, int S, float scale) {
+  fa_fwd_t<3>(q, k, v, out, lse, B, Hq, Hkv, S, scale);
+}

@@ -13,6 +13,7 @@
 #include "philox_random.hip"
 #include "gemm.hip"
 #include "attention_bwd.hip"
+#include "attention_fwd.hip"
 
 #include <vector>
 
@@ -353,6 +354,99 @@ std::vector<at::Tensor> fa_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   return {dq, dk, dv};
 }
 
+// ------------------------------ flash-attention forward --------------------
+extern "C" __global__ void permlane_probe_kernel(int* out) {
+  int x = (int)threadIdx.x;
+  auto r = __builtin_amdgcn_permlane32_swap(x, x, false, false);
+  out[threadIdx.x] = (r[0] << 16) | (r[1] & 0xFFFF);
+}
+
+// mfma_f32_32x32x16_bf16 layout probe: builds A/B frags per the ASSUMED
+// layout (operand row/col = lane&31, k = (lane>>5)*8 + e) and dumps D for
+// three tests; see tools/fa_fwd_check.py for decoding.
+extern "C" __global__ void mfma32_probe_kernel(float* out) {
+  typedef short sx8 __attribute__((ext_vector_type(8)));
+  typedef float fx16 __attribute__((ext_vector_type(16)));
+  int lane = threadIdx.x & 63;
+  int l31 = lane & 31, half = lane >> 5;
+  sx8 af, bf, a1, bj, ak, bk;
+  for (int e = 0; e < 8; ++e) {
+    int k = half * 8 + e;
+    af[e] = (short)f32_to_bf16((float)((k == 0) ? l31 : 0));
+    bf[e] = (short)f32_to_bf16((float)((k == 0) ? 1 : 0));
+    a1[e] = bf[e];
+    bj[e] = (short)f32_to_bf16((float)((k == 0) ? l31 : 0));
+    ak[e] = (short)f32_to_bf16((float)((k == 5) ? 1 : 0));
+    bk[e] = (short)f32_to_bf16((float)((k == 5) ? l31 : 0));
+  }
+  fx16 z = (fx16)(0.f);
+  fx16 d1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, z, 0, 0, 0);
+  fx16 d2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, bj, z, 0, 0, 0);
+  fx16 d3 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ak, bk, z, 0, 0, 0);
+  for (int r = 0; r < 16; ++r) {
+    out[0 * 64 * 16 + lane * 16 + r] = d1[r];
+    out[1 * 64 * 16 + lane * 16 + r] = d2[r];
+    out[2 * 64 * 16 + lane * 16 + r] = d3[r];
+  }
+}
+
+at::Tensor mfma32_probe() {
+  auto out = at::empty({3, 64, 16}, at::TensorOptions().dtype(at::kFloat).device(at::kCUDA));
+  hipLaunchKernelGGL(mfma32_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     out.data_ptr<float>());
+  return out;
+}
+
+at::Tensor permlane_probe() {
+  auto out = at::empty({64}, at::TensorOptions().dtype(at::kInt).device(at::kCUDA));
+  hipLaunchKernelGGL(permlane_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     out.data_ptr<int>());
+  return out;
+}
+
+std::vector<at::Tensor> fa_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                               double scale) {
+  // q: [B,Hq,S,D]; k,v: [B,Hkv,S,D]; causal, D=128, S % 256 == 0
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
+  int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2), D = (int)q.size(3);
+  int Hkv = (int)k.size(1);
+  TORCH_CHECK(D == FF_D, "fa_fwd supports head_dim 128");
+  TORCH_CHECK(S % FF_QTILE == 0, "seq must be a multiple of 256");
+  TORCH_CHECK(Hq % Hkv == 0, "Hq must be a multiple of Hkv");
+  auto out = at::empty_like(q);
+  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  dim3 grid(S / FF_QTILE, Hq, B);
+  hipLaunchKernelGGL(fa_fwd_bf16, grid, dim3(FF_THREADS), 0, cur_stream(),
+                     (const unsigned short*)q.data_ptr(),
+                     (const unsigned short*)k.data_ptr(),
+                     (const unsigned short*)v.data_ptr(),
+                     (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                     B, Hq, Hkv, S, (float)scale);
+  return {out, lse};
+}
+
+std::vector<at::Tensor> fa_fwd_ablate(at::Tensor q, at::Tensor k, at::Tensor v,
+                                      double scale, int64_t mode) {
+  // ablation-timing variants of fa_fwd (NOT numerically meaningful for
+  // mode != 0); see attention_fwd.hip template MODE docs
+  int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2);
+  int Hkv = (int)k.size(1);
+  auto out = at::empty_like(q);
+  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  dim3 grid(S / FF_QTILE, Hq, B);
+  auto kern = mode == 1 ? fa_fwd_bf16_ab1 : mode == 2 ? fa_fwd_bf16_ab2
+              : mode == 3 ? fa_fwd_bf16_ab3 : fa_fwd_bf16;
+  hipLaunchKernelGGL(kern, grid, dim3(FF_THREADS), 0, cur_stream(),
+                     (const unsigned short*)q.data_ptr(),
+                     (const unsigned short*)k.data_ptr(),
+                     (const unsigned short*)v.data_ptr(),
+                     (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                     B, Hq, Hkv, S, (float)scale);
+  return {out, lse};
+}
+
 // ------------------------------ philox random ------------------------------
 ShardDesc make_desc(const std::vector<int64_t>& gshape,
                     const std::vector<int64_t>& lshape,
@@ -455,6 +549,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scale_", &scale_);
   m.def("gemm_tn", &gemm_tn, pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("variant") = 0);
   m.def("fa_bwd", &fa_bwd);
+  m.def("fa_fwd", &fa_fwd);
+  m.def("fa_fwd_ablate", &fa_fwd_ablate);
+  m.def("permlane_probe", &permlane_probe);
+  m.def("mfma32_probe", &mfma32_probe);
   m.def("philox_uniform_", &philox_uniform_);
   m.def("philox_normal_", &philox_normal_);
   m.def("philox_dropout", &philox_dropout);
