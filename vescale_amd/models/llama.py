@@ -19,6 +19,7 @@ import torch.nn.functional as F
 
 from ..ops import (
     build_rope_table,
+    flash_attention_causal,
     fused_cross_entropy,
     rmsnorm,
     rope_qkv,
@@ -106,7 +107,9 @@ class Attention(nn.Module):
         q = q.transpose(1, 2)  # [B, H, S, D]
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)
-        o = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+        # our MFMA flash forward + library backward (ops/functional.py);
+        # falls back to SDPA off-GPU or for unsupported shapes
+        o = flash_attention_causal(q, k, v)
         o = o.transpose(1, 2).reshape(B, S, self.hq * cfg.head_dim)
         return reduce_from_tp(self.wo(o), self.tp)
 

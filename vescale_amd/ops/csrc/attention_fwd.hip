@@ -13,10 +13,14 @@
 //   - P -> PV B-operand relayout in-register: v_cvt_pk_bf16_f32 pairs +
 //     permlane32_swap (guide T12: one swap fills both half-kv words; no
 //     LDS round trip, no divergent branch)
-//   - V staged in a [4 kv][16 d] SUBTILED layout (stride 72 elems) and the
-//     PV A-operand (V^T) read with ds_read_b64_tr_b16 (guide T10): vector
-//     stores stay coalesced/conflict-free and the transpose is free in HW
-//     (the naive scalar V^T staging was a 16-way bank conflict, ~5.7x)
+//   - V staged EXACTLY like K (row-major + XOR swizzle, vector stores)
+//     and the PV A-operand (V^T) read with ds_read_b64_tr_b16 (guide
+//     T10): probe-measured semantics (tools/tr_probe_check.py) are a 4x4
+//     transpose within each aligned 4-lane group — lane l elem j receives
+//     the (l&3)-th element of the 64b read issued by lane (l&~3)+j — so
+//     pointing lanes of a group at 4 consecutive kv rows (same d column
+//     group) yields the transposed fragment for free; the row-XOR swizzle
+//     also spreads the 4 rows across banks (2-way worst case)
 //   - K/V tiles DOUBLE-BUFFERED in LDS: tile j+1's global loads issue
 //     before tile j's compute, stores land in the other buffer, ONE
 //     barrier per tile (guide T3/T4 phase overlap, structured form)
@@ -35,8 +39,6 @@
 #define FF_QTILE (FF_QBLK * FF_WAVES)  // 256 q rows per block
 #define FF_KV 64               // kv tile
 #define FF_THREADS (FF_WAVES * 64)
-#define FF_VSUB 72             // V subtile stride in elems ([4][16] + pad 8)
-#define FF_VSZ (128 * FF_VSUB) // (64/4 kv-subs) * (128/16 d-subs) = 128
 
 typedef float ff_floatx4 __attribute__((ext_vector_type(4)));
 typedef float ff_floatx16 __attribute__((ext_vector_type(16)));
@@ -60,12 +62,7 @@ DEV int ff_kswz(int byte_off) {
   return byte_off ^ ((row & 7) << 4);
 }
 
-// V subtiled address (elems): kv in [0,64), d in [0,128)
-DEV int ff_vaddr(int kv, int d) {
-  return ((kv >> 2) * 8 + (d >> 4)) * FF_VSUB + (kv & 3) * 16 + (d & 15);
-}
-
-template <int MODE>  // 0=full, 1=no-softmax, 2=no-PV(+relayout), 3=no-ST (ablation timing)
+template <int MODE>  // 0=full, 1=no-softmax, 2=no-PV, 3=no-ST, 4=no-gating, 5=exp-domain (bisect)
 DEV void fa_fwd_t(const unsigned short* __restrict__ q,
             const unsigned short* __restrict__ k,
             const unsigned short* __restrict__ v,
@@ -73,7 +70,7 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
             float* __restrict__ lse,
             int B, int Hq, int Hkv, int S, float scale) {
   __shared__ unsigned short lk[2][FF_KV * FF_D];  // K, swizzled
-  __shared__ unsigned short lv[2][FF_VSZ];        // V, subtiled [4][16]
+  __shared__ unsigned short lv[2][FF_KV * FF_D];  // V, swizzled (same as K)
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -91,6 +88,7 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
   const int64_t kbase = (((int64_t)b * Hkv + hkv) * S) * FF_D;
   const int q0 = qt * FF_QTILE + wave * FF_QBLK;   // wave's first q row
   const int my_q = q0 + l31;                        // this lane's q row
+  const float scale2 = (MODE == 5) ? scale : scale * 1.44269504088896340736f;
 
   // ---- Q fragments in registers: B-operand B[j=q][k=d] ----
   // lane holds Q[my_q][ (half*8 + e) + 16*ks ] for ks = 0..7
@@ -130,7 +128,7 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
   for (int j = 0; j < 4; ++j) {
     int e = (tid + j * FF_THREADS) * 8;
     *reinterpret_cast<ff_shortx8*>((char*)lk[0] + ff_kswz(e * 2)) = kreg[j];
-    *reinterpret_cast<ff_shortx8*>(lv[0] + ff_vaddr(e >> 7, e & 127)) = vreg[j];
+    *reinterpret_cast<ff_shortx8*>((char*)lv[0] + ff_kswz(e * 2)) = vreg[j];
   }
   __syncthreads();
 
@@ -138,6 +136,13 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
     const int kv0 = jkv * FF_KV;
     const int buf = jkv & 1;
     const bool have_next = (jkv + 1 < last_kv);
+    // wave-uniform activity: a wave whose whole q-range is above this kv
+    // tile skips ALL compute (it still stages + hits the barrier). Active
+    // tiles always satisfy kv0 <= q0 <= min(my_q), so no lane is ever
+    // fully masked and no -inf/-inf NaN guards are needed.
+    const bool active = (MODE == 4) || (kv0 <= q0 + FF_QBLK - 1);
+    // wave-uniform mask need: only diagonal tiles pay the per-element cmp
+    const bool need_mask = (MODE == 4) || (kv0 + FF_KV - 1 > q0);
     // issue next tile's global loads BEFORE compute (latency overlap)
     if (have_next) {
       const unsigned short* kn = kg + (int64_t)(kv0 + FF_KV) * FF_D;
@@ -150,6 +155,7 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
       }
     }
 
+    if (active) {
     // ---- ST = K . Q^T : 2 kv-subtiles of 32, contraction d (8 ksteps) ----
     ff_floatx16 st[2];
 #pragma unroll
@@ -171,34 +177,52 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
       // ablation: skip softmax; keep state plausible so loop carries on
       l_run += 1.f; m_run = 0.f;
     } else {
-    // ---- causal mask + online softmax (per q = l31) ----
-    // lane holds ST[kv][q=l31] at kv = s2*32 + (r&3)+8*(r>>2)+4*half
+    // ---- causal mask + online softmax, exp2 domain (per q = l31) ----
+    // lane holds ST[kv][q=l31] at kv = s2*32 + (r&3)+8*(r>>2)+4*half.
+    // Work in log2: x2 = s * scale*log2(e); p = exp2(x2 - m2) maps to a
+    // bare v_exp_f32 (no hidden 1/ln2 multiply per element).
     float pmax = -INFINITY;
+    if (need_mask) {
 #pragma unroll
-    for (int s2 = 0; s2 < 2; ++s2)
+      for (int s2 = 0; s2 < 2; ++s2)
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int kv = kv0 + s2 * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
-        float x = (kv <= my_q) ? st[s2][r] * scale : -INFINITY;
-        st[s2][r] = x;
-        pmax = fmaxf(pmax, x);
-      }
+        for (int r = 0; r < 16; ++r) {
+          int kv = kv0 + s2 * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+          float x = (kv <= my_q) ? st[s2][r] * scale2 : -INFINITY;
+          st[s2][r] = x;
+          pmax = fmaxf(pmax, x);
+        }
+    } else {
+#pragma unroll
+      for (int s2 = 0; s2 < 2; ++s2)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float x = st[s2][r] * scale2;
+          st[s2][r] = x;
+          pmax = fmaxf(pmax, x);
+        }
+    }
     // combine halves: each (q) lives in lanes l31 and l31+32
     pmax = fmaxf(pmax, __int_as_float(ff_swap_other(__float_as_int(pmax), half)));
-    float m_new = fmaxf(m_run, pmax);
-    bool dead = (m_new == -INFINITY);  // fully-masked tile for this lane
-    float corr = (m_run == -INFINITY) ? 0.f : __expf(m_run - (dead ? 0.f : m_new));
-    if (dead) corr = 1.f;
+    float m_new = fmaxf(m_run, pmax);  // finite: active tiles have kv0 <= my_q
+    bool dead = false;
+    if constexpr (MODE == 4) dead = (m_new == -INFINITY);
+    float corr;
+    if constexpr (MODE == 5) corr = __expf(m_run - m_new);
+    else corr = __builtin_amdgcn_exp2f(m_run - m_new);  // m_run=-inf -> 0
+    if (dead) { corr = 1.f; m_new = m_run; }
     float psum = 0.f;
 #pragma unroll
     for (int s2 = 0; s2 < 2; ++s2)
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        float p = (dead || st[s2][r] == -INFINITY) ? 0.f : __expf(st[s2][r] - m_new);
+        float p;
+        if (dead) p = 0.f;
+        else if constexpr (MODE == 5) p = (st[s2][r] == -INFINITY) ? 0.f : __expf(st[s2][r] - m_new);
+        else p = __builtin_amdgcn_exp2f(st[s2][r] - m_new);  // -inf -> 0
         st[s2][r] = p;
         psum += p;
       }
-    if (dead) m_new = m_run;
     psum += __int_as_float(ff_swap_other(__float_as_int(psum), half));
     l_run = l_run * corr + psum;
     m_run = m_new;
@@ -251,23 +275,30 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
     }
 
     // ---- PV: O[d][q] += V^T . P  (contraction kv, 4 ksteps of 16) ----
-    // A-frag (V^T rows d = ds*32 + l31, kv = ks*16 + half*8 + e) via
-    // ds_read_b64_tr_b16 on the subtiled V: per 16-lane group the HW reads
-    // lds[(l&15) + j*16 + (l>>4)*64]; our per-lane base cancels the
-    // (l>>4)*64 term and selects the right [4][16] subtile; two reads
-    // (+0 / +8*FF_VSUB elems) give the 8 kv elems of the fragment.
+    // A-frag (V^T rows d = ds*32 + l31, kv = ks*16 + half*8 + e) via the
+    // probe-measured group transpose (tools/tr_probe_check.py):
+    //   OUT[l&15][j] = IN[4j + ((l&15)>>2)][l&3]   per 16-lane group,
+    // so lane y reads V row kvbase + ((y&15)>>2) at byte column
+    // dbase + (y&3)*4 elems, and the HW hands lane l the d = ds*32 + l31
+    // column of rows kvbase..kvbase+3. Two reads (kvbase, kvbase+4) fill
+    // the 8 kv elems; ks walks rows in +16 steps = +4096 bytes, which
+    // commutes with the bits-4..6 row swizzle (row&7 unchanged).
 #pragma unroll
     for (int ds = 0; ds < 4; ++ds) {
-      ff_lds_p base = (ff_lds_p)(lv[buf] +
-          ((half * 16 + ds * 2 + (g16 & 1)) * FF_VSUB - g16 * 64));
+      const int col2 = (ds * 32 + (l31 & 16) + (lane & 3) * 4) * 2;
+      const int row0 = half * 8 + ((lane & 15) >> 2);  // kv row, jj = 0
+      const int a0 = (row0 * 256 + col2) ^ ((row0 & 7) << 4);
+      const int row1 = row0 + 4;                       // jj = 1
+      const int a1 = (row1 * 256 + col2) ^ ((row1 & 7) << 4);
+      ff_lds_p b0 = (ff_lds_p)((const char*)lv[buf] + a0);
+      ff_lds_p b1 = (ff_lds_p)((const char*)lv[buf] + a1);
       ff_shortx4 t[4][2];
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks) {
-        asm volatile("ds_read_b64_tr_b16 %0, %2 offset:%c3\n\t"
-                     "ds_read_b64_tr_b16 %1, %2 offset:%c4"
+        asm volatile("ds_read_b64_tr_b16 %0, %2 offset:%c4\n\t"
+                     "ds_read_b64_tr_b16 %1, %3 offset:%c4"
                      : "=v"(t[ks][0]), "=v"(t[ks][1])
-                     : "v"(base), "i"(ks * 4 * 8 * FF_VSUB * 2),
-                       "i"((ks * 4 + 1) * 8 * FF_VSUB * 2));
+                     : "v"(b0), "v"(b1), "i"(ks * 16 * 256));
       }
       asm volatile("s_waitcnt lgkmcnt(0)"
                    : "+v"(t[0][0]), "+v"(t[0][1]), "+v"(t[1][0]), "+v"(t[1][1]),
@@ -282,6 +313,7 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
     }
 
     }  // end relayout+PV (MODE != 2)
+    }  // end if (active)
 
     // store next tile into the other buffer, then one barrier
     if (have_next) {
@@ -289,13 +321,13 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
       for (int j = 0; j < 4; ++j) {
         int e = (tid + j * FF_THREADS) * 8;
         *reinterpret_cast<ff_shortx8*>((char*)lk[buf ^ 1] + ff_kswz(e * 2)) = kreg[j];
-        *reinterpret_cast<ff_shortx8*>(lv[buf ^ 1] + ff_vaddr(e >> 7, e & 127)) = vreg[j];
+        *reinterpret_cast<ff_shortx8*>((char*)lv[buf ^ 1] + ff_kswz(e * 2)) = vreg[j];
       }
     }
     __syncthreads();
   }
 
-  // ---- epilogue: O[q][d] = acc / l ; lse = m + ln(l) ----
+  // ---- epilogue: O[q][d] = acc / l ; lse = m2*ln2 + ln(l) (m2 log2-dom) ----
   float inv_l = (l_run > 0.f) ? 1.0f / l_run : 0.f;
   unsigned short* og = out + qbase + (int64_t)my_q * FF_D;
 #pragma unroll
@@ -306,7 +338,8 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
       og[d] = f32_to_bf16(oacc[ds][r] * inv_l);
     }
   if (half == 0) {
-    lse[((int64_t)b * Hq + h) * S + my_q] = m_run + __logf(l_run);
+    lse[((int64_t)b * Hq + h) * S + my_q] =
+        m_run * (MODE == 5 ? 1.0f : 0.69314718055994530942f) + __logf(l_run);
   }
 }
 
@@ -334,4 +367,16 @@ fa_fwd_bf16_ab3(const unsigned short* q, const unsigned short* k,
                 const unsigned short* v, unsigned short* out, float* lse,
                 int B, int Hq, int Hkv, int S, float scale) {
   fa_fwd_t<3>(q, k, v, out, lse, B, Hq, Hkv, S, scale);
+}
+extern "C" __global__ void __launch_bounds__(FF_THREADS, 2)
+fa_fwd_bf16_ab4(const unsigned short* q, const unsigned short* k,
+                const unsigned short* v, unsigned short* out, float* lse,
+                int B, int Hq, int Hkv, int S, float scale) {
+  fa_fwd_t<4>(q, k, v, out, lse, B, Hq, Hkv, S, scale);
+}
+extern "C" __global__ void __launch_bounds__(FF_THREADS, 2)
+fa_fwd_bf16_ab5(const unsigned short* q, const unsigned short* k,
+                const unsigned short* v, unsigned short* out, float* lse,
+                int B, int Hq, int Hkv, int S, float scale) {
+  fa_fwd_t<5>(q, k, v, out, lse, B, Hq, Hkv, S, scale);
 }

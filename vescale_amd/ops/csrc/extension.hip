@@ -397,6 +397,56 @@ at::Tensor mfma32_probe() {
   return out;
 }
 
+// tr-read addressing probe for the 4x4-group transpose scheme: stage a
+// pattern (value = kv*128 + d) through the SAME swizzled store the kernel
+// uses, then perform the kernel's exact PV reads; host checks lane l elem
+// e == (ks*16 + half*8 + e)*128 + ds*32 + l31 (tools/tr_probe_check.py).
+extern "C" __global__ void tr_probe_kernel(unsigned short* out) {
+  __shared__ unsigned short lvp[FF_KV * FF_D];
+  int tid = threadIdx.x;
+  for (int i = tid * 8; i < FF_KV * FF_D; i += 64 * 8) {
+    ff_shortx8 vv;
+#pragma unroll
+    for (int x = 0; x < 8; ++x) vv[x] = (short)(i + x);
+    *reinterpret_cast<ff_shortx8*>((char*)lvp + ff_kswz(i * 2)) = vv;
+  }
+  __syncthreads();
+  int lane = tid & 63;
+  int l31 = lane & 31;
+  int half = lane >> 5;
+  for (int ds = 0; ds < 4; ++ds) {
+    const int col2 = (ds * 32 + (l31 & 16) + (lane & 3) * 4) * 2;
+    const int row0 = half * 8 + ((lane & 15) >> 2);
+    const int a0 = (row0 * 256 + col2) ^ ((row0 & 7) << 4);
+    const int row1 = row0 + 4;
+    const int a1 = (row1 * 256 + col2) ^ ((row1 & 7) << 4);
+    ff_lds_p b0 = (ff_lds_p)((const char*)lvp + a0);
+    ff_lds_p b1 = (ff_lds_p)((const char*)lvp + a1);
+    ff_shortx4 t0, t1;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      switch (ks) {
+        case 0: asm volatile("ds_read_b64_tr_b16 %0, %2 offset:0\n\tds_read_b64_tr_b16 %1, %3 offset:0" : "=v"(t0), "=v"(t1) : "v"(b0), "v"(b1)); break;
+        case 1: asm volatile("ds_read_b64_tr_b16 %0, %2 offset:4096\n\tds_read_b64_tr_b16 %1, %3 offset:4096" : "=v"(t0), "=v"(t1) : "v"(b0), "v"(b1)); break;
+        case 2: asm volatile("ds_read_b64_tr_b16 %0, %2 offset:8192\n\tds_read_b64_tr_b16 %1, %3 offset:8192" : "=v"(t0), "=v"(t1) : "v"(b0), "v"(b1)); break;
+        default: asm volatile("ds_read_b64_tr_b16 %0, %2 offset:12288\n\tds_read_b64_tr_b16 %1, %3 offset:12288" : "=v"(t0), "=v"(t1) : "v"(b0), "v"(b1)); break;
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(t0), "+v"(t1));
+      unsigned short* o = out + ((ds * 4 + ks) * 64 + lane) * 8;
+#pragma unroll
+      for (int x = 0; x < 4; ++x) { o[x] = (unsigned short)t0[x]; o[4 + x] = (unsigned short)t1[x]; }
+    }
+  }
+}
+
+at::Tensor tr_probe() {
+  auto out = at::empty({4, 4, 64, 8},
+                       at::TensorOptions().dtype(at::kShort).device(at::kCUDA));
+  hipLaunchKernelGGL(tr_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     (unsigned short*)out.data_ptr());
+  return out;
+}
+
 at::Tensor permlane_probe() {
   auto out = at::empty({64}, at::TensorOptions().dtype(at::kInt).device(at::kCUDA));
   hipLaunchKernelGGL(permlane_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
@@ -437,7 +487,8 @@ std::vector<at::Tensor> fa_fwd_ablate(at::Tensor q, at::Tensor k, at::Tensor v,
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   dim3 grid(S / FF_QTILE, Hq, B);
   auto kern = mode == 1 ? fa_fwd_bf16_ab1 : mode == 2 ? fa_fwd_bf16_ab2
-              : mode == 3 ? fa_fwd_bf16_ab3 : fa_fwd_bf16;
+              : mode == 3 ? fa_fwd_bf16_ab3 : mode == 4 ? fa_fwd_bf16_ab4
+              : mode == 5 ? fa_fwd_bf16_ab5 : fa_fwd_bf16;
   hipLaunchKernelGGL(kern, grid, dim3(FF_THREADS), 0, cur_stream(),
                      (const unsigned short*)q.data_ptr(),
                      (const unsigned short*)k.data_ptr(),
@@ -553,6 +604,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fa_fwd_ablate", &fa_fwd_ablate);
   m.def("permlane_probe", &permlane_probe);
   m.def("mfma32_probe", &mfma32_probe);
+  m.def("tr_probe", &tr_probe);
   m.def("philox_uniform_", &philox_uniform_);
   m.def("philox_normal_", &philox_normal_);
   m.def("philox_dropout", &philox_dropout);

@@ -5,6 +5,8 @@ from __future__ import annotations
 
 from typing import Optional, Tuple
 
+import os
+
 import torch
 
 
@@ -303,12 +305,16 @@ def fused_cross_entropy(
 # bottleneck (~255 TF effective); causal, head_dim 128, GQA.
 # ---------------------------------------------------------------------------
 class _FlashAttention(torch.autograd.Function):
+    """Causal flash attention: OUR MFMA forward kernel (fa_fwd: ~565 TF at
+    the Llama-8B shape vs the stock AOTriton forward's ~477 TF) paired with
+    the library flash backward, fed with our logsumexp (same ln-domain
+    [B,H,S] fp32 convention).  Our own fa_bwd kernel exists (attention_bwd
+    .hip) but is slower than the library's for now — see NOTES_ROUND2.md."""
+
     @staticmethod
     def forward(ctx, q, k, v, scale):
         # q,k,v: [B, H, S, D] contiguous bf16
-        out, lse, *_ = torch.ops.aten._scaled_dot_product_flash_attention(
-            q, k, v, 0.0, True, False, scale=scale
-        )
+        out, lse = _ext().fa_fwd(q, k, v, scale)
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.scale = scale
         return out
@@ -316,8 +322,12 @@ class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
-        dq, dk, dv = _ext().fa_bwd(
-            q, k, v, out.contiguous(), dout, lse.float(), ctx.scale
+        S = q.shape[2]
+        dev = q.device
+        philox = torch.zeros((), device=dev, dtype=torch.int64)
+        dq, dk, dv = torch.ops.aten._scaled_dot_product_flash_attention_backward(
+            dout.contiguous(), q, k, v, out, lse, None, None, S, S,
+            0.0, True, philox, philox, scale=ctx.scale
         )
         return dq, dk, dv, None
 
@@ -328,7 +338,12 @@ def flash_attention_causal(q, k, v):
     import math
 
     scale = 1.0 / math.sqrt(q.shape[-1])
-    if _use_hip(q, k, v) and q.shape[-1] == 128 and q.shape[2] % 64 == 0:
+    if (
+        _use_hip(q, k, v)
+        and q.shape[-1] == 128
+        and q.shape[2] % 256 == 0
+        and os.environ.get("VESCALE_FA", "hip") == "hip"
+    ):
         return _FlashAttention.apply(
             q.contiguous(), k.contiguous(), v.contiguous(), scale
         )
