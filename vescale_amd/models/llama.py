@@ -103,10 +103,8 @@ class Attention(nn.Module):
         cfg = self.cfg
         x = copy_to_tp(x, self.tp)
         qkv = self.wqkv(x)
+        # rope_qkv emits [B, H, S, D] directly (attention's native layout)
         q, k, v = rope_qkv(qkv, rope_table, self.hq, self.hkv, cfg.head_dim)
-        q = q.transpose(1, 2)  # [B, H, S, D]
-        k = k.transpose(1, 2)
-        v = v.transpose(1, 2)
         # our MFMA flash forward + library backward (ops/functional.py);
         # falls back to SDPA off-GPU or for unsupported shapes
         o = flash_attention_causal(q, k, v)

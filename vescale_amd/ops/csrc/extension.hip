@@ -93,9 +93,9 @@ std::vector<at::Tensor> rope_qkv_fwd(at::Tensor qkv, at::Tensor table,
   check_bf16_contig(qkv, "qkv");
   int64_t T = B * S;
   auto opt = qkv.options();
-  auto q = at::empty({B, S, Hq, D}, opt);
-  auto k = at::empty({B, S, Hkv, D}, opt);
-  auto v = at::empty({B, S, Hkv, D}, opt);
+  auto q = at::empty({B, Hq, S, D}, opt);
+  auto k = at::empty({B, Hkv, S, D}, opt);
+  auto v = at::empty({B, Hkv, S, D}, opt);
   int64_t total = T * (Hq + 2 * Hkv) * (D / 2);
   int grid = grid_for(total, 256);
   hipLaunchKernelGGL(rope_qkv_fwd_bf16, dim3(grid), dim3(256), 0,

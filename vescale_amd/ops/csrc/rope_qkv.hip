@@ -1,15 +1,17 @@
 // Fused QKV-split + RoPE — CDNA4.
 //
 // Consumes the packed output of the fused wqkv GEMM [B, S, (Hq+2*Hkv)*D]
-// and emits contiguous q/k/v in [B, S, H, D] layout with RoPE applied to
-// q and k — one read + one write instead of (3 slice-copies + 2 rope
-// passes).  Backward packs dq/dk/dv back (inverse rotation on dq/dk).
+// and emits contiguous q/k/v in [B, H, S, D] layout (the attention
+// kernels' native layout — saves three transpose-copies per layer) with
+// RoPE applied to q and k — one read + one write instead of (3
+// slice-copies + 2 rope passes + 3 transposes).  Backward packs dq/dk/dv
+// back (inverse rotation on dq/dk).
 #include "common.h"
 
 #define BLOCK 256
 
-// forward: qkv [T, (Hq+2Hkv)*D] -> q [T, Hq*D] (roped), k [T, Hkv*D]
-// (roped), v [T, Hkv*D].  T = B*S tokens; table [S, D/2, 2].
+// forward: qkv [T, (Hq+2Hkv)*D] -> q [B,Hq,S,D] (roped), k [B,Hkv,S,D]
+// (roped), v [B,Hkv,S,D].  T = B*S tokens; table [S, D/2, 2].
 extern "C" __global__ void __launch_bounds__(BLOCK)
 rope_qkv_fwd_bf16(const unsigned short* __restrict__ qkv,
                   unsigned short* __restrict__ q,
@@ -31,17 +33,18 @@ rope_qkv_fwd_bf16(const unsigned short* __restrict__ qkv,
     int h = (int)(th % heads_total);
     int64_t tok = th / heads_total;
     int64_t s = tok % seq;
+    int64_t b = tok / seq;
     const unsigned short* src = qkv + tok * row_in + (int64_t)h * D;
     float x0 = bf16_to_f32(src[d]);
     float x1 = bf16_to_f32(src[d + rot]);
     unsigned short* dst;
     bool do_rope = true;
     if (h < Hq) {
-      dst = q + (tok * Hq + h) * D;
+      dst = q + (((b * Hq + h) * seq) + s) * D;
     } else if (h < Hq + Hkv) {
-      dst = k + (tok * Hkv + (h - Hq)) * D;
+      dst = k + (((b * Hkv + (h - Hq)) * seq) + s) * D;
     } else {
-      dst = v + (tok * Hkv + (h - Hq - Hkv)) * D;
+      dst = v + (((b * Hkv + (h - Hq - Hkv)) * seq) + s) * D;
       do_rope = false;
     }
     if (do_rope) {
@@ -77,14 +80,15 @@ rope_qkv_bwd_bf16(const unsigned short* __restrict__ dq,
     int h = (int)(th % heads_total);
     int64_t tok = th / heads_total;
     int64_t s = tok % seq;
+    int64_t b = tok / seq;
     const unsigned short* src;
     bool do_rope = true;
     if (h < Hq) {
-      src = dq + (tok * Hq + h) * D;
+      src = dq + (((b * Hq + h) * seq) + s) * D;
     } else if (h < Hq + Hkv) {
-      src = dk + (tok * Hkv + (h - Hq)) * D;
+      src = dk + (((b * Hkv + (h - Hq)) * seq) + s) * D;
     } else {
-      src = dv + (tok * Hkv + (h - Hq - Hkv)) * D;
+      src = dv + (((b * Hkv + (h - Hq - Hkv)) * seq) + s) * D;
       do_rope = false;
     }
     float x0 = bf16_to_f32(src[d]);

@@ -151,18 +151,29 @@ class _RoPEQKV(torch.autograd.Function):
         kv = n_kv_heads * head_dim
         q = qkv[..., :d].reshape(B, S, n_heads, head_dim)
         k = qkv[..., d : d + kv].reshape(B, S, n_kv_heads, head_dim)
-        v = qkv[..., d + kv :].reshape(B, S, n_kv_heads, head_dim).contiguous()
+        v = qkv[..., d + kv :].reshape(B, S, n_kv_heads, head_dim)
         q = _rope_ref(q, table, pos_offset, False)
         k = _rope_ref(k, table, pos_offset, False)
-        return q, k, v
+        # match the HIP kernel's [B, H, S, D] output layout
+        return (
+            q.permute(0, 2, 1, 3).contiguous(),
+            k.permute(0, 2, 1, 3).contiguous(),
+            v.permute(0, 2, 1, 3).contiguous(),
+        )
 
     @staticmethod
     def backward(ctx, dq, dk, dv):
         (table,) = ctx.saved_tensors
         B, S, Hq, Hkv, D, pos = ctx.dims
         if ctx.hip:
-            dqkv = _ext().rope_qkv_bwd(dq, dk, dv, table, B, S, Hq, Hkv, D, pos)
+            dqkv = _ext().rope_qkv_bwd(
+                dq.contiguous(), dk.contiguous(), dv.contiguous(),
+                table, B, S, Hq, Hkv, D, pos,
+            )
             return dqkv, None, None, None, None, None
+        dq = dq.permute(0, 2, 1, 3)  # [B,H,S,D] -> [B,S,H,D]
+        dk = dk.permute(0, 2, 1, 3)
+        dv = dv.permute(0, 2, 1, 3)
         dqr = _rope_ref(dq, table, pos, True).reshape(B, S, Hq * D)
         dkr = _rope_ref(dk, table, pos, True).reshape(B, S, Hkv * D)
         dvr = dv.reshape(B, S, Hkv * D)
@@ -171,8 +182,8 @@ class _RoPEQKV(torch.autograd.Function):
 
 def rope_qkv(qkv: torch.Tensor, table: torch.Tensor, n_heads: int,
              n_kv_heads: int, head_dim: int, pos_offset: int = 0):
-    """qkv: [B, S, (Hq+2*Hkv)*D] packed -> (q, k, v) in [B, S, H, D], RoPE
-    applied to q and k."""
+    """qkv: [B, S, (Hq+2*Hkv)*D] packed -> (q, k, v) in [B, H, S, D] (the
+    attention kernels' native layout), RoPE applied to q and k."""
     return _RoPEQKV.apply(qkv, table, n_heads, n_kv_heads, head_dim, pos_offset)
 
 
