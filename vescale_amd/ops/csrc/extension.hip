@@ -14,6 +14,7 @@
 #include "gemm.hip"
 #include "attention_bwd.hip"
 #include "attention_fwd.hip"
+#include "attention_bwd2.hip"
 
 #include <vector>
 
@@ -498,6 +499,78 @@ std::vector<at::Tensor> fa_fwd_ablate(at::Tensor q, at::Tensor k, at::Tensor v,
   return {out, lse};
 }
 
+// ------------------------------ flash-attention backward v2 ---------------
+std::vector<at::Tensor> fa_bwd2(at::Tensor q, at::Tensor k, at::Tensor v,
+                                at::Tensor o, at::Tensor dout, at::Tensor lse,
+                                double scale) {
+  // 32x32-MFMA backward (attention_bwd2.hip): preprocess + dq + dv + dk
+  // (+ GQA reduction). q,o,dout: [B,Hq,S,D]; k,v: [B,Hkv,S,D]; lse ln-dom.
+  check_bf16_contig(q, "q");
+  check_bf16_contig(k, "k");
+  check_bf16_contig(v, "v");
+  check_bf16_contig(o, "o");
+  int B = (int)q.size(0), Hq = (int)q.size(1), S = (int)q.size(2), D = (int)q.size(3);
+  int Hkv = (int)k.size(1);
+  TORCH_CHECK(D == FB_D, "fa_bwd2 supports head_dim 128");
+  TORCH_CHECK(S % FB_TILE == 0, "seq must be a multiple of 128");
+  TORCH_CHECK(lse.scalar_type() == at::kFloat);
+  auto lse_c = lse.contiguous();
+  auto dout_c = dout.contiguous();
+  auto delta = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  int64_t rows = (int64_t)B * Hq * S;
+  hipLaunchKernelGGL(fa_bwd_preprocess, dim3(grid_for(rows, 1, 4096)), dim3(256),
+                     0, cur_stream(),
+                     (const unsigned short*)dout_c.data_ptr(),
+                     (const unsigned short*)o.data_ptr(),
+                     delta.data_ptr<float>(), rows);
+  auto dq = at::empty_like(q);
+  dim3 grid(S / FB_TILE, Hq, B);
+  hipLaunchKernelGGL(fa2_dq_bf16, grid, dim3(FB_THREADS), 0, cur_stream(),
+                     (const unsigned short*)q.data_ptr(),
+                     (const unsigned short*)k.data_ptr(),
+                     (const unsigned short*)v.data_ptr(),
+                     (const unsigned short*)dout_c.data_ptr(),
+                     lse_c.data_ptr<float>(), delta.data_ptr<float>(),
+                     (unsigned short*)dq.data_ptr(), B, Hq, Hkv, S,
+                     (float)scale);
+  auto dv_part = at::empty({B, Hq, S, D}, q.options());
+  auto dk_part = at::empty({B, Hq, S, D}, q.options());
+  hipLaunchKernelGGL(fa2_dv_bf16, grid, dim3(FB_THREADS), 0, cur_stream(),
+                     (const unsigned short*)q.data_ptr(),
+                     (const unsigned short*)k.data_ptr(),
+                     (const unsigned short*)dout_c.data_ptr(),
+                     lse_c.data_ptr<float>(),
+                     (unsigned short*)dv_part.data_ptr(), B, Hq, Hkv, S,
+                     (float)scale);
+  hipLaunchKernelGGL(fa2_dk_bf16, grid, dim3(FB_THREADS), 0, cur_stream(),
+                     (const unsigned short*)q.data_ptr(),
+                     (const unsigned short*)k.data_ptr(),
+                     (const unsigned short*)v.data_ptr(),
+                     (const unsigned short*)dout_c.data_ptr(),
+                     lse_c.data_ptr<float>(), delta.data_ptr<float>(),
+                     (unsigned short*)dk_part.data_ptr(), B, Hq, Hkv, S,
+                     (float)scale);
+  at::Tensor dk, dv;
+  if (Hkv == Hq) {
+    dk = dk_part;
+    dv = dv_part;
+  } else {
+    dk = at::empty({B, Hkv, S, D}, q.options());
+    dv = at::empty({B, Hkv, S, D}, q.options());
+    int64_t SD = (int64_t)S * D;
+    int rgrid = grid_for((int64_t)B * Hkv * SD / 8, 256);
+    hipLaunchKernelGGL(fa_bwd_reduce_gqa, dim3(rgrid), dim3(256), 0,
+                       cur_stream(),
+                       (const unsigned short*)dk_part.data_ptr(),
+                       (unsigned short*)dk.data_ptr(), B, Hq, Hkv, SD);
+    hipLaunchKernelGGL(fa_bwd_reduce_gqa, dim3(rgrid), dim3(256), 0,
+                       cur_stream(),
+                       (const unsigned short*)dv_part.data_ptr(),
+                       (unsigned short*)dv.data_ptr(), B, Hq, Hkv, SD);
+  }
+  return {dq, dk, dv};
+}
+
 // ------------------------------ philox random ------------------------------
 ShardDesc make_desc(const std::vector<int64_t>& gshape,
                     const std::vector<int64_t>& lshape,
@@ -600,6 +673,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scale_", &scale_);
   m.def("gemm_tn", &gemm_tn, pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("variant") = 0);
   m.def("fa_bwd", &fa_bwd);
+  m.def("fa_bwd2", &fa_bwd2);
   m.def("fa_fwd", &fa_fwd);
   m.def("fa_fwd_ablate", &fa_fwd_ablate);
   m.def("permlane_probe", &permlane_probe);

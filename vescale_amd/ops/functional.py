@@ -316,11 +316,11 @@ def fused_cross_entropy(
 # bottleneck (~255 TF effective); causal, head_dim 128, GQA.
 # ---------------------------------------------------------------------------
 class _FlashAttention(torch.autograd.Function):
-    """Causal flash attention: OUR MFMA forward kernel (fa_fwd: ~565 TF at
-    the Llama-8B shape vs the stock AOTriton forward's ~477 TF) paired with
-    the library flash backward, fed with our logsumexp (same ln-domain
-    [B,H,S] fp32 convention).  Our own fa_bwd kernel exists (attention_bwd
-    .hip) but is slower than the library's for now — see NOTES_ROUND2.md."""
+    """Causal flash attention, fully on our MFMA kernels: fa_fwd
+    (attention_fwd.hip, ~565 TF at the Llama-8B shape vs the stock AOTriton
+    forward's ~477 TF) + fa_bwd2 (attention_bwd2.hip, ~18.7 ms vs the
+    library backward's ~20.9 ms at that shape).  Set VESCALE_FA=aten to
+    fall back to the torch/AOTriton kernels."""
 
     @staticmethod
     def forward(ctx, q, k, v, scale):
@@ -333,13 +333,7 @@ class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
-        S = q.shape[2]
-        dev = q.device
-        philox = torch.zeros((), device=dev, dtype=torch.int64)
-        dq, dk, dv = torch.ops.aten._scaled_dot_product_flash_attention_backward(
-            dout.contiguous(), q, k, v, out, lse, None, None, S, S,
-            0.0, True, philox, philox, scale=ctx.scale
-        )
+        dq, dk, dv = _ext().fa_bwd2(q, k, v, out, dout, lse, ctx.scale)
         return dq, dk, dv, None
 
 
