@@ -118,22 +118,21 @@ DEV void fb_relayout(const fb_floatx16 st[2], fb_shortx8 pb[4], int half) {
     const int a1 = (row1 * 256 + col2) ^ ((row1 & 7) << 4);                 \
     ff_lds_p b0 = (ff_lds_p)((const char*)(lsrc) + a0);                     \
     ff_lds_p b1 = (ff_lds_p)((const char*)(lsrc) + a1);                     \
-    asm volatile("ds_read_b64_tr_b16 %0, %2 offset:0\n\t"                   \
-                 "ds_read_b64_tr_b16 %1, %3 offset:0"                       \
-                 : "=v"(t[0][0]), "=v"(t[0][1]) : "v"(b0), "v"(b1));        \
-    asm volatile("ds_read_b64_tr_b16 %0, %2 offset:4096\n\t"                \
-                 "ds_read_b64_tr_b16 %1, %3 offset:4096"                    \
-                 : "=v"(t[1][0]), "=v"(t[1][1]) : "v"(b0), "v"(b1));        \
-    asm volatile("ds_read_b64_tr_b16 %0, %2 offset:8192\n\t"                \
-                 "ds_read_b64_tr_b16 %1, %3 offset:8192"                    \
-                 : "=v"(t[2][0]), "=v"(t[2][1]) : "v"(b0), "v"(b1));        \
-    asm volatile("ds_read_b64_tr_b16 %0, %2 offset:12288\n\t"               \
-                 "ds_read_b64_tr_b16 %1, %3 offset:12288"                   \
-                 : "=v"(t[3][0]), "=v"(t[3][1]) : "v"(b0), "v"(b1));        \
-    asm volatile("s_waitcnt lgkmcnt(0)"                                     \
-                 : "+v"(t[0][0]), "+v"(t[0][1]), "+v"(t[1][0]),             \
-                   "+v"(t[1][1]), "+v"(t[2][0]), "+v"(t[2][1]),             \
-                   "+v"(t[3][0]), "+v"(t[3][1]));                           \
+    /* single asm: 8 reads + drain (see attention_fwd.hip note on the   \
+       interposed-copy race; earlyclobber keeps dsts off b0/b1) */         \
+    asm volatile(                                                           \
+        "ds_read_b64_tr_b16 %0, %8 offset:0\n\t"                           \
+        "ds_read_b64_tr_b16 %1, %9 offset:0\n\t"                           \
+        "ds_read_b64_tr_b16 %2, %8 offset:4096\n\t"                        \
+        "ds_read_b64_tr_b16 %3, %9 offset:4096\n\t"                        \
+        "ds_read_b64_tr_b16 %4, %8 offset:8192\n\t"                        \
+        "ds_read_b64_tr_b16 %5, %9 offset:8192\n\t"                        \
+        "ds_read_b64_tr_b16 %6, %8 offset:12288\n\t"                       \
+        "ds_read_b64_tr_b16 %7, %9 offset:12288\n\t"                       \
+        "s_waitcnt lgkmcnt(0)"                                              \
+        : "=&v"(t[0][0]), "=&v"(t[0][1]), "=&v"(t[1][0]), "=&v"(t[1][1]),   \
+          "=&v"(t[2][0]), "=&v"(t[2][1]), "=&v"(t[3][0]), "=&v"(t[3][1])    \
+        : "v"(b0), "v"(b1));                                                \
   }
 
 DEV fb_shortx8 fb_cat(const fb_shortx4 a, const fb_shortx4 b) {
@@ -165,9 +164,11 @@ fa2_dq_bf16(const unsigned short* __restrict__ q,
   const int l31 = lane & 31;
   const int half = lane >> 5;
 
-  const int qt = blockIdx.x;
-  const int h = blockIdx.y;
-  const int b = blockIdx.z;
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  ff_xcd_remap(bx, by, bz);
+  const int qt = bx;
+  const int h = by;
+  const int b = bz;
   const int hkv = h / (Hq / Hkv);
   const int64_t qbase = (((int64_t)b * Hq + h) * S) * FB_D;
   const int64_t kbase = (((int64_t)b * Hkv + hkv) * S) * FB_D;
@@ -291,9 +292,11 @@ fa2_dv_bf16(const unsigned short* __restrict__ q,
   const int l31 = lane & 31;
   const int half = lane >> 5;
 
-  const int kt = blockIdx.x;
-  const int h = blockIdx.y;
-  const int b = blockIdx.z;
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  ff_xcd_remap(bx, by, bz);
+  const int kt = bx;
+  const int h = by;
+  const int b = bz;
   const int hkv = h / (Hq / Hkv);
   const int64_t qbase = (((int64_t)b * Hq + h) * S) * FB_D;
   const int64_t kbase = (((int64_t)b * Hkv + hkv) * S) * FB_D;
@@ -400,9 +403,11 @@ fa2_dk_bf16(const unsigned short* __restrict__ q,
   const int l31 = lane & 31;
   const int half = lane >> 5;
 
-  const int kt = blockIdx.x;
-  const int h = blockIdx.y;
-  const int b = blockIdx.z;
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  ff_xcd_remap(bx, by, bz);
+  const int kt = bx;
+  const int h = by;
+  const int b = bz;
   const int hkv = h / (Hq / Hkv);
   const int64_t qbase = (((int64_t)b * Hq + h) * S) * FB_D;
   const int64_t kbase = (((int64_t)b * Hkv + hkv) * S) * FB_D;

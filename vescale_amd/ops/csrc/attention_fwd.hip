@@ -56,6 +56,22 @@ DEV int ff_swap_other(int x, int half) {
   return half ? r[0] : r[1];
 }
 
+// XCD-aware block remap (guide T1): the 8 XCDs have private L2s and the
+// HW round-robins consecutive workgroups across them; remapping so each
+// XCD owns a CONTIGUOUS chunk of the flat grid keeps all q-tiles of the
+// same (batch, head) — which stream the same K/V — on one XCD's L2.
+// Bijective when nwg % 8 == 0 (else identity).
+DEV void ff_xcd_remap(int& bx, int& by, int& bz) {
+  int gx = gridDim.x, gy = gridDim.y;
+  int nwg = gx * gy * (int)gridDim.z;
+  if ((nwg & 7) != 0) return;
+  int f = bx + gx * (by + gy * bz);
+  f = (f & 7) * (nwg >> 3) + (f >> 3);
+  bx = f % gx; f /= gx;
+  by = f % gy;
+  bz = f / gy;
+}
+
 // K-tile LDS swizzle (guide G4): XOR byte-bits 4-6 slot with row bits 0-2
 DEV int ff_kswz(int byte_off) {
   int row = byte_off >> 8;  // 256B rows ([64][128] bf16)
@@ -79,9 +95,11 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
   const int half = lane >> 5;  // 0 | 1
   const int g16 = lane >> 4;   // 16-lane group, for tr reads
 
-  const int qt = blockIdx.x;       // q macro-tile (256 rows)
-  const int h = blockIdx.y;
-  const int b = blockIdx.z;
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  ff_xcd_remap(bx, by, bz);
+  const int qt = bx;               // q macro-tile
+  const int h = by;
+  const int b = bz;
   const int hkv = h / (Hq / Hkv);
 
   const int64_t qbase = (((int64_t)b * Hq + h) * S) * FF_D;
@@ -293,16 +311,24 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
       ff_lds_p b0 = (ff_lds_p)((const char*)lv[buf] + a0);
       ff_lds_p b1 = (ff_lds_p)((const char*)lv[buf] + a1);
       ff_shortx4 t[4][2];
-#pragma unroll
-      for (int ks = 0; ks < 4; ++ks) {
-        asm volatile("ds_read_b64_tr_b16 %0, %2 offset:%c4\n\t"
-                     "ds_read_b64_tr_b16 %1, %3 offset:%c4"
-                     : "=v"(t[ks][0]), "=v"(t[ks][1])
-                     : "v"(b0), "v"(b1), "i"(ks * 16 * 256));
-      }
-      asm volatile("s_waitcnt lgkmcnt(0)"
-                   : "+v"(t[0][0]), "+v"(t[0][1]), "+v"(t[1][0]), "+v"(t[1][1]),
-                     "+v"(t[2][0]), "+v"(t[2][1]), "+v"(t[3][0]), "+v"(t[3][1]));
+      // ONE asm statement for all 8 reads + the drain: with separate
+      // statements the compiler may interpose copies of t between a read
+      // and the waitcnt, capturing the register BEFORE the LDS data lands
+      // (observed as run-to-run nondeterminism in O). Earlyclobber outputs
+      // keep the async dsts from aliasing the address registers.
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %8 offset:0\n\t"
+          "ds_read_b64_tr_b16 %1, %9 offset:0\n\t"
+          "ds_read_b64_tr_b16 %2, %8 offset:4096\n\t"
+          "ds_read_b64_tr_b16 %3, %9 offset:4096\n\t"
+          "ds_read_b64_tr_b16 %4, %8 offset:8192\n\t"
+          "ds_read_b64_tr_b16 %5, %9 offset:8192\n\t"
+          "ds_read_b64_tr_b16 %6, %8 offset:12288\n\t"
+          "ds_read_b64_tr_b16 %7, %9 offset:12288\n\t"
+          "s_waitcnt lgkmcnt(0)"
+          : "=&v"(t[0][0]), "=&v"(t[0][1]), "=&v"(t[1][0]), "=&v"(t[1][1]),
+            "=&v"(t[2][0]), "=&v"(t[2][1]), "=&v"(t[3][0]), "=&v"(t[3][1])
+          : "v"(b0), "v"(b1));
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks) {
         ff_shortx8 vf;
