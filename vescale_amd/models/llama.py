@@ -20,6 +20,7 @@ import torch.nn.functional as F
 from ..ops import (
     build_rope_table,
     flash_attention_causal,
+    fused_add_rmsnorm,
     fused_cross_entropy,
     rmsnorm,
     rope_qkv,
@@ -135,10 +136,18 @@ class TransformerBlock(nn.Module):
         self.ffn_norm = RMSNorm(cfg.dim, cfg.norm_eps)
         self.ffn = FeedForward(cfg, tp)
 
-    def forward(self, x, rope_table):
-        x = x + self.attn(self.attn_norm(x), rope_table)
-        x = x + self.ffn(self.ffn_norm(x))
-        return x
+    def forward(self, h, delta, rope_table):
+        # residual stream (h) with a PENDING delta: the add is fused into
+        # the next norm's read pass (ops.fused_add_rmsnorm) — saves a full
+        # read+write of the hidden state per residual add
+        y, h = fused_add_rmsnorm(
+            h if delta is None else delta,
+            None if delta is None else h,
+            self.attn_norm.weight, self.attn_norm.eps,
+        )
+        a = self.attn(y, rope_table)
+        y2, h = fused_add_rmsnorm(a, h, self.ffn_norm.weight, self.ffn_norm.eps)
+        return h, self.ffn(y2)
 
 
 class LlamaModel(nn.Module):
@@ -178,12 +187,19 @@ class LlamaModel(nn.Module):
     ):
         h = self.tok_embeddings(tokens)
         tab = self.rope_table
+        delta = None
         for layer in self.layers:
             if self.activation_checkpointing and self.training:
-                h = torch.utils.checkpoint.checkpoint(layer, h, tab, use_reentrant=False)
+                h, delta = torch.utils.checkpoint.checkpoint(
+                    layer, h, delta, tab, use_reentrant=False
+                )
             else:
-                h = layer(h, tab)
-        h = self.norm(h)
+                h, delta = layer(h, delta, tab)
+        h, _ = fused_add_rmsnorm(
+            h if delta is None else delta,
+            None if delta is None else h,
+            self.norm.weight, self.norm.eps,
+        )
         logits = self.output(h)
         if targets is None:
             return logits

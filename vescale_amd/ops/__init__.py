@@ -39,6 +39,7 @@ def require_ext():
 from .functional import (  # noqa: E402,F401
     flash_attention_causal,
     fused_cross_entropy,
+    fused_add_rmsnorm,
     rmsnorm,
     rope_apply,
     rope_qkv,

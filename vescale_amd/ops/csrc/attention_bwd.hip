@@ -36,28 +36,25 @@ extern "C" __global__ void __launch_bounds__(256)
 fa_bwd_preprocess(const unsigned short* __restrict__ dout,
                   const unsigned short* __restrict__ o,
                   float* __restrict__ delta, int64_t rows) {
-  __shared__ float lds[4];
-  int64_t row = blockIdx.x;
-  const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
-  for (; row < rows; row += gridDim.x) {
-    const unsigned short* pd = dout + row * FA_D;
-    const unsigned short* po = o + row * FA_D;
-    // 256 threads x 8 elems = 2048 > 128: use first 16 lanes of wave 0..3
+  // 16 lanes per row (8 elems each), 16 rows per 256-thread block per
+  // iteration; fully coalesced loads, shfl-xor reduce inside each 16-lane
+  // group, no barriers — HBM-roofline for this pure-bandwidth pass.
+  const int sub = threadIdx.x & 15;         // lane within row
+  int64_t row = (int64_t)blockIdx.x * 16 + (threadIdx.x >> 4);
+  const int64_t stride = (int64_t)gridDim.x * 16;
+  for (; row < rows; row += stride) {
+    const unsigned short* pd = dout + row * FA_D + sub * 8;
+    const unsigned short* po = o + row * FA_D + sub * 8;
+    fa_shortx8 a = *reinterpret_cast<const fa_shortx8*>(pd);
+    fa_shortx8 b = *reinterpret_cast<const fa_shortx8*>(po);
     float s = 0.f;
-    int i = threadIdx.x * 8;
-    if (i < FA_D) {
-      fa_shortx8 a = *reinterpret_cast<const fa_shortx8*>(pd + i);
-      fa_shortx8 b = *reinterpret_cast<const fa_shortx8*>(po + i);
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        s += bf16_to_f32((unsigned short)a[j]) * bf16_to_f32((unsigned short)b[j]);
-    }
-    s = wave_reduce_sum(s);
-    if (lane == 0) lds[wave] = s;
-    __syncthreads();
-    if (threadIdx.x == 0) delta[row] = lds[0] + lds[1] + lds[2] + lds[3];
-    __syncthreads();
+    for (int j = 0; j < 8; ++j)
+      s += bf16_to_f32((unsigned short)a[j]) * bf16_to_f32((unsigned short)b[j]);
+#pragma unroll
+    for (int off = 8; off >= 1; off >>= 1)
+      s += __shfl_xor(s, off, 64);
+    if (sub == 0) delta[row] = s;
   }
 }
 

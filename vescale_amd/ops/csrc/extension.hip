@@ -49,7 +49,8 @@ std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
 }
 
 std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
-                                    at::Tensor rrms) {
+                                    at::Tensor rrms,
+                                    c10::optional<at::Tensor> dres) {
   check_bf16_contig(dy, "dy");
   check_bf16_contig(x, "x");
   int hidden = (int)x.size(-1);
@@ -58,13 +59,48 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
   auto dx = at::empty_like(x);
   auto dw = at::zeros({hidden}, x.options().dtype(at::kFloat));
   int grid = (int)std::min<int64_t>(rows, 1024);
+  const unsigned short* dres_p = nullptr;
+  at::Tensor dres_c;
+  if (dres.has_value()) {
+    dres_c = dres->contiguous();
+    dres_p = (const unsigned short*)dres_c.data_ptr();
+  }
   hipLaunchKernelGGL(rmsnorm_bwd_bf16, dim3(grid), dim3(256), 0, cur_stream(),
                      (const unsigned short*)dy.data_ptr(),
                      (const unsigned short*)x.data_ptr(),
                      (const unsigned short*)w.data_ptr(),
-                     rrms.data_ptr<float>(), (unsigned short*)dx.data_ptr(),
+                     rrms.data_ptr<float>(), dres_p,
+                     (unsigned short*)dx.data_ptr(),
                      dw.data_ptr<float>(), rows, hidden);
   return {dx, dw};
+}
+
+std::vector<at::Tensor> rmsnorm_res_fwd(at::Tensor x,
+                                        c10::optional<at::Tensor> res,
+                                        at::Tensor w, double eps) {
+  // fused residual add + rmsnorm: returns {norm_out, res_new, rrms}
+  check_bf16_contig(x, "x");
+  check_bf16_contig(w, "w");
+  int hidden = (int)x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  int64_t rows = x.numel() / hidden;
+  auto out = at::empty_like(x);
+  auto res_out = at::empty_like(x);
+  auto rrms = at::empty({rows}, x.options().dtype(at::kFloat));
+  const unsigned short* res_p = nullptr;
+  at::Tensor res_c;
+  if (res.has_value()) {
+    res_c = res->contiguous();
+    res_p = (const unsigned short*)res_c.data_ptr();
+  }
+  int grid = (int)std::min<int64_t>(rows, 2048);
+  hipLaunchKernelGGL(rmsnorm_res_fwd_bf16, dim3(grid), dim3(256), 0,
+                     cur_stream(), (const unsigned short*)x.data_ptr(), res_p,
+                     (const unsigned short*)w.data_ptr(),
+                     (unsigned short*)out.data_ptr(),
+                     (unsigned short*)res_out.data_ptr(),
+                     rrms.data_ptr<float>(), rows, hidden, (float)eps);
+  return {out, res_out, rrms};
 }
 
 // ------------------------------ RoPE ------------------------------
@@ -658,7 +694,10 @@ std::vector<at::Tensor> philox_dropout(at::Tensor x, std::vector<int64_t> gshape
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd);
-  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, pybind11::arg("dy"), pybind11::arg("x"),
+        pybind11::arg("w"), pybind11::arg("rrms"),
+        pybind11::arg("dres") = pybind11::none());
+  m.def("rmsnorm_res_fwd", &rmsnorm_res_fwd);
   m.def("rope", &rope);
   m.def("rope_qkv_fwd", &rope_qkv_fwd);
   m.def("rope_qkv_bwd", &rope_qkv_bwd);

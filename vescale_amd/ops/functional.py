@@ -125,6 +125,60 @@ def _rope_ref(x, table, pos_offset, backward):
     return torch.cat([o0, o1], dim=-1).to(x.dtype)
 
 
+class _FusedAddRMSNorm(torch.autograd.Function):
+    """Residual-stream add fused into RMSNorm: (x, res) -> (norm(x+res),
+    x+res).  res may be None (start of the stream).  Backward fuses the
+    residual grad accumulation into the norm backward (kernel dres path).
+    The grad w.r.t. x and res is the same tensor (add distributes)."""
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, res, weight: torch.Tensor, eps: float):
+        ctx.has_res = res is not None
+        if _use_hip(x, weight) and (res is None or res.dtype == x.dtype):
+            out, res_new, rrms = _ext().rmsnorm_res_fwd(
+                x.contiguous(), res.contiguous() if res is not None else None,
+                weight.contiguous(), eps,
+            )
+            ctx.save_for_backward(res_new, weight, rrms)
+            ctx.hip = True
+            return out, res_new
+        res_new = x if res is None else (x + res)
+        xf = res_new.float()
+        rrms = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+        out = (xf * rrms * weight.float()).to(x.dtype)
+        ctx.save_for_backward(res_new, weight, rrms.squeeze(-1))
+        ctx.hip = False
+        return out, res_new
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor, dres_new):
+        res_new, w, rrms = ctx.saved_tensors
+        if ctx.hip:
+            dx, dw = _ext().rmsnorm_bwd(
+                dy.contiguous(), res_new, w, rrms,
+                dres_new.contiguous() if dres_new is not None else None,
+            )
+            dw = dw.to(w.dtype)
+        else:
+            xf = res_new.float()
+            dyf = dy.float()
+            r = rrms.unsqueeze(-1)
+            wf = w.float()
+            h = res_new.shape[-1]
+            dot = (dyf * wf * xf).sum(-1, keepdim=True)
+            dx = (r * wf * dyf - xf * (dot * r.pow(3) / h)).to(res_new.dtype)
+            dw = (dyf * xf * r).reshape(-1, h).sum(0).to(w.dtype)
+            if dres_new is not None:
+                dx = dx + dres_new
+        dres = dx if ctx.has_res else None
+        return dx, dres, dw, None
+
+
+def fused_add_rmsnorm(x, res, weight, eps):
+    """(norm(x + res), x + res); res=None starts the residual stream."""
+    return _FusedAddRMSNorm.apply(x, res, weight, eps)
+
+
 def rope_apply(x: torch.Tensor, table: torch.Tensor, pos_offset: int = 0) -> torch.Tensor:
     """x: [B, S, H, D]."""
     return _RoPE.apply(x, table, pos_offset)
