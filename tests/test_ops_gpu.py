@@ -257,6 +257,35 @@ def test_gemm_tn_mfma():
 
 
 @requires_gpu
+def test_fused_add_rmsnorm():
+    """Fused residual-add + rmsnorm (fwd + dres-fused bwd) vs fp32 ref."""
+    import vescale_amd.ops as ops
+    from vescale_amd.ops.functional import fused_add_rmsnorm
+
+    ops.require_ext()
+    torch.manual_seed(5)
+    x = torch.randn(64, 4096, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    r = torch.randn(64, 4096, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(4096, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y, res_new = fused_add_rmsnorm(x, r, w, 1e-5)
+    dy = torch.randn_like(y)
+    dr2 = torch.randn_like(y)
+    (y * dy + res_new * dr2).sum().backward()
+
+    xf = x.detach().float().requires_grad_()
+    rf = r.detach().float().requires_grad_()
+    wf = w.detach().float().requires_grad_()
+    rn = xf + rf
+    yf = rn * torch.rsqrt(rn.pow(2).mean(-1, keepdim=True) + 1e-5) * wf
+    (yf * dy.float() + rn * dr2.float()).sum().backward()
+    assert torch.allclose(y.float(), yf.float(), atol=5e-2, rtol=5e-2)
+    assert torch.allclose(res_new.float(), rn.float(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(x.grad.float(), xf.grad, atol=8e-2, rtol=8e-2)
+    assert torch.allclose(r.grad.float(), rf.grad, atol=8e-2, rtol=8e-2)
+    assert torch.allclose(w.grad.float(), wf.grad, atol=2.0, rtol=5e-2)
+
+
+@requires_gpu
 def test_flash_attention_fwd_kernel():
     """fa_fwd MFMA forward kernel (attention_fwd.hip): O and logsumexp vs
     fp32 reference (causal, GQA, D=128). The kernel's layout assumptions
