@@ -122,7 +122,9 @@ def main():
         dist.barrier()
     t1 = time.perf_counter()
 
-    elapsed = torch.tensor([t1 - t0], dtype=torch.float64)
+    # MAX over ranks (nccl backend: reduce on-device — a CPU tensor would
+    # raise under RCCL)
+    elapsed = torch.tensor([t1 - t0], dtype=torch.float64, device=device)
     if world_size > 1:
         dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
     sec = float(elapsed.item())
