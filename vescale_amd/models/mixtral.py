@@ -66,6 +66,12 @@ class MoELayer(nn.Module):
         self.dispatcher = None
         self.local_expert_ids: List[int] = []
 
+    def make_expert(self) -> nn.Module:
+        """Fresh expert on this layer's device (dynamic re-placement)."""
+        dev = self.router.weight.device
+        dt = self.router.weight.dtype
+        return Expert(self.cfg).to(device=dev, dtype=dt)
+
     def enable_expert_parallel(self, group, allocator, dispatcher, local_ids):
         self._ep = True
         self.ep_group = group

@@ -37,19 +37,26 @@ class MoELayerParamBuffer:
         self.grad_buffer = torch.zeros(total, dtype=torch.float32, device=dev)
         off = 0
         self._views = []
+        self._view_by_param: Dict[int, torch.Tensor] = {}
         for p in self.params:
             n = p.numel()
             view = self.grad_buffer.narrow(0, off, n).view(p.shape)
             p.main_grad = view
             self._views.append((p, view))
-            p.register_post_accumulate_grad_hook(self._make_hook(p, view))
+            self._view_by_param[id(p)] = view
+            p.register_post_accumulate_grad_hook(self._make_hook(p))
             off += n
         self._work = None
 
-    def _make_hook(self, p, view):
+    def _make_hook(self, p):
+        # resolve the view at CALL time: rebuild() replaces the buffer and
+        # views after dynamic expert re-placement, and a hook that captured
+        # the old view would accumulate into freed storage
         def hook(param):
             if param.grad is not None:
-                view.add_(param.grad.float())
+                view = self._view_by_param.get(id(param))
+                if view is not None:
+                    view.add_(param.grad.float())
                 param.grad = None
 
         return hook
@@ -77,6 +84,32 @@ class MoELayerParamBuffer:
     def zero_grad(self):
         self.grad_buffer.zero_()
 
+    def rebuild(self, layer: nn.Module):
+        """Re-derive the flat buffer after dynamic expert re-placement
+        (moe.api.rebalance_experts): moved-away params vanish with their
+        modules; newly-arrived params get views + hooks.  Params that
+        stayed keep their hook (re-registering would double-accumulate),
+        so hooks are tracked by id."""
+        hooked = set(self._view_by_param.keys())
+        self.params = [
+            p for n, p in layer.named_parameters() if "experts" in n
+        ] or list(layer.parameters())
+        total = sum(p.numel() for p in self.params)
+        dev = self.params[0].device if self.params else torch.device("cpu")
+        self.grad_buffer = torch.zeros(total, dtype=torch.float32, device=dev)
+        off = 0
+        self._views = []
+        self._view_by_param = {}
+        for p in self.params:
+            n = p.numel()
+            view = self.grad_buffer.narrow(0, off, n).view(p.shape)
+            p.main_grad = view
+            self._views.append((p, view))
+            self._view_by_param[id(p)] = view
+            if id(p) not in hooked:
+                p.register_post_accumulate_grad_hook(self._make_hook(p))
+            off += n
+
 
 class MoEParamBuffer:
     """All MoE layers' buffers (parity: MoEParamBuffer :405)."""
@@ -98,3 +131,9 @@ class MoEParamBuffer:
     def zero_grad(self):
         for b in self.layer_buffers.values():
             b.zero_grad()
+
+    def rebuild(self, model: nn.Module, layer_cls: str = "MoELayer"):
+        """Re-derive all layer buffers after dynamic expert re-placement."""
+        for name, mod in model.named_modules():
+            if type(mod).__name__ == layer_cls and name in self.layer_buffers:
+                self.layer_buffers[name].rebuild(mod)

@@ -53,3 +53,99 @@ class _RemoteExpert(nn.Module):
 
     def forward(self, x):  # pragma: no cover
         raise RuntimeError("expert not resident on this rank")
+
+
+# ---------------------------------------------------------------------------
+# dynamic expert re-placement
+# ---------------------------------------------------------------------------
+def rebalance_experts(
+    layer: nn.Module,
+    token_counts,
+    optimizer: Optional[torch.optim.Optimizer] = None,
+) -> bool:
+    """Dynamic expert re-placement (reference experts_allocator dynamic
+    policy, legacy/vescale/moe/experts_allocator.py): all-reduce the
+    observed per-expert token counts over the EP group, ask the allocator
+    for a new placement, and MOVE re-placed experts' parameters (and, when
+    an optimizer is given, their Adam-style state) between ranks with
+    point-to-point sends on the EP group.
+
+    The layer must expose `make_expert()` (a fresh expert module on the
+    layer's device) — MoELayer does.  Every EP rank must call this with
+    its LOCAL counts (list/tensor of length n_experts).  Returns True if
+    the placement changed.  If the model also uses a MoEParamBuffer for a
+    DP dim, call `buffer.rebuild(model)` afterwards.
+    """
+    alloc = layer.allocator
+    group = layer.ep_group
+    if alloc is None:
+        return False
+    dist_up = dist.is_initialized() and group is not None
+    rank = dist.get_rank(group) if dist_up else 0
+    counts = torch.as_tensor(token_counts, dtype=torch.float64).clone()
+    if dist_up:
+        dist.all_reduce(counts, group=group)
+    old = [alloc.owner_of(e) for e in range(alloc.n_experts)]
+    if not alloc.update(counts.tolist()):
+        return False
+    new = [alloc.owner_of(e) for e in range(alloc.n_experts)]
+
+    opt_keys = ("exp_avg", "exp_avg_sq", "step")
+    for e in range(alloc.n_experts):
+        if old[e] == new[e]:
+            continue
+        src_g = dist.get_global_rank(group, old[e]) if dist_up else old[e]
+        dst_g = dist.get_global_rank(group, new[e]) if dist_up else new[e]
+        if rank == old[e]:
+            mod = layer.experts[e]
+            for p in mod.parameters():
+                dist.send(p.data.contiguous(), dst=dst_g, group=group)
+                if optimizer is not None:
+                    st = optimizer.state.get(p, {})
+                    flags = torch.tensor(
+                        [1 if k in st else 0 for k in opt_keys], dtype=torch.int64
+                    )
+                    dist.send(flags, dst=dst_g, group=group)
+                    for k in opt_keys:
+                        if k in st:
+                            t = st[k]
+                            t = t if torch.is_tensor(t) else torch.tensor(float(t))
+                            dist.send(
+                                t.to(torch.float32).reshape(-1).contiguous(),
+                                dst=dst_g, group=group,
+                            )
+                    optimizer.state.pop(p, None)
+            if optimizer is not None:
+                plist = optimizer.param_groups[0]["params"]
+                mine = set(id(q) for q in mod.parameters())
+                optimizer.param_groups[0]["params"] = [
+                    q for q in plist if id(q) not in mine
+                ]
+            layer.experts[e] = _RemoteExpert()
+        elif rank == new[e]:
+            mod = layer.make_expert()
+            for p in mod.parameters():
+                buf = torch.empty_like(p.data)
+                dist.recv(buf, src=src_g, group=group)
+                with torch.no_grad():
+                    p.data.copy_(buf)
+                if optimizer is not None:
+                    flags = torch.empty(len(opt_keys), dtype=torch.int64)
+                    dist.recv(flags, src=src_g, group=group)
+                    st = {}
+                    for j, k in enumerate(opt_keys):
+                        if int(flags[j]):
+                            n = p.numel() if k != "step" else 1
+                            t = torch.empty(n, dtype=torch.float32)
+                            dist.recv(t, src=src_g, group=group)
+                            st[k] = (
+                                t.reshape(p.shape).to(p.device)
+                                if k != "step"
+                                else t.reshape(()).clone()
+                            )
+                    optimizer.state[p] = st
+            if optimizer is not None:
+                optimizer.param_groups[0]["params"].extend(mod.parameters())
+            layer.experts[e] = mod
+    layer.local_expert_ids = alloc.experts_of(rank)
+    return True

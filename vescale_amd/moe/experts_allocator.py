@@ -37,3 +37,45 @@ class BasicExpertsAllocator(ExpertsAllocator):
     def owner_of(self, expert_id: int) -> int:
         per = max(1, self.n_experts // self.ep_world)
         return min(expert_id // per, self.ep_world - 1)
+
+
+class LoadBalancedExpertsAllocator(ExpertsAllocator):
+    """DYNAMIC placement (reference's dynamic allocator policy): starts
+    blocked like Basic, and `update(token_counts)` re-places experts with
+    a greedy longest-processing-time bin packing over the observed load,
+    capped at ceil(E/W) experts per rank (bounds per-rank memory).
+    Returns True when the placement changed; the caller moves the params
+    (moe.api.rebalance_experts)."""
+
+    def __init__(self, n_experts: int, ep_world: int):
+        super().__init__(n_experts, ep_world)
+        per = max(1, n_experts // ep_world)
+        self.placement: List[int] = [
+            min(e // per, ep_world - 1) for e in range(n_experts)
+        ]
+
+    def owner_of(self, expert_id: int) -> int:
+        return self.placement[expert_id]
+
+    def update(self, token_counts: Optional[List[int]] = None) -> bool:
+        if not token_counts or self.ep_world <= 1:
+            return False
+        cap = -(-self.n_experts // self.ep_world)  # ceil(E/W)
+        order = sorted(
+            range(self.n_experts), key=lambda e: -float(token_counts[e])
+        )
+        load = [0.0] * self.ep_world
+        slots = [0] * self.ep_world
+        new: List[int] = [0] * self.n_experts
+        for e in order:
+            # least-loaded rank with a free slot; ties -> keep current owner
+            cands = [r for r in range(self.ep_world) if slots[r] < cap]
+            cands.sort(key=lambda r: (load[r], r != self.placement[e]))
+            r = cands[0]
+            new[e] = r
+            load[r] += float(token_counts[e])
+            slots[r] += 1
+        if new == self.placement:
+            return False
+        self.placement = new
+        return True
