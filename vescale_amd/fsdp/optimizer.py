@@ -106,13 +106,29 @@ class FlatAdamW:
         from ..dtensor.dtensor import DTensor
         from ..dtensor.placement_types import RaggedShard, TensorMeta
 
+        import torch.distributed as dist
+
+        from ..dtensor.device_mesh import DeviceMesh
+
         mesh = self.engine.mesh
         assert mesh is not None or self.engine.world_size == 1, (
             "sharded_state_dict needs the FSDP engine built on a DeviceMesh"
         )
+        suffix = ""
         if mesh is not None and mesh.ndim != 1:
-            mesh = None  # nD meshes: fall back to local tensors (TP-aware
-            # keys are round-2 work)
+            # FSDP on ONE dim of an nD mesh (e.g. TP x DP): each orthogonal
+            # slice (TP rank) has its OWN flat state — qualify the keys by
+            # the orthogonal coordinate so concurrent DCP saves don't
+            # collide, and express the FSDP-dim sharding over the 1-D
+            # submesh.  Same-topology reload; cross-TP resharding is the
+            # remaining nD gap (NOTES_ROUND2).
+            coord = mesh.get_coordinate()
+            fd = self.engine.mesh_dim
+            other = [str(c) for d, c in enumerate(coord) if d != fd]
+            suffix = ".mp" + "_".join(other)
+            pg = self.engine.ag_pg
+            ranks = dist.get_process_group_ranks(pg)
+            mesh = DeviceMesh(mesh.device_type, ranks, pg=pg)
         out = {"step": self.step_count}
         for u in self.engine.units:
             st = self.state[u.name]
@@ -124,7 +140,7 @@ class FlatAdamW:
                 placement = RaggedShard((0,), units)
                 tm = TensorMeta(torch.Size((u.flat_numel,)), (1,), t.dtype)
                 spec = DTensorSpec(mesh, (placement,), tm)
-                out[f"{u.name}.{key}"] = DTensor(t, spec, requires_grad=False)
+                out[f"{u.name}.{key}{suffix}"] = DTensor(t, spec, requires_grad=False)
         return out
 
     def load_sharded_state_dict(self, sd):
