@@ -35,7 +35,7 @@ from ..dtensor.device_mesh import DeviceMesh
 from ..dtensor.dtensor import DTensor
 from ..dtensor.placement_types import RaggedShard, TensorMeta
 from ..dtensor._dtensor_spec import DTensorSpec
-from ..ndtimeline import ndtimeit, predefined as ndm
+from ..ndtimeline import ndtimeit, ndtimeit_stream as nd_stream, predefined as ndm
 
 logger = logging.getLogger(__name__)
 
@@ -313,14 +313,23 @@ class FSDP(nn.Module):
     def _unshard(self, unit: FSDPUnit, async_on_stream: bool = True):
         if unit._is_unsharded or self.world_size == 1:
             return
-        _t = ndtimeit(ndm.UNSHARD_AG); _t.__enter__()
+        use_stream = self._on_gpu and async_on_stream
+        # host-side span only on the synchronous path; the async path
+        # records its span on the comm stream below (no double count)
+        _t = ndtimeit(ndm.UNSHARD_AG) if not use_stream else None
+        if _t: _t.__enter__()
         unit.full = torch.empty(
             unit.flat_numel, dtype=unit.param_dtype, device=self.device
         )
         if self._on_gpu and async_on_stream:
             self._ag_stream.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(self._ag_stream):
-                dist.all_gather_into_tensor(unit.full, unit.shard, group=self.ag_pg)
+                # device-side span recorded ON the comm stream (ndtimeline
+                # comm-stream events — reference patch #4 equivalent)
+                with nd_stream(ndm.UNSHARD_AG, self._ag_stream):
+                    dist.all_gather_into_tensor(
+                        unit.full, unit.shard, group=self.ag_pg
+                    )
                 unit._ag_event = torch.cuda.Event()
                 unit._ag_event.record(self._ag_stream)
             unit.full.record_stream(self._ag_stream)
@@ -329,7 +338,7 @@ class FSDP(nn.Module):
             unit._ag_event = None
         unit._attach_param_views()
         unit._is_unsharded = True
-        _t.__exit__(None, None, None)
+        if _t: _t.__exit__(None, None, None)
 
     def _wait_unshard(self, unit: FSDPUnit):
         if unit._ag_event is not None:
@@ -411,6 +420,8 @@ class FSDP(nn.Module):
 
     def _finish_unit_grads(self, unit: FSDPUnit):
         # all grads of this unit accumulated into unit.grad_full
+        # (host-side GRAD_RS span kept: it measures issue+accumulate; the
+        # collective itself is additionally timed on the rs stream)
         with ndtimeit(ndm.GRAD_RS):
             for p in unit.params:
                 p.grad = None
@@ -428,7 +439,10 @@ class FSDP(nn.Module):
             if self._on_gpu:
                 self._rs_stream.wait_stream(torch.cuda.current_stream())
                 with torch.cuda.stream(self._rs_stream):
-                    dist.reduce_scatter_tensor(target, gf, op=op, group=self.rs_pg)
+                    with nd_stream(ndm.GRAD_RS, self._rs_stream):
+                        dist.reduce_scatter_tensor(
+                            target, gf, op=op, group=self.rs_pg
+                        )
                     if accumulate:
                         # grad ACCUMULATION across micro-backwards
                         unit.grad_shard.add_(target)

@@ -5,6 +5,7 @@ from .timer import (
     NDMetricLevel,
     NDTimerManager,
     ndtimeit,
+    ndtimeit_stream,
     ndtimeit_p2p,
     ndtimer,
 )

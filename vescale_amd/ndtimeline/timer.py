@@ -228,6 +228,31 @@ class ndtimeit:
         return False
 
 
+class ndtimeit_stream(ndtimeit):
+    """Stream-aware variant: records the start/stop HIP events on a GIVEN
+    stream (e.g. the FSDP all-gather/reduce-scatter comm streams), so the
+    span measures the collective's device-side occupancy on its own stream
+    rather than issue-side time on the compute stream (closes the
+    reference's #4 comm-stream-accessor patch at engine level)."""
+
+    def __init__(self, metric: str, stream):
+        super().__init__(metric)
+        self.stream = stream
+
+    def __enter__(self):
+        mgr = NDTimerManager.current()
+        self.t = mgr.timer(self.metric) if mgr and mgr.enabled else None
+        if self.t:
+            self.t.start(stream=self.stream)
+        return self
+
+    def __exit__(self, *exc):
+        if self.t:
+            mgr = NDTimerManager.current()
+            self.t.stop(stream=self.stream, step=mgr.step if mgr else None)
+        return False
+
+
 def ndtimeit_p2p(metric: str, peer: int):
     """P2P-op timing with peer annotation (reference timer.py:730)."""
 
