@@ -35,7 +35,7 @@
 
 #define FB_D 128
 #define FB_OWN 32             // rows owned per wave (q for dq; kv for dv/dk)
-#define FB_WAVES 4
+#define FB_WAVES 8
 #define FB_TILE (FB_OWN * FB_WAVES)  // 128 own-rows per block
 #define FB_OTH 64             // other-side tile (kv for dq; q for dv/dk)
 #define FB_THREADS (FB_WAVES * 64)
@@ -46,12 +46,12 @@ typedef short fb_shortx4 __attribute__((ext_vector_type(4)));
 typedef short fb_shortx8 __attribute__((ext_vector_type(8)));
 typedef int fb_intx4 __attribute__((ext_vector_type(4)));
 
-// stage a 64x128 bf16 tile row-major + ff_kswz swizzle (block-cooperative,
-// 256 threads x 4 chunks of 8)
+// stage a 64x128 bf16 tile row-major + ff_kswz swizzle (block-cooperative)
+#define FB_CHUNKS (FB_OTH * FB_D / (FB_THREADS * 8))
 DEV void fb_stage64(const unsigned short* __restrict__ g,
                     unsigned short* l, int tid) {
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
+  for (int j = 0; j < FB_CHUNKS; ++j) {
     int e = (tid + j * FB_THREADS) * 8;
     fb_shortx8 v = *reinterpret_cast<const fb_shortx8*>(g + e);
     *reinterpret_cast<fb_shortx8*>((char*)l + ff_kswz(e * 2)) = v;
@@ -189,11 +189,11 @@ fa2_dq_bf16(const unsigned short* __restrict__ q,
     for (int ks = 0; ks < 8; ++ks)
       qf[ks] = *reinterpret_cast<const fb_shortx8*>(qg + ks * 16 + half * 8);
   }
-  // stage the block's 128 dO rows once (swizzled; 8 chunks at 256 threads)
+  // stage the block's FB_TILE dO rows once (swizzled)
   {
     const unsigned short* dg = dout + qbase + (int64_t)qt * FB_TILE * FB_D;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
+    for (int j = 0; j < FB_TILE * FB_D / (FB_THREADS * 8); ++j) {
       int e = (tid + j * FB_THREADS) * 8;
       fb_shortx8 vv = *reinterpret_cast<const fb_shortx8*>(dg + e);
       *reinterpret_cast<fb_shortx8*>((char*)ldo + ff_kswz(e * 2)) = vv;

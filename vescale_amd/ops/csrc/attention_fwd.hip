@@ -134,16 +134,17 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
 
   const unsigned short* kg = k + kbase;
   const unsigned short* vg = v + kbase;
-  // staging registers: 4 chunks of 8 elems per thread (64*128 / 256 / 8)
-  ff_shortx8 kreg[4], vreg[4];
+  // staging registers: tile elems / threads / 8 chunks per thread
+#define FF_CHUNKS (FF_KV * FF_D / (FF_THREADS * 8))
+  ff_shortx8 kreg[FF_CHUNKS], vreg[FF_CHUNKS];
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
+  for (int j = 0; j < FF_CHUNKS; ++j) {
     int e = (tid + j * FF_THREADS) * 8;
     kreg[j] = *reinterpret_cast<const ff_shortx8*>(kg + e);
     vreg[j] = *reinterpret_cast<const ff_shortx8*>(vg + e);
   }
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
+  for (int j = 0; j < FF_CHUNKS; ++j) {
     int e = (tid + j * FF_THREADS) * 8;
     *reinterpret_cast<ff_shortx8*>((char*)lk[0] + ff_kswz(e * 2)) = kreg[j];
     *reinterpret_cast<ff_shortx8*>((char*)lv[0] + ff_kswz(e * 2)) = vreg[j];
@@ -166,7 +167,7 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
       const unsigned short* kn = kg + (int64_t)(kv0 + FF_KV) * FF_D;
       const unsigned short* vn = vg + (int64_t)(kv0 + FF_KV) * FF_D;
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
+      for (int j = 0; j < FF_CHUNKS; ++j) {
         int e = (tid + j * FF_THREADS) * 8;
         kreg[j] = *reinterpret_cast<const ff_shortx8*>(kn + e);
         vreg[j] = *reinterpret_cast<const ff_shortx8*>(vn + e);
@@ -344,7 +345,7 @@ DEV void fa_fwd_t(const unsigned short* __restrict__ q,
     // store next tile into the other buffer, then one barrier
     if (have_next) {
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
+      for (int j = 0; j < FF_CHUNKS; ++j) {
         int e = (tid + j * FF_THREADS) * 8;
         *reinterpret_cast<ff_shortx8*>((char*)lk[buf ^ 1] + ff_kswz(e * 2)) = kreg[j];
         *reinterpret_cast<ff_shortx8*>((char*)lv[buf ^ 1] + ff_kswz(e * 2)) = vreg[j];
