@@ -280,3 +280,52 @@ def test_sock_streamer():
         streamer.stop()
         assert len(got) == 2
         assert got[0].metric == "forward-compute" and got[1].rank == 1
+
+
+def test_emulator_chunk_size_model():
+    """RCCL chunk accounting (algo/proto selection + ring loop geometry,
+    reference emulator/calculate_chunk_size.py parity)."""
+    import torch
+
+    from vescale_amd.emulator import (
+        calc_byte_per_step,
+        compute_last_chunk_size,
+        ring_chunk_geometry,
+        run_ring_all_reduce,
+        topo_get_algo_info,
+    )
+
+    # protocol step economics: Simple > LL128 > LL payload per step
+    s = calc_byte_per_step("Simple")
+    assert s == (1 << 22) // 8
+    assert calc_byte_per_step("LL128") < s
+    assert calc_byte_per_step("LL") < calc_byte_per_step("LL128")
+
+    # tuner shape: small->tree/LL, mid->ring/LL128, large->ring/Simple
+    assert topo_get_algo_info(4 << 10, 8)[:2] == ("tree", "LL")
+    assert topo_get_algo_info(256 << 10, 8)[:2] == ("ring", "LL128")
+    assert topo_get_algo_info(64 << 20, 8)[:2] == ("ring", "Simple")
+
+    # geometry covers the buffer exactly, loops of nranks*chunk
+    geo = ring_chunk_geometry(10_000, 4, 4)
+    covered = sum(sz for loop in geo.loops for _, sz in loop)
+    assert covered == 10_000
+    flatsegs = [seg for loop in geo.loops for seg in loop]
+    offs = [o for o, _ in flatsegs]
+    assert offs == sorted(offs)
+    assert compute_last_chunk_size(10_000, 4, geo.chunk_elems) <= geo.chunk_elems
+
+    # chunked ring: deterministic, correct, and a DIFFERENT fp ordering
+    # than the unchunked ring for multi-loop buffers
+    torch.manual_seed(0)
+    bufs = [torch.randn(5000, dtype=torch.float32) for _ in range(4)]
+    ref64 = sum(b.double() for b in bufs)
+    out1 = run_ring_all_reduce([b.clone() for b in bufs], chunk_bytes=1024)
+    out1b = run_ring_all_reduce([b.clone() for b in bufs], chunk_bytes=1024)
+    out2 = run_ring_all_reduce([b.clone() for b in bufs])
+    assert torch.equal(out1[0], out1b[0])           # deterministic
+    assert all(torch.equal(out1[0], o) for o in out1)  # all ranks equal
+    assert torch.allclose(out1[0].double(), ref64, atol=1e-3)
+    assert torch.allclose(out2[0].double(), ref64, atol=1e-3)
+    # orderings genuinely differ between geometries (fp non-associativity)
+    assert not torch.equal(out1[0], out2[0])

@@ -4,6 +4,12 @@ from .all_reduce import (
     run_ring_all_reduce,
     run_tree_all_reduce,
 )
+from .calculate_chunk_size import (
+    calc_byte_per_step,
+    compute_last_chunk_size,
+    ring_chunk_geometry,
+    topo_get_algo_info,
+)
 from .topo import double_binary_trees, ring_order
 from .mesh_collectives import (
     emu_all_gather,
@@ -18,6 +24,10 @@ __all__ = [
     "run_tree_all_reduce",
     "run_direct_all_reduce",
     "ring_order",
+    "topo_get_algo_info",
+    "calc_byte_per_step",
+    "compute_last_chunk_size",
+    "ring_chunk_geometry",
     "double_binary_trees",
     "emu_all_gather",
     "emu_reduce_scatter",

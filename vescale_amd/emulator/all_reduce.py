@@ -17,36 +17,67 @@ import torch
 from .topo import double_binary_trees, ring_order, tree_children
 
 
-def run_ring_all_reduce(buffers: List[torch.Tensor]) -> List[torch.Tensor]:
+def run_ring_all_reduce(
+    buffers: List[torch.Tensor], chunk_bytes: int = 0
+) -> List[torch.Tensor]:
     """Ring allreduce: reduce-scatter pass then all-gather pass.
     buffers[r] is rank r's input; returns the reduced buffers (all equal,
-    each element reduced in ring order starting from its chunk owner)."""
+    each element reduced in ring order starting from its chunk owner).
+
+    chunk_bytes > 0 enables the CHUNKED geometry the real collective uses
+    (emulator/calculate_chunk_size.py): the buffer is processed in loops
+    of nranks * chunk elements and each element's reduction order follows
+    its position WITHIN ITS LOOP — which is what decides the floating
+    point addition order on hardware for buffers larger than one loop."""
     W = len(buffers)
     if W == 1:
         return buffers
     n = buffers[0].numel()
     flats = [b.reshape(-1) for b in buffers]
-    # chunk boundaries (NCCL splits into W chunks, remainder to the front)
-    base = n // W
-    rem = n % W
-    sizes = [base + (1 if i < rem else 0) for i in range(W)]
-    offs = [0]
-    for s in sizes[:-1]:
-        offs.append(offs[-1] + s)
-
-    # reduce-scatter then all-gather, modeled as: chunk c enters the ring at
-    # rank (c+1)%W and accumulates sequentially around it — the exact
-    # element-wise addition order of the ring algorithm
     acc = [f.clone() for f in flats]
-    for c in range(W):
-        o, s = offs[c], sizes[c]
-        # chunk c starts at rank (c+1)%W and travels the ring accumulating
-        cur = flats[(c + 1) % W][o : o + s].clone()
-        for step in range(1, W):
-            r = (c + 1 + step) % W
-            cur = cur + flats[r][o : o + s]
-        for r in range(W):
-            acc[r][o : o + s] = cur
+
+    def loop_segments():
+        if chunk_bytes > 0:
+            eb = buffers[0].element_size()
+            chunk = max(1, chunk_bytes // eb)
+            loop_elems = W * chunk
+            loops = []
+            off = 0
+            while off < n:
+                this_loop = min(loop_elems, n - off)
+                base = this_loop // W
+                rem = this_loop % W
+                segs = []
+                o = off
+                for c in range(W):
+                    s = base + (1 if c < rem else 0)
+                    segs.append((o, s))
+                    o += s
+                loops.append(segs)
+                off += this_loop
+            return loops
+        # unchunked: one loop spanning the buffer, W chunks
+        base = n // W
+        rem = n % W
+        segs = []
+        o = 0
+        for c in range(W):
+            s = base + (1 if c < rem else 0)
+            segs.append((o, s))
+            o += s
+        return [segs]
+
+    for segs in loop_segments():
+        for c, (o, s) in enumerate(segs):
+            if s == 0:
+                continue
+            # chunk c starts at rank (c+1)%W and travels the ring
+            cur = flats[(c + 1) % W][o : o + s].clone()
+            for step in range(1, W):
+                r = (c + 1 + step) % W
+                cur = cur + flats[r][o : o + s]
+            for r in range(W):
+                acc[r][o : o + s] = cur
     return [a.reshape(buffers[0].shape) for a in acc]
 
 
