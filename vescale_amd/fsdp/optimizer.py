@@ -96,12 +96,28 @@ class FlatAdamW:
             for n, t in st.items():
                 self.state[k][n].copy_(t)
 
-    def sharded_state_dict(self):
-        """Optimizer state as DTensors in the flat RaggedShard layout (one
-        1-D global extent per unit buffer, this rank holding its contiguous
-        shard) — DCP-savable and reshardable across world sizes, mirroring
-        the reference's OptimizerStateSpec flat-range checkpointing
-        (optim/distributed_optimizer.py:51)."""
+    def sharded_state_dict(self, per_param: bool = True):
+        """Optimizer state for DCP.
+
+        per_param=True (default): PER-PARAMETER entries addressed as boxes
+        in each parameter's GLOBAL (unsharded) index space
+        (checkpoint/flat_state.py) — reshardable across BOTH data-parallel
+        and tensor-parallel topology changes, mirroring the reference's
+        OptimizerStateSpec flat-range checkpointing
+        (optim/distributed_optimizer.py:51).
+
+        per_param=False: legacy flat RaggedShard layout (one 1-D global
+        extent per unit buffer; DP-reshardable only, TP-qualified keys on
+        nD meshes)."""
+        if per_param:
+            from ..checkpoint.flat_state import build_param_state_entries
+
+            out = {"step": self.step_count}
+            for u in self.engine.units:
+                out.update(
+                    build_param_state_entries(u, self.state[u.name], u.name)
+                )
+            return out
         from ..dtensor._dtensor_spec import DTensorSpec
         from ..dtensor.dtensor import DTensor
         from ..dtensor.placement_types import RaggedShard, TensorMeta

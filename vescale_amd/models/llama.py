@@ -98,6 +98,18 @@ class Attention(nn.Module):
         # fused qkv projection: one GEMM instead of three (TP: local heads)
         self.wqkv = nn.Linear(d, (self.hq + 2 * self.hkv) * hd, bias=False)
         self.wo = nn.Linear(self.hq * hd, d, bias=False)
+        if self.tp.world > 1:
+            # TP metadata for topology-independent (cross-TP reshardable)
+            # checkpointing: (dim, global_size, [(local_start, n, global_start)])
+            w, r = self.tp.world, self.tp.rank
+            q, kv = self.hq * hd, self.hkv * hd
+            self.wqkv.weight._tp_shard = (
+                0, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd,
+                [(0, q, r * q),
+                 (q, kv, cfg.n_heads * hd + r * kv),
+                 (q + kv, kv, (cfg.n_heads + cfg.n_kv_heads) * hd + r * kv)],
+            )
+            self.wo.weight._tp_shard = (1, cfg.n_heads * hd, [(0, q, r * q)])
 
     def forward(self, x: torch.Tensor, rope_table: torch.Tensor) -> torch.Tensor:
         B, S, _ = x.shape
@@ -122,6 +134,13 @@ class FeedForward(nn.Module):
         # w1 (gate) and w3 (up) fused into one GEMM (TP: local ffn slice)
         self.w13 = nn.Linear(cfg.dim, 2 * self.ffn_local, bias=False)
         self.w2 = nn.Linear(self.ffn_local, cfg.dim, bias=False)
+        if self.tp.world > 1:
+            fl, r = self.ffn_local, self.tp.rank
+            self.w13.weight._tp_shard = (
+                0, 2 * cfg.ffn_dim,
+                [(0, fl, r * fl), (fl, fl, cfg.ffn_dim + r * fl)],
+            )
+            self.w2.weight._tp_shard = (1, cfg.ffn_dim, [(0, fl, r * fl)])
 
     def forward(self, x):
         x = copy_to_tp(x, self.tp)
