@@ -342,6 +342,21 @@ def test_flash_attention_train_path():
 
 
 @requires_gpu
+def test_gemm8_tn():
+    """8-phase 256x256 MFMA GEMM (gemm8.hip) vs fp32 reference."""
+    import vescale_amd.ops as ops
+
+    C = ops.require_ext()
+    torch.manual_seed(2)
+    a = torch.randn(512, 256, device="cuda", dtype=torch.bfloat16) / 8
+    b = torch.randn(256, 256, device="cuda", dtype=torch.bfloat16) / 8
+    c = C.gemm_tn8(a, b)
+    ref = a.float() @ b.float().t()
+    rel = ((c.float() - ref).abs() / ref.abs().clamp_min(1e-2)).max().item()
+    assert rel < 0.05, rel
+
+
+@requires_gpu
 def test_flash_attention_bwd_kernel():
     """Direct test of our fa_bwd MFMA kernel (attention_bwd.hip) — kept as
     an alternative backward (slower than the library's today, see

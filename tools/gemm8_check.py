@@ -1,0 +1,39 @@
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import time, torch
+import vescale_amd.ops as ops
+C = ops.require_ext()
+
+def check(M, N, K):
+    torch.manual_seed(1)
+    a = torch.randn(M, K, device="cuda", dtype=torch.bfloat16) / 8
+    b = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) / 8
+    c = C.gemm_tn8(a, b)
+    ref = a.float() @ b.float().t()
+    err = (c.float() - ref).abs()
+    rel = (err / ref.abs().clamp_min(1e-2)).max().item()
+    print(f"{M}x{N}x{K}: max_abs={err.max().item():.4f} rel={rel:.4f}")
+    return rel < 0.05
+
+ok = check(512, 512, 256) and check(512, 256, 128) and check(1024, 512, 4096)
+print("REFCHECK", "OK" if ok else "FAILED")
+if not ok: raise SystemExit(1)
+
+def bench(M, N, K, n=20):
+    a = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+    bt = b.t()
+    def ours(): C.gemm_tn8(a, b)
+    def lib(): torch.matmul(a, bt)
+    for fn, name in ((ours, "gemm8"), (lib, "hipblaslt")):
+        for _ in range(3): fn()
+        torch.cuda.synchronize(); t = time.perf_counter()
+        for _ in range(n): fn()
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t) / n
+        print(f"{M}x{N}x{K} {name}: {dt*1e3:7.3f} ms {2*M*N*K/dt/1e12:6.0f} TF")
+
+bench(4096, 4096, 4096)
+bench(8192, 8192, 8192)
+bench(32768, 6144, 4096)   # llama wqkv
+bench(32768, 4096, 14336)  # llama w2 fwd

@@ -12,6 +12,7 @@
 #include "cross_entropy.hip"
 #include "philox_random.hip"
 #include "gemm.hip"
+#include "gemm8.hip"
 #include "attention_bwd.hip"
 #include "attention_fwd.hip"
 #include "attention_bwd2.hip"
@@ -320,6 +321,23 @@ at::Tensor gemm_tn(at::Tensor a, at::Tensor b, int64_t variant) {
   if (variant == 3) kern = gemm_tn_bf16_v3;
   if (variant == 4) kern = gemm_tn_bf16_v4;
   hipLaunchKernelGGL(kern, dim3(grid), dim3(threads), 0, cur_stream(),
+                     (const unsigned short*)a.data_ptr(),
+                     (const unsigned short*)b.data_ptr(),
+                     (unsigned short*)c.data_ptr(), (int)M, (int)N, (int)K);
+  return c;
+}
+
+at::Tensor gemm_tn8(at::Tensor a, at::Tensor b) {
+  // C[M,N] = A[M,K] . B[N,K]^T via the 8-phase 256x256 template (gemm8.hip)
+  check_bf16_contig(a, "a");
+  check_bf16_contig(b, "b");
+  int64_t M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(b.size(1) == K, "K mismatch");
+  TORCH_CHECK(M % 256 == 0 && N % 256 == 0 && K % 64 == 0,
+              "gemm_tn8 requires M,N % 256 == 0 and K % 64 == 0");
+  auto c = at::empty({M, N}, a.options());
+  dim3 grid((unsigned)(M / 256), (unsigned)(N / 256), 1);
+  hipLaunchKernelGGL(gemm8_tn_bf16, grid, dim3(G8_THREADS), 0, cur_stream(),
                      (const unsigned short*)a.data_ptr(),
                      (const unsigned short*)b.data_ptr(),
                      (unsigned short*)c.data_ptr(), (int)M, (int)N, (int)K);
@@ -711,6 +729,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("l2norm_sq", &l2norm_sq);
   m.def("scale_", &scale_);
   m.def("gemm_tn", &gemm_tn, pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("variant") = 0);
+  m.def("gemm_tn8", &gemm_tn8);
   m.def("fa_bwd", &fa_bwd);
   m.def("fa_bwd2", &fa_bwd2);
   m.def("fa_fwd", &fa_fwd);
