@@ -382,3 +382,28 @@ def _t_explicit_collectives(rank, ws):
 
 def test_explicit_collectives():
     spawn(2, _t_explicit_collectives)
+
+
+def test_rope_qkv_layout_cpu():
+    """rope_qkv emits [B, H, S, D] (the attention kernels' native layout);
+    CPU fallback must match an explicit slice+rope+permute reference."""
+    import math
+
+    import torch
+
+    from vescale_amd.ops.functional import _rope_ref, build_rope_table, rope_qkv
+
+    torch.manual_seed(3)
+    B, S, Hq, Hkv, D = 2, 8, 4, 2, 16
+    qkv = torch.randn(B, S, (Hq + 2 * Hkv) * D)
+    table = build_rope_table(S, D, 10000.0)
+    q, k, v = rope_qkv(qkv, table, Hq, Hkv, D)
+    assert q.shape == (B, Hq, S, D) and k.shape == (B, Hkv, S, D)
+    qr = qkv[..., : Hq * D].reshape(B, S, Hq, D)
+    kr = qkv[..., Hq * D : (Hq + Hkv) * D].reshape(B, S, Hkv, D)
+    vr = qkv[..., (Hq + Hkv) * D :].reshape(B, S, Hkv, D)
+    qe = _rope_ref(qr, table, 0, False).permute(0, 2, 1, 3)
+    ke = _rope_ref(kr, table, 0, False).permute(0, 2, 1, 3)
+    assert torch.allclose(q, qe, atol=1e-6)
+    assert torch.allclose(k, ke, atol=1e-6)
+    assert torch.allclose(v, vr.permute(0, 2, 1, 3), atol=1e-6)
