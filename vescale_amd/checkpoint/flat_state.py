@@ -15,7 +15,7 @@ several segments).  See models/llama.py for the annotations.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Sequence, Tuple
 
 import torch
 
@@ -160,16 +160,14 @@ class FlatParamStateShard(torch.Tensor):
         ]
 
     def __get_tensor_shard__(self, index) -> torch.Tensor:
+        # box flat starts are relative to the PARAM's local tensor;
+        # self._local starts at the first covered element (_flat_base)
         for (off, sz), fs in self._boxes:
             if torch.Size(off) == index.offset:
-                return self._local.narrow(0, fs_rel(self, fs), box_numel((off, sz))).view(sz)
+                return self._local.narrow(
+                    0, fs - self._flat_base, box_numel((off, sz))
+                ).view(sz)
         raise ValueError(f"no state box at {index.offset} for {index.fqn}")
-
-
-def fs_rel(shard: "FlatParamStateShard", fs: int) -> int:
-    """Local flat start is stored relative to the PARAM's local tensor;
-    shard._local starts at the first covered element."""
-    return fs - shard._flat_base
 
 
 def build_param_state_entries(
