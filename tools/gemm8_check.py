@@ -24,8 +24,9 @@ def bench(M, N, K, n=20):
     b = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
     bt = b.t()
     def ours(): C.gemm_tn8(a, b)
+    def lock(): C.gemm_tn8(a, b, 1)
     def lib(): torch.matmul(a, bt)
-    for fn, name in ((ours, "gemm8"), (lib, "hipblaslt")):
+    for fn, name in ((ours, "gemm8"), (lock, "lockstep"), (lib, "hipblaslt")):
         for _ in range(3): fn()
         torch.cuda.synchronize(); t = time.perf_counter()
         for _ in range(n): fn()

@@ -327,7 +327,7 @@ at::Tensor gemm_tn(at::Tensor a, at::Tensor b, int64_t variant) {
   return c;
 }
 
-at::Tensor gemm_tn8(at::Tensor a, at::Tensor b) {
+at::Tensor gemm_tn8(at::Tensor a, at::Tensor b, int64_t mode) {
   // C[M,N] = A[M,K] . B[N,K]^T via the 8-phase 256x256 template (gemm8.hip)
   check_bf16_contig(a, "a");
   check_bf16_contig(b, "b");
@@ -337,7 +337,8 @@ at::Tensor gemm_tn8(at::Tensor a, at::Tensor b) {
               "gemm_tn8 requires M,N % 256 == 0 and K % 64 == 0");
   auto c = at::empty({M, N}, a.options());
   dim3 grid((unsigned)(M / 256), (unsigned)(N / 256), 1);
-  hipLaunchKernelGGL(gemm8_tn_bf16, grid, dim3(G8_THREADS), 0, cur_stream(),
+  auto kern = mode == 1 ? gemm8_tn_bf16_lockstep : gemm8_tn_bf16;
+  hipLaunchKernelGGL(kern, grid, dim3(G8_THREADS), 0, cur_stream(),
                      (const unsigned short*)a.data_ptr(),
                      (const unsigned short*)b.data_ptr(),
                      (unsigned short*)c.data_ptr(), (int)M, (int)N, (int)K);
@@ -729,7 +730,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("l2norm_sq", &l2norm_sq);
   m.def("scale_", &scale_);
   m.def("gemm_tn", &gemm_tn, pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("variant") = 0);
-  m.def("gemm_tn8", &gemm_tn8);
+  m.def("gemm_tn8", &gemm_tn8, pybind11::arg("a"), pybind11::arg("b"), pybind11::arg("mode") = 0);
   m.def("fa_bwd", &fa_bwd);
   m.def("fa_bwd2", &fa_bwd2);
   m.def("fa_fwd", &fa_fwd);

@@ -61,11 +61,12 @@ DEV g8_shortx8 g8_frag(const unsigned short* l, int frag, int kc, int lane) {
   return *reinterpret_cast<const g8_shortx8*>((const char*)l + g8_swz(byte));
 }
 
-extern "C" __global__ void __launch_bounds__(G8_THREADS, 1)
-gemm8_tn_bf16(const unsigned short* __restrict__ A,
-              const unsigned short* __restrict__ B,
-              unsigned short* __restrict__ C,
-              int M, int N, int K) {
+template <int MODE>  // 0 = tile-boundary barriers only; 1 = per-phase
+                     // barrier pairs (the guide's lockstep schedule, m196)
+DEV void g8_body(const unsigned short* __restrict__ A,
+                 const unsigned short* __restrict__ B,
+                 unsigned short* __restrict__ C,
+                 int M, int N, int K) {
   __shared__ unsigned short lA[2][2][128 * G8_BK];  // [dbuf][mhalf]
   __shared__ unsigned short lB[2][2][128 * G8_BK];  // [dbuf][nhalf]
 
@@ -141,7 +142,8 @@ gemm8_tn_bf16(const unsigned short* __restrict__ A,
       // reads, so the compiler inserts MINIMAL counted waits and can
       // software-pipeline reads across the unrolled phases
 
-      // (d) 16 mfmas under raised priority (T5)
+      // (d) 16 mfmas under raised priority (T5); MODE 1 = lockstep
+      if constexpr (MODE == 1) __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < 4; ++i)
@@ -150,6 +152,8 @@ gemm8_tn_bf16(const unsigned short* __restrict__ A,
           acc[mh2 * 4 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[i], bf[j], acc[mh2 * 4 + i][j], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
+
+      if constexpr (MODE == 1) __builtin_amdgcn_s_barrier();
 
       // (e) K-tile boundary: all of kt+1's half-tiles must have LANDED
       // (LDS writes from vm loads become block-visible via drain+barrier)
@@ -175,4 +179,15 @@ gemm8_tn_bf16(const unsigned short* __restrict__ A,
         cg[(int64_t)r * N] = f32_to_bf16(acc[i][j][r]);
     }
   }
+}
+
+extern "C" __global__ void __launch_bounds__(G8_THREADS, 1)
+gemm8_tn_bf16(const unsigned short* A, const unsigned short* B,
+              unsigned short* C, int M, int N, int K) {
+  g8_body<0>(A, B, C, M, N, K);
+}
+extern "C" __global__ void __launch_bounds__(G8_THREADS, 1)
+gemm8_tn_bf16_lockstep(const unsigned short* A, const unsigned short* B,
+                       unsigned short* C, int M, int N, int K) {
+  g8_body<1>(A, B, C, M, N, K);
 }
