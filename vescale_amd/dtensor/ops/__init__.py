@@ -3,7 +3,17 @@ from __future__ import annotations
 
 
 def register_all(dispatcher):
-    from . import attention, embedding, math_ops, matrix, pointwise, random_ops, tensor_ops, view_ops
+    from . import (
+        attention,
+        conv_ops,
+        embedding,
+        math_ops,
+        matrix,
+        pointwise,
+        random_ops,
+        tensor_ops,
+        view_ops,
+    )
 
     pointwise.register(dispatcher)
     matrix.register(dispatcher)
@@ -11,5 +21,6 @@ def register_all(dispatcher):
     math_ops.register(dispatcher)
     tensor_ops.register(dispatcher)
     embedding.register(dispatcher)
+    conv_ops.register(dispatcher)
     random_ops.register(dispatcher)
     attention.register(dispatcher)
