@@ -1,6 +1,9 @@
 """Loss-parity artifact: tiny-llama 30 fixed steps; run on GPU (bf16) and
 CPU (fp32); losses written to JSON for the parity report."""
-import json, sys, torch
+import json, os, sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import torch
 from vescale_amd.fsdp import FSDP, FlatAdamW
 from vescale_amd.models.llama import LlamaModel, llama_tiny
 
