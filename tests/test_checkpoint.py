@@ -291,3 +291,39 @@ def test_optimizer_checkpoint_cross_tp_reshard():
     with tempfile.TemporaryDirectory() as td:
         spawn(4, _t_opt_ckpt_xtp, td, "save")
         spawn(2, _t_opt_ckpt_xtp, td, "load")
+
+
+def test_flat_state_tp_segment_boxes():
+    """Unit test of the TP-segmented global-box mapping (flat_state.py):
+    the packed-QKV layout (3 dim-0 segments) and a dim-1 row-parallel
+    shard both land on the right global coordinates."""
+    from vescale_amd.checkpoint.flat_state import _global_boxes
+
+    # packed qkv: local [4, 6]; rows 0-1 are "q" -> global rows 2-3,
+    # rows 2-3 are "k" -> global rows 6-7 (global dim-0 size 8)
+    p = torch.zeros(4, 6)
+    p._tp_shard = (0, 8, [(0, 2, 2), (2, 2, 6)])
+    boxes = _global_boxes((4, 6), 3, 21, p)
+    assert [(tuple(o), tuple(s), fs) for (o, s), fs in boxes] == [
+        ((2, 3), (1, 3), 3),
+        ((3, 0), (1, 6), 6),
+        ((6, 0), (1, 6), 12),
+        ((7, 0), (1, 3), 18),
+    ]
+
+    # row-parallel: local [4, 3] = cols 3-5 of a global [4, 9]
+    q = torch.zeros(4, 3)
+    q._tp_shard = (1, 9, [(0, 3, 3)])
+    boxes = _global_boxes((4, 3), 1, 8, q)
+    assert [(tuple(o), tuple(s), fs) for (o, s), fs in boxes] == [
+        ((0, 4), (1, 2), 1),
+        ((1, 3), (1, 3), 3),
+        ((2, 3), (1, 2), 6),
+    ]
+
+    # no TP metadata: plain row-major boxes
+    r = torch.zeros(4, 4)
+    boxes = _global_boxes((4, 4), 2, 10, r)
+    assert [(tuple(o), tuple(s)) for (o, s), _ in boxes] == [
+        ((0, 2), (1, 2)), ((1, 0), (1, 4)), ((2, 0), (1, 2)),
+    ]
