@@ -342,6 +342,28 @@ def test_flash_attention_train_path():
 
 
 @requires_gpu
+def test_flash_attention_aten_fallback():
+    """VESCALE_FA=aten must route flash_attention_causal to the library
+    kernels and agree with the HIP path."""
+    import importlib
+    import os
+
+    import vescale_amd.ops.functional as F_
+
+    torch.manual_seed(4)
+    q = torch.randn(1, 4, 256, 128, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(1, 2, 256, 128, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(1, 2, 256, 128, device="cuda", dtype=torch.bfloat16)
+    hip = F_.flash_attention_causal(q, k, v)
+    os.environ["VESCALE_FA"] = "aten"
+    try:
+        lib = F_.flash_attention_causal(q, k, v)
+    finally:
+        os.environ.pop("VESCALE_FA", None)
+    assert torch.allclose(hip.float(), lib.float(), atol=3e-2, rtol=3e-2)
+
+
+@requires_gpu
 def test_gemm8_tn():
     """8-phase 256x256 MFMA GEMM (gemm8.hip) vs fp32 reference."""
     import vescale_amd.ops as ops
