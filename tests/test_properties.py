@@ -98,3 +98,58 @@ def test_ring_chunk_geometry_covers(numel, nranks):
             pos += sz
             covered += sz
     assert covered == numel
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    rows=st.integers(1, 12),
+    cols=st.integers(1, 12),
+    nseg=st.integers(1, 3),
+    data=st.data(),
+)
+def test_global_boxes_roundtrip(rows, cols, nseg, data):
+    """TP-segmented global boxes cover exactly the local flat range, and
+    mapping each box back through the segment table recovers the local
+    coordinates (checkpoint/flat_state.py)."""
+    from vescale_amd.checkpoint.flat_state import _global_boxes
+
+    nseg = min(nseg, rows)
+    # partition local rows into nseg segments with disjoint global rows
+    cuts = sorted(
+        data.draw(
+            st.lists(
+                st.integers(1, rows - 1), min_size=nseg - 1, max_size=nseg - 1,
+                unique=True,
+            )
+        )
+    ) if nseg > 1 else []
+    bounds = [0] + cuts + [rows]
+    segs = []
+    g = 0
+    gdim = 0
+    for i in range(nseg):
+        l0, l1 = bounds[i], bounds[i + 1]
+        gap = data.draw(st.integers(0, 4))
+        g += gap
+        segs.append((l0, l1 - l0, g))
+        g += l1 - l0
+        gdim = g
+    p = torch.zeros(rows, cols)
+    p._tp_shard = (0, gdim, segs)
+    n = rows * cols
+    a = data.draw(st.integers(0, n))
+    b = data.draw(st.integers(a, n))
+    boxes = _global_boxes((rows, cols), a, b, p)
+    # coverage + local-flat ordering
+    pos = a
+    for (goff, sz), fs in boxes:
+        assert fs == pos
+        nel = 1
+        for x in sz:
+            nel *= x
+        # map global rows back to local rows through the segment table
+        for gr in range(goff[0], goff[0] + sz[0]):
+            hits = [l0 + (gr - g0) for l0, ln, g0 in segs if g0 <= gr < g0 + ln]
+            assert len(hits) == 1
+        pos += nel
+    assert pos == b
