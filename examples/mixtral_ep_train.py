@@ -20,6 +20,9 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--batch", type=int, default=4)
     ap.add_argument("--seq", type=int, default=512)
+    ap.add_argument("--rebalance-every", type=int, default=0,
+                    help="dynamically re-place experts by observed load "
+                         "every N steps (moe.rebalance_experts)")
     args = ap.parse_args()
 
     on_gpu = torch.cuda.is_available()
@@ -30,7 +33,11 @@ def main():
 
     from vescale_amd.dtensor import init_device_mesh
     from vescale_amd.models.mixtral import MixtralModel, mixtral_8x7b, mixtral_tiny
-    from vescale_amd.moe import parallelize_experts
+    from vescale_amd.moe import (
+        BasicExpertsAllocator,
+        LoadBalancedExpertsAllocator,
+        parallelize_experts,
+    )
 
     mesh = init_device_mesh(device.type, (world,), mesh_dim_names=("EP",))
     torch.manual_seed(0)
@@ -39,7 +46,11 @@ def main():
         args.seq = 32
     model = MixtralModel(cfg).to(device)
     model.init_weights()
-    parallelize_experts(model, mesh)
+    parallelize_experts(
+        model, mesh,
+        allocator_cls=(LoadBalancedExpertsAllocator if args.rebalance_every
+                       else BasicExpertsAllocator),
+    )
     # non-expert params are replicated -> plain DDP-style allreduce of their
     # grads; expert grads already carry all EP ranks' token contributions
     opt = torch.optim.AdamW([p for p in model.parameters()], lr=3e-4)
@@ -57,6 +68,22 @@ def main():
                 p.grad /= world
         opt.step()
         opt.zero_grad()
+        # dynamic expert re-placement from observed routing load
+        if args.rebalance_every and (step + 1) % args.rebalance_every == 0:
+            from vescale_amd.moe import rebalance_experts
+
+            for layer in model.layers:
+                moe = layer.moe
+                with torch.no_grad():
+                    logits = moe.router(
+                        model.tok_embeddings(x).reshape(-1, cfg.dim)
+                    )
+                    ids = torch.topk(logits, cfg.top_k, dim=-1).indices.reshape(-1)
+                    counts = torch.bincount(ids.cpu(), minlength=cfg.n_experts)
+                moved = rebalance_experts(moe, counts.tolist(), optimizer=opt)
+                if moved and rank == 0:
+                    print(f"step {step}: re-placed experts "
+                          f"{moe.allocator.placement}", flush=True)
         if rank == 0 and step % 5 == 0:
             print(f"step {step} loss {float(loss):.4f}")
     dist.destroy_process_group()
