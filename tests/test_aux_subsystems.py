@@ -329,3 +329,39 @@ def test_emulator_chunk_size_model():
     assert torch.allclose(out2[0].double(), ref64, atol=1e-3)
     # orderings genuinely differ between geometries (fp non-associativity)
     assert not torch.equal(out1[0], out2[0])
+
+
+def _t_disable_redistribute(rank, ws):
+    """VESCALE_DISABLE_REDISTRIBUTE raises on IMPLICIT dispatch-time
+    redistribution while explicit .redistribute() still works."""
+    import os
+
+    import torch
+
+    from vescale_amd.dtensor import DTensor, init_device_mesh
+    from vescale_amd.dtensor.placement_types import Replicate, Shard
+
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("D",))
+    a = DTensor.from_local(torch.randn(2, 4), mesh, [Shard(0)])
+    b = DTensor.from_local(torch.randn(2, 3), mesh, [Shard(1)])
+    os.environ["VESCALE_DISABLE_REDISTRIBUTE"] = "1"
+    try:
+        # mm with mismatched shardings needs an implicit redistribute
+        raised = False
+        try:
+            torch.mm(a, b.redistribute(placements=[Shard(0)]))
+            # Shard(0) @ Shard(0): strategy must move one side -> raises
+        except RuntimeError as e:
+            raised = "implicit redistribute disabled" in str(e)
+        assert raised, "expected implicit-redistribute error"
+        # explicit redistribute unaffected
+        r = a.redistribute(placements=[Replicate()])
+        assert r.to_local().shape == (2 * ws, 4)
+    finally:
+        os.environ.pop("VESCALE_DISABLE_REDISTRIBUTE", None)
+
+
+def test_disable_redistribute_flag():
+    from tests.common import spawn
+
+    spawn(2, _t_disable_redistribute)

@@ -17,6 +17,7 @@ bucketed allreduce in _grad_sync.py.
 from __future__ import annotations
 
 import re
+import os
 from dataclasses import dataclass, field
 from typing import Any, Dict, List, Optional, Sequence, Union
 
@@ -79,7 +80,9 @@ def parallelize_module(
                 continue
             if spec.is_local:
                 d = DTensor.from_local(
-                    p.data, device_mesh, spec.placements, run_check=spec.run_check
+                    p.data, device_mesh, spec.placements,
+                    run_check=spec.run_check
+                    and not os.environ.get("VESCALE_DISABLE_RUN_CHECK")
                 )
             else:
                 d = distribute_tensor(p.data, device_mesh, spec.placements)

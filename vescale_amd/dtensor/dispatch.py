@@ -24,6 +24,21 @@ from ._op_schema import OpSchema, OutputSharding
 from .placement_types import Partial, Placement, Replicate, Shard, TensorMeta
 from .redistribute import redistribute_local_tensor
 
+
+def _implicit_redistribute_guard(op, src_spec, tgt):
+    """VESCALE_DISABLE_REDISTRIBUTE (reference dtensor/README.md:96-98):
+    when set, any IMPLICIT redistribution inside op dispatch raises so
+    hidden communication can be found and planned away.  Explicit
+    DTensor.redistribute() calls are unaffected (they don't go through
+    the dispatcher's input-target path)."""
+    import os
+
+    if os.environ.get("VESCALE_DISABLE_REDISTRIBUTE"):
+        raise RuntimeError(
+            f"implicit redistribute disabled (VESCALE_DISABLE_REDISTRIBUTE): "
+            f"op {op} wants {tuple(tgt)} but input is {src_spec.placements}"
+        )
+
 aten = torch.ops.aten
 
 _STRICT = os.environ.get("VESCALE_AMD_STRICT", "0") == "1"
@@ -144,6 +159,7 @@ class OpDispatcher:
                         for i, tgt in enumerate(sharding.input_targets):
                             if tgt is None or tuple(tgt) == specs[i].placements:
                                 continue
+                            _implicit_redistribute_guard(op, specs[i], tgt)
                             tgt_spec = DTensorSpec(mesh, tuple(tgt), specs[i].tensor_meta)
                             locals_[i] = redistribute_local_tensor(
                                 locals_[i], specs[i], tgt_spec
@@ -252,6 +268,7 @@ class OpDispatcher:
             for i, tgt in enumerate(sharding.input_targets):
                 if tgt is None or tuple(tgt) == specs[i].placements:
                     continue
+                _implicit_redistribute_guard(op, specs[i], tgt)
                 tgt_spec = DTensorSpec(mesh, tuple(tgt), specs[i].tensor_meta)
                 locals_[i] = redistribute_local_tensor(locals_[i], specs[i], tgt_spec)
                 specs[i] = tgt_spec

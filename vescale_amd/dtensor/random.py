@@ -129,10 +129,17 @@ class ThreadBasedRNGTracker(OffsetBasedRNGTracker):
     def exec_random_op(self, op, local_args, local_kwargs, spec: DTensorSpec):
         """Run a random aten op with global philox indexing; returns the
         local result or NotImplemented to fall back."""
+        import os
+
         import torch as _t
 
         from ..ops import has_ext
 
+        # VESCALE_SINGLE_DEVICE_RAND=0 (reference dtensor/random.py:381):
+        # opt OUT of the single-device-parity philox path and use plain
+        # per-rank RNG instead
+        if os.environ.get("VESCALE_SINGLE_DEVICE_RAND", "1") in ("0", "off"):
+            return NotImplemented
         aten = _t.ops.aten
         x = local_args[0]
         if not (isinstance(x, _t.Tensor) and x.is_cuda and has_ext()):

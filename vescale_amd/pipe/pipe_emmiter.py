@@ -60,6 +60,21 @@ class ScheduleEngine:
 
     # ------------------------------------------------------------------
     def build_schedule(self, n_mb: int) -> List[Instr]:
+        sched = self._build_schedule_impl(n_mb)
+        # VESCALE_DUMP_INSTRUCTION (reference env flag): print this
+        # stage's instruction stream for schedule debugging
+        import os
+
+        if os.environ.get("VESCALE_DUMP_INSTRUCTION"):
+            import sys
+
+            print(f"[pipe stage {self.s}] schedule ({len(sched)} instrs):",
+                  file=sys.stderr)
+            for i, ins in enumerate(sched):
+                print(f"  {i:3d}: {ins}", file=sys.stderr)
+        return sched
+
+    def _build_schedule_impl(self, n_mb: int) -> List[Instr]:
         st = self.plan.schedule_type
         if st == PipelineScheduleType.SIMPLE_1F1B:
             if self.V != 1:
