@@ -12,6 +12,7 @@ chunk decompositions (incl. ragged flat ranges as N-D boxes).
 from __future__ import annotations
 
 import concurrent.futures
+import logging
 import os
 from typing import Any, Dict, Optional
 
@@ -52,6 +53,12 @@ def _materialize_state(obj):
     return obj
 
 
+logger = logging.getLogger("vescale_amd.checkpoint")
+if os.environ.get("VESCALE_CHECKPOINT_LOGGING_LEVEL"):
+    # reference env flag: set the checkpoint subsystem's log level by name
+    logger.setLevel(os.environ["VESCALE_CHECKPOINT_LOGGING_LEVEL"].upper())
+
+
 def save(
     path: str,
     checkpoint_state: Dict[str, Any],
@@ -71,6 +78,7 @@ def save(
             os.makedirs(comp_path, exist_ok=True)
         if dist.is_initialized():
             dist.barrier()
+        logger.debug("checkpoint save: component %s -> %s", key, comp_path)
         if async_checkpoint and not dist.is_initialized():
             # stage to CPU (pinned pool on GPU), write in background
             cpu_sd = _to_cpu(sd)

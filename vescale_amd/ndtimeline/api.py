@@ -21,6 +21,16 @@ def init_ndtimers(
     log: bool = False,
 ) -> NDTimerManager:
     global _chrome
+    # VESCALE_NDTIMELINE_LOG_LEVEL (reference env): override the metric
+    # level by name, e.g. DEBUG/INFO/WARNING
+    import os
+
+    env_level = os.environ.get("VESCALE_NDTIMELINE_LOG_LEVEL")
+    if env_level:
+        try:
+            level = NDMetricLevel[env_level.upper()]
+        except KeyError:
+            pass
     GlobalReferenceTime.calibrate()
     mgr = NDTimerManager(level)
     if chrome_trace_path:
