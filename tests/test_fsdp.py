@@ -53,13 +53,16 @@ def _train(rank, ws, out_path, n_steps):
             json.dump(losses, f)
 
 
-@pytest.mark.parametrize("n_steps", [4])
-def test_fsdp_ws2_parity(n_steps):
+@pytest.mark.parametrize("ws", [2, 4])
+def test_fsdp_ws_parity(ws):
+    """ws=1 vs ws=N loss parity — N=4 exercises multi-unit RaggedShard
+    math beyond the pairwise case (insurance for the 8-GPU scale run)."""
+    n_steps = 4
     with tempfile.TemporaryDirectory() as td:
         p1 = os.path.join(td, "ws1.json")
-        p2 = os.path.join(td, "ws2.json")
+        p2 = os.path.join(td, "wsN.json")
         spawn(1, _train, p1, n_steps)
-        spawn(2, _train, p2, n_steps)
+        spawn(ws, _train, p2, n_steps)
         l1 = json.load(open(p1))
         l2 = json.load(open(p2))
         assert len(l1) == len(l2) == n_steps
@@ -125,3 +128,29 @@ def _t_grad_accumulation(rank, ws):
 
 def test_fsdp_grad_accumulation():
     spawn(2, _t_grad_accumulation)
+
+
+def _t_ws8_step(rank, ws):
+    """One step at ws=8 (the scale-run world size) — shard math, comms
+    and optimizer must hold beyond the small worlds."""
+    from vescale_amd.dtensor import init_device_mesh
+
+    torch.manual_seed(42)
+    cfg = llama_tiny()
+    model = LlamaModel(cfg)
+    model.init_weights()
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("DP",))
+    eng = FSDP(model, mesh, param_dtype=torch.float32, device=torch.device("cpu"))
+    opt = FlatAdamW(eng, lr=1e-3, grad_clip=1.0)
+    g = torch.Generator().manual_seed(5)
+    x = torch.randint(0, cfg.vocab_size, (8, 32), generator=g)
+    xs = torch.chunk(x, ws)[rank]
+    loss = eng(xs, torch.roll(xs, -1, 1))
+    loss.backward()
+    opt.step()
+    l2 = eng(xs, torch.roll(xs, -1, 1))
+    assert float(l2) < float(loss), "loss should drop after one step"
+
+
+def test_fsdp_ws8_step():
+    spawn(8, _t_ws8_step)
