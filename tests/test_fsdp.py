@@ -154,3 +154,34 @@ def _t_ws8_step(rank, ws):
 
 def test_fsdp_ws8_step():
     spawn(8, _t_ws8_step)
+
+
+def test_llama70b_memory_plan_fits_mi355x():
+    """BASELINE config 'Llama-3 70B FSDP+TP 2D mesh on 8x MI355X':
+    construct the 70B architecture on the meta device (deferred init, no
+    allocation) and check the per-GPU steady-state memory plan fits the
+    288 GB HBM3E budget with FSDP=8 sharding of params/grads/opt state."""
+    from vescale_amd.initialize import deferred_init, is_deferred
+    from vescale_amd.models.llama import LlamaModel, llama3_70b
+
+    cfg = llama3_70b()
+    model = deferred_init(LlamaModel, cfg)
+    assert is_deferred(model)
+    n_params = sum(p.numel() for p in model.parameters())
+    assert 68e9 < n_params < 72e9, n_params
+
+    ws = 8
+    GB = 1024**3
+    param_shard = n_params * 2 / ws            # bf16 shard
+    grad_shard = n_params * 2 / ws             # bf16 grad shard (ZeRO-2+)
+    opt_state = n_params * 12 / ws             # fp32 master + m + v
+    # transient: largest unit gathered full (embeddings+head ~= 2*vocab*d)
+    full_unit = 2 * cfg.vocab_size * cfg.dim * 2
+    # activations, bs1 x seq8192, full recompute granularity: per-layer
+    # boundary activation + one layer live
+    act = 8192 * cfg.dim * 2 * (cfg.n_layers + 6)
+    total = param_shard + grad_shard + opt_state + full_unit + act
+    assert total < 288 * GB, f"plan needs {total/GB:.0f} GB"
+    # and it does NOT fit a 141 GB (H200-class) device -> the 288 GB HBM
+    # is what makes single-node 70B FSDP8 viable
+    assert total > 141 * GB
