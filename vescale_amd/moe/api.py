@@ -90,6 +90,9 @@ def rebalance_experts(
         return False
     new = [alloc.owner_of(e) for e in range(alloc.n_experts)]
 
+    # NOTE: optimizer state transfer covers param_groups[0] and the Adam
+    # family keys below — matching how the MoE examples/tests construct
+    # the expert optimizer (a single param group)
     opt_keys = ("exp_avg", "exp_avg_sq", "step")
     for e in range(alloc.n_experts):
         if old[e] == new[e]:
