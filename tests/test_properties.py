@@ -153,3 +153,20 @@ def test_global_boxes_roundtrip(rows, cols, nseg, data):
             assert len(hits) == 1
         pos += nel
     assert pos == b
+
+
+@settings(max_examples=80, deadline=None)
+@given(nwg=st.integers(1, 4096))
+def test_xcd_bijective_remap(nwg):
+    """The m204 bijective XCD remap used by gemm8.hip (and the simpler
+    %8==0 form in the attention kernels) must be a bijection on
+    [0, nwg) for ANY grid size — the original m157 form was not
+    (guide ERRATA #11)."""
+    q, r = nwg >> 3, nwg & 7
+    seen = set()
+    for f in range(nwg):
+        xcd, off = f & 7, f >> 3
+        g = (xcd * (q + 1) if xcd < r else r * (q + 1) + (xcd - r) * q) + off
+        assert 0 <= g < nwg
+        seen.add(g)
+    assert len(seen) == nwg
