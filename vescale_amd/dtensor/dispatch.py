@@ -15,13 +15,13 @@ tuples without pytree, and cache hits skip all rule logic.
 from __future__ import annotations
 
 import os
-from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple
+from typing import Any, Callable, Dict, List, Optional
 
 import torch
 
 from ._dtensor_spec import DTensorSpec
 from ._op_schema import OpSchema, OutputSharding
-from .placement_types import Partial, Placement, Replicate, Shard, TensorMeta
+from .placement_types import Replicate, Shard, TensorMeta
 from .redistribute import redistribute_local_tensor
 
 

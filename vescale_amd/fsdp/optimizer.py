@@ -12,7 +12,7 @@ from typing import Dict, Optional
 import torch
 import torch.distributed as dist
 
-from ..ops import adamw_step_flat, l2norm_sq, scale_flat_
+from ..ops import adamw_step_flat
 from .engine import FSDP
 
 

@@ -8,7 +8,7 @@ same all-to-all machinery.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 import torch.distributed as dist
 
