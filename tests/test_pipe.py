@@ -227,3 +227,33 @@ def _t_shared_params(rank, ws):
 
 def test_pp_shared_params_sync():
     spawn(2, _t_shared_params)
+
+
+def test_wgrad_store_true_split():
+    """Zero-bubble W/B split is REAL: after a B-phase backward the Linear
+    weight grads are absent (deferred closures queued), input grads are
+    exact, and the W-phase produces grads identical to a plain backward —
+    with NO retain_graph (the graph is freed at B time)."""
+    from vescale_amd.pipe.wgrad_store import WeightGradStore, zb_patch_linears
+
+    torch.manual_seed(0)
+    ref = nn.Sequential(nn.Linear(6, 5), nn.GELU(), nn.Linear(5, 4))
+    zb = nn.Sequential(nn.Linear(6, 5), nn.GELU(), nn.Linear(5, 4))
+    zb.load_state_dict(ref.state_dict())
+    store = WeightGradStore()
+    assert zb_patch_linears(zb, store) == 2
+    x = torch.randn(3, 6, requires_grad=True)
+    xr = x.detach().clone().requires_grad_()
+    ref(xr).square().sum().backward()
+    store.begin()
+    zb(x).square().sum().backward()   # no retain_graph
+    store.end((0, 0))
+    assert zb[0].weight.grad is None and zb[2].weight.grad is None
+    assert torch.allclose(x.grad, xr.grad, atol=1e-6)
+    store.pop_run((0, 0))
+    for a, b in zip(zb.parameters(), ref.parameters()):
+        assert torch.allclose(a.grad, b.grad, atol=1e-6)
+    # inactive store: patched layers compute weight grads inline
+    zb.zero_grad()
+    zb(x.detach()).sum().backward()
+    assert zb[0].weight.grad is not None
