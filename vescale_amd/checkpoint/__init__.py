@@ -21,7 +21,8 @@ import torch.distributed as dist
 import torch.distributed.checkpoint as dcp
 
 from . import dtensor_dcp  # installs DTensor DCP hooks  # noqa: F401
-from .mem_pool import (
+from .mem_pool import (  # noqa: F401
+    mem_checkpoint_path,
     GLOBAL_POOL,
     PinnedStoragePool,
     copy_gpu_tensor_to_cpu_pinned_mem_pool,

@@ -59,3 +59,25 @@ def release_cpu_tensor(t: torch.Tensor):
     buf = getattr(t, "_pool_buf", None)
     if buf is not None:
         GLOBAL_POOL.release(buf)
+
+
+def mem_checkpoint_path(name: str = "vescale_amd_ckpt") -> str:
+    """An IN-MEMORY checkpoint directory (tmpfs).
+
+    Parity: the reference ships a gRPC in-memory file server
+    (checkpoint/utilities/server/mem_server_lib.py) so checkpoints can be
+    staged to RAM.  On the MI355X single-node deployment the same
+    capability is the host's tmpfs: DCP save/load against /dev/shm is an
+    in-memory checkpoint with zero extra machinery (cross-host restore
+    still goes through the shared filesystem path).
+    """
+    import os
+
+    base = "/dev/shm" if os.path.isdir("/dev/shm") else None
+    if base is None:
+        import tempfile
+
+        base = tempfile.gettempdir()
+    path = os.path.join(base, name)
+    os.makedirs(path, exist_ok=True)
+    return path
