@@ -229,11 +229,7 @@ def test_pp_shared_params_sync():
     spawn(2, _t_shared_params)
 
 
-def test_wgrad_store_true_split():
-    """Zero-bubble W/B split is REAL: after a B-phase backward the Linear
-    weight grads are absent (deferred closures queued), input grads are
-    exact, and the W-phase produces grads identical to a plain backward —
-    with NO retain_graph (the graph is freed at B time)."""
+def _wgrad_split_body():
     from vescale_amd.pipe.wgrad_store import WeightGradStore, zb_patch_linears
 
     torch.manual_seed(0)
@@ -257,3 +253,25 @@ def test_wgrad_store_true_split():
     zb.zero_grad()
     zb(x.detach()).sum().backward()
     assert zb[0].weight.grad is not None
+    print("WB_OK")
+
+
+def test_wgrad_store_true_split():
+    """Zero-bubble W/B split is REAL: after a B-phase backward the Linear
+    weight grads are absent (deferred closures queued), input grads are
+    exact, and the W-phase produces grads identical to a plain backward —
+    with NO retain_graph.  Subprocess-isolated: parent-side autograd
+    backward warms engine threads that poison later fork+gloo tests
+    (tests/README.md fork-safety rule)."""
+    import os
+    import subprocess
+    import sys
+
+    code = (
+        "import sys; sys.path.insert(0, %r); "
+        "from tests.test_pipe import _wgrad_split_body; _wgrad_split_body()"
+    ) % os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                         text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert "WB_OK" in out.stdout

@@ -30,7 +30,11 @@ def _implicit_redistribute_guard(op, src_spec, tgt):
     when set, any IMPLICIT redistribution inside op dispatch raises so
     hidden communication can be found and planned away.  Explicit
     DTensor.redistribute() calls are unaffected (they don't go through
-    the dispatcher's input-target path)."""
+    the dispatcher's input-target path).  NOTE a deliberate default
+    difference: the reference ships with implicit redistribution DISABLED
+    by default (_diff.py:24 reads the env with default "1"); this
+    framework allows it by default (friendlier eager UX) and the flag
+    opts INTO strictness."""
     import os
 
     if os.environ.get("VESCALE_DISABLE_REDISTRIBUTE"):

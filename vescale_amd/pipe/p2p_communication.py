@@ -76,7 +76,33 @@ def drain_send_reqs():
     _SEND_QUEUE.clear()
 
 
+def _log_p2p(ops) -> None:
+    """VESCALE_DUMMY_P2P / debug tracing (reference dtensor/_diff.py:32
+    dummy_p2p): when VESCALE_DUMMY_P2P is set, log every p2p batch through
+    DebugLogger (per-rank files / stderr) — schedule-debugging visibility
+    without reading tensors."""
+    import os
+
+    if not os.environ.get("VESCALE_DUMMY_P2P"):
+        return
+    import logging
+
+    try:
+        desc = ", ".join(
+            f"{'send' if o.op is dist.isend else 'recv'}"
+            f"(peer={o.peer}, shape={tuple(o.tensor.shape)})"
+            for o in ops
+        )
+    except Exception:
+        desc = f"{len(ops)} ops"
+    rank = dist.get_rank() if dist.is_initialized() else 0
+    logging.getLogger("vescale_amd.pipe.p2p").warning(
+        "[rank %d] p2p: %s", rank, desc
+    )
+
+
 def _run_p2p_ops(ops: List[dist.P2POp], pg):
+    _log_p2p(ops)
     """Post all ops; WAIT only on receives, queue sends for a later drain.
     Pre-posting-free deadlock safety: a send can then never participate in
     a rendezvous cycle on the host side."""
