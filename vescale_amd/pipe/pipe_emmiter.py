@@ -85,6 +85,11 @@ class ScheduleEngine:
         if st == PipelineScheduleType.INTERLEAVED_1F1B:
             return interleaved_1f1b_schedule(self.s, self.P, n_mb, self.V)
         if st == PipelineScheduleType.ZERO_BUBBLE:
+            if self.V != 1:
+                raise ValueError(
+                    "zero-bubble schedule requires virtual_chunks == 1 "
+                    "(ZB-V with virtual chunks is future work)"
+                )
             from .zero_bubble import zero_bubble_schedule
 
             return zero_bubble_schedule(self.s, self.P, n_mb)
