@@ -107,6 +107,61 @@ def test_do_state_dict():
     spawn(2, _t_do_state_dict)
 
 
+def _t_do_multibucket_rs_semantics(rank, ws, n_steps=3):
+    """Regression: DistributedOptimizer range maps must be PER BUCKET.
+
+    GradBuffer reduce-scatters each bucket independently, so after grad sync
+    only [bucket.offset + rank*shard, bucket.offset + (rank+1)*shard) holds
+    reduced data on RCCL.  gloo falls back to all_reduce (whole buffer valid),
+    which masked the round-1 bug — so here we POISON every non-owned region
+    of every bucket with NaN after sync to emulate reduce-scatter validity,
+    then verify the optimizer still reproduces the single-process baseline.
+    """
+    torch.manual_seed(3)
+    net = Net(d=64)
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("DP",))
+    # tiny buckets => several buckets (old code read unreduced data here)
+    ddp = DDP(net, mesh, use_distributed_optimizer=True, bucket_size=4096)
+    gb0 = next(iter(ddp.grad_buffers.values()))
+    assert len(gb0.buckets) > 1, "test needs multiple buckets"
+    inner = torch.optim.AdamW(net.parameters(), lr=1e-2)
+    opt = DistributedOptimizer(inner, [ddp])
+
+    g = torch.Generator().manual_seed(11)
+    data = [torch.randn(8, 64, generator=g) for _ in range(n_steps)]
+
+    for x in data:
+        loss = ddp(torch.chunk(x, ws)[rank]).pow(2).mean()
+        loss.backward()
+        ddp.finish_grad_sync()
+        for gbuf in ddp.grad_buffers.values():
+            for b in gbuf.buckets:
+                s = b.data.numel() // ws
+                for r in range(ws):
+                    if r != rank:
+                        b.data.narrow(0, r * s, s).fill_(float("nan"))
+        opt.step()
+        opt.zero_grad()
+
+    # single-process baseline
+    torch.manual_seed(3)
+    ref = Net(d=64)
+    ropt = torch.optim.AdamW(ref.parameters(), lr=1e-2)
+    for x in data:
+        l = ref(x).pow(2).mean()
+        l.backward()
+        ropt.step()
+        ropt.zero_grad()
+
+    for (n1, p1), (n2, p2) in zip(net.named_parameters(), ref.named_parameters()):
+        assert torch.isfinite(p1).all(), f"{n1} has non-finite values"
+        assert torch.allclose(p1, p2, atol=1e-5), (n1, (p1 - p2).abs().max())
+
+
+def test_do_multibucket_reduce_scatter_semantics():
+    spawn(2, _t_do_multibucket_rs_semantics)
+
+
 def _t_2d_tp_dp(rank, ws):
     """DP x TP 2x2: DModule TP inside, DDP outside — loss parity vs single."""
     from vescale_amd.dmodule import parallelize_module

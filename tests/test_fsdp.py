@@ -206,3 +206,47 @@ def test_llama70b_memory_plan_fits_mi355x():
     )
     assert out.returncode == 0, out.stderr[-1500:]
     assert "PLAN_OK" in out.stdout
+
+
+def _t_frozen_param(rank, ws):
+    """Regression: a unit containing an unused requires_grad
+    param must still reduce-scatter the grads it DID produce, and the
+    readiness state must not leak into the next step."""
+    import torch.nn as nn
+    from vescale_amd.dtensor import init_device_mesh
+
+    torch.manual_seed(7)
+
+    class Branchy(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.used = nn.Linear(8, 8, bias=False)
+            self.unused = nn.Linear(8, 8, bias=False)  # requires_grad, no grad
+
+        def forward(self, x):
+            return self.used(x)
+
+    model = Branchy()
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("DP",))
+    eng = FSDP(model, mesh, param_dtype=torch.float32,
+               device=torch.device("cpu"))
+    opt = FlatAdamW(eng, lr=1e-2, weight_decay=0.0)
+    g = torch.Generator().manual_seed(5)
+    for step in range(3):
+        used_before = model.used.weight.detach().clone()
+        unused_before = model.unused.weight.detach().clone()
+        x = torch.randn(4, 8, generator=g)
+        out = eng(torch.chunk(x, ws)[rank])
+        loss = out.pow(2).mean()
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        # the used param's grads must be reduced + applied EVERY step (the
+        # old counter wedged the unit after step 1 -> no update), and the
+        # zero-grad param must not drift
+        assert not torch.allclose(model.used.weight, used_before), step
+        assert torch.allclose(model.unused.weight, unused_before), step
+
+
+def test_fsdp_unused_param_unit_flush():
+    spawn(2, _t_frozen_param)

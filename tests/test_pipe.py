@@ -127,9 +127,12 @@ def test_fx_pipe_parser():
     assert torch.allclose(h, ref, atol=1e-6)
 
 
-def _t_pp_dp_3d(rank, ws):
+def _t_pp_dp_3d(rank, ws, sched="1f1b"):
     """PP=2 x DP=2 (+ DistributedOptimizer ZeRO-2): loss parity vs single
-    device (reference 4D-alignment methodology, minus TP)."""
+    device (reference 4D-alignment methodology, minus TP).  Run with
+    sched="zero_bubble_v" to cover deferred W-phase grads flowing into the
+    DDP main_grad buffer (regression: closures once wrote param.grad
+    directly, silently dropping Linear weight grads from DP reduction)."""
     import torch.distributed as dist
     from vescale_amd.ddp import DistributedDataParallel as DDP
     from vescale_amd.dtensor import init_device_mesh
@@ -145,7 +148,7 @@ def _t_pp_dp_3d(rank, ws):
     mods = _make_modules()
     plan = PipelineParallelPlan(
         num_stages=2,
-        schedule_type=PipelineScheduleType.SIMPLE_1F1B,
+        schedule_type=PipelineScheduleType(sched),
         split_method=PipelineSplitMethodType.UNIFORM,
     )
     stage = construct_pipeline_stage(mods, plan, pp_rank)
@@ -197,6 +200,10 @@ def _t_pp_dp_3d(rank, ws):
 
 def test_pp_dp_zero2_3d():
     spawn(4, _t_pp_dp_3d)
+
+
+def test_pp_dp_zero_bubble_wgrads_reduced():
+    spawn(4, _t_pp_dp_3d, "zero_bubble_v")
 
 
 def _t_shared_params(rank, ws):

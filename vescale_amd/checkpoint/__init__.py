@@ -80,10 +80,23 @@ def save(
         if dist.is_initialized():
             dist.barrier()
         logger.debug("checkpoint save: component %s -> %s", key, comp_path)
-        if async_checkpoint and not dist.is_initialized():
-            # stage to CPU (pinned pool on GPU), write in background
-            cpu_sd = _to_cpu(sd)
-            fut = _executor.submit(_do_save, comp_path, cpu_sd, process_group)
+        if async_checkpoint:
+            if dist.is_initialized():
+                # dcp.async_save stages (D2H) synchronously, then the
+                # serialization + write run in a background thread with the
+                # plan collectives coordinated safely (reference async
+                # futures, api/vescale_checkpointer.py:85)
+                fut = dcp.async_save(
+                    sd,
+                    storage_writer=dcp.FileSystemWriter(comp_path),
+                    planner=VeScaleSavePlanner(),
+                    process_group=process_group,
+                )
+            else:
+                # single process: stage to CPU (pinned pool on GPU), write in
+                # a background thread
+                cpu_sd = _to_cpu(sd)
+                fut = _executor.submit(_do_save, comp_path, cpu_sd, process_group)
             futures.append(fut)
             _pending.append(fut)
         else:
@@ -118,6 +131,27 @@ def wait_pending():
     _pending.clear()
 
 
+def _iter_state_tensors(obj):
+    """Deterministic-order walk over every tensor leaf of a state dict
+    (sorted keys, so all ranks traverse identically)."""
+    if isinstance(obj, dict):
+        for k in sorted(obj.keys(), key=str):
+            yield from _iter_state_tensors(obj[k])
+    elif isinstance(obj, (list, tuple)):
+        for v in obj:
+            yield from _iter_state_tensors(v)
+    elif isinstance(obj, torch.Tensor):
+        yield obj
+
+
+def _broadcast_state(sd, group) -> None:
+    """Broadcast every tensor leaf in-place from group-rank 0."""
+    src = dist.get_global_rank(group, 0) if group is not None else 0
+    for t in _iter_state_tensors(sd):
+        local = t._local_tensor if hasattr(t, "_local_tensor") else t
+        dist.broadcast(local, src=src, group=group)
+
+
 def load(
     path: str,
     checkpoint_state: Dict[str, Any],
@@ -126,19 +160,41 @@ def load(
     process_group=None,
 ):
     """In-place load into the given model/optimizer objects (resharding as
-    needed).  broadcast_checkpoint: only DP-rank-0 reads from storage and
-    results are broadcast over DP (reference api/vescale_checkpointer.py:160
-    broadcast-load) — here DCP reads per-rank chunks directly, which on a
-    shared filesystem is equivalent; flag kept for API parity."""
+    needed).
+
+    broadcast_checkpoint: only rank 0 of `process_group` (default: WORLD)
+    reads from storage (single-rank DCP read of its local plan) and every
+    tensor leaf is broadcast in-place to the other ranks (reference
+    api/vescale_checkpointer.py:160-214 DP-broadcast load).  Contract: the
+    state must be REPLICATED across the group (DDP replicas over DP) — for
+    DP-sharded state (FSDP/RaggedShard over the same group) use the default
+    per-rank DCP read, which fetches exactly each rank's chunks.
+    """
+    use_bcast = (
+        broadcast_checkpoint
+        and dist.is_initialized()
+        and dist.get_world_size(process_group) > 1
+    )
+    group_rank = dist.get_rank(process_group) if use_bcast else 0
     for key, obj in checkpoint_state.items():
         comp_path = os.path.join(path, key)
         sd = _materialize_state(obj)
-        dcp.load(
-            sd,
-            storage_reader=dcp.FileSystemReader(comp_path),
-            planner=VeScaleLoadPlanner(),
-            process_group=process_group,
-        )
+        if use_bcast:
+            if group_rank == 0:
+                dcp.load(
+                    sd,
+                    storage_reader=dcp.FileSystemReader(comp_path),
+                    planner=VeScaleLoadPlanner(),
+                    no_dist=True,
+                )
+            _broadcast_state(sd, process_group)
+        else:
+            dcp.load(
+                sd,
+                storage_reader=dcp.FileSystemReader(comp_path),
+                planner=VeScaleLoadPlanner(),
+                process_group=process_group,
+            )
         # push back into stateful objects that need it
         if hasattr(obj, "load_sharded_state_dict"):
             obj.load_sharded_state_dict(sd)

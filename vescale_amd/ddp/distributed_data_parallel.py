@@ -9,6 +9,7 @@ plain modules.
 """
 from __future__ import annotations
 
+from contextlib import contextmanager
 from typing import Dict, List, Optional, Union
 
 import torch
@@ -70,7 +71,12 @@ class DistributedDataParallel(nn.Module):
                 self._grad_buffer_of[p] = gb
         for p in params:
             p.main_grad = self._grad_buffer_of[p].get_main_grad_view(p)
-            p.register_post_accumulate_grad_hook(self._make_param_hook(p))
+            hook = self._make_param_hook(p)
+            p.register_post_accumulate_grad_hook(hook)
+            # exposed so out-of-autograd grad producers (zero-bubble deferred
+            # W-phase closures, pipe/wgrad_store.py) route through the SAME
+            # main_grad-accumulate + register_grad_ready path
+            p._ddp_param_hook = hook
 
     # ------------------------------------------------------------------
     def _make_param_hook(self, param):
@@ -94,6 +100,21 @@ class DistributedDataParallel(nn.Module):
     # ------------------------------------------------------------------
     def forward(self, *args, **kwargs):
         return self.module(*args, **kwargs)
+
+    def set_is_last_microbatch(self, flag: bool):
+        """With gradient accumulation (PP microbatches / ZB W-phases), call
+        with False for all but the final microbatch so bucket-complete hooks
+        don't start the DP sync early (Megatron is_last_microbatch flag)."""
+        for gb in self.grad_buffers.values():
+            gb.set_is_last_microbatch(flag)
+
+    @contextmanager
+    def no_sync(self):
+        self.set_is_last_microbatch(False)
+        try:
+            yield
+        finally:
+            self.set_is_last_microbatch(True)
 
     def zero_grad_buffer(self):
         for gb in self.grad_buffers.values():

@@ -40,6 +40,10 @@ class Bucket:
         self.overlap_grad_reduce = overlap_grad_reduce
         self.comm_handle = None
         self.comm_issued = False
+        # Megatron-style flag: with gradient accumulation (PP microbatches,
+        # zero-bubble W-phases) hooks fire once per microbatch; sync must only
+        # start on the LAST one.  Default True = one backward per step.
+        self.is_last_microbatch = True
 
     def reset(self):
         self.params_with_grad = set()
@@ -83,7 +87,12 @@ class Bucket:
     def register_grad_ready(self, param):
         assert param in self.params
         self.params_with_grad.add(param)
-        if self.overlap_grad_reduce and len(self.params_with_grad) == len(self.params):
+        if (
+            self.overlap_grad_reduce
+            and self.is_last_microbatch
+            and not self.comm_issued
+            and len(self.params_with_grad) == len(self.params)
+        ):
             self.start_grad_sync()
 
 
@@ -169,6 +178,10 @@ class GradBuffer:
 
     def register_grad_ready(self, param):
         self.bucket_of_param[param].register_grad_ready(param)
+
+    def set_is_last_microbatch(self, flag: bool):
+        for b in self.buckets:
+            b.is_last_microbatch = flag
 
 
 def _numel_local(p: torch.nn.Parameter) -> int:

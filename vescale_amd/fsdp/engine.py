@@ -116,7 +116,7 @@ class FSDPUnit:
         self.params: List[nn.Parameter] = [p for _, p in params]
         self.grad_full: Optional[torch.Tensor] = None
         self.grad_shard: Optional[torch.Tensor] = None
-        self._grads_ready = 0
+        self._grads_ready: set = set()
         self._ag_event: Optional[torch.cuda.Event] = None
         self._rs_work = None
         self._is_unsharded = False
@@ -410,10 +410,13 @@ class FSDP(nn.Module):
     # ---------------- gradient path ----------------
     def _make_grad_hook(self, unit: FSDPUnit):
         def hook(param):
-            unit._grads_ready += 1
-            if unit._grads_ready < len(unit.params):
+            # per-param readiness SET, not a counter: an unused requires_grad
+            # param (frozen/conditional branch) must not wedge the unit, and
+            # a repeated hook on the same param must not double-count
+            unit._grads_ready.add(id(param))
+            if len(unit._grads_ready) < len(unit.params):
                 return
-            unit._grads_ready = 0
+            unit._grads_ready.clear()
             self._finish_unit_grads(unit)
 
         return hook
@@ -429,10 +432,8 @@ class FSDP(nn.Module):
                 unit.grad_shard = unit.grad_full
                 return
             accumulate = unit.grad_shard is not None
-            target = (
-                torch.empty(unit.shard_numel, dtype=unit.param_dtype, device=self.device)
-                if accumulate
-                else torch.empty(unit.shard_numel, dtype=unit.param_dtype, device=self.device)
+            target = torch.empty(
+                unit.shard_numel, dtype=unit.param_dtype, device=self.device
             )
             gf = unit.grad_full
             op = dist.ReduceOp.AVG if hasattr(dist.ReduceOp, "AVG") else dist.ReduceOp.SUM
@@ -468,6 +469,13 @@ class FSDP(nn.Module):
         """Wait for all grad reduce-scatters (call after loss.backward())."""
         self._in_backward = False
         self._exec_order_done = True
+        # flush units whose readiness never hit len(params) this backward
+        # (unused requires_grad params): their produced grads are in
+        # grad_full (zero elsewhere) and must still be reduced
+        for u in self.units:
+            if u._grads_ready and u.grad_full is not None:
+                u._grads_ready.clear()
+                self._finish_unit_grads(u)
         if self._on_gpu and self.world_size > 1:
             torch.cuda.current_stream().wait_stream(self._rs_stream)
         # re-scale: RCCL without AVG support -> grads were summed
@@ -493,7 +501,7 @@ class FSDP(nn.Module):
         for u in self.units:
             u.grad_full = None
             u.grad_shard = None
-            u._grads_ready = 0
+            u._grads_ready.clear()
             for p in u.params:
                 p.grad = None
 

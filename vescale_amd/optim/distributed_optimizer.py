@@ -66,29 +66,44 @@ class DistributedOptimizer:
             for name, p in target.named_parameters():
                 self._fqn_of[id(p)] = name
 
-        # build range maps + master slices per grad buffer
+        # build range maps + master slices per grad buffer.
+        # CRITICAL: ranges are PER BUCKET, not one contiguous range over the
+        # whole buffer — GradBuffer reduce-scatters each bucket independently,
+        # so this rank's reduced grads land at
+        # [bucket.offset + rank*shard, bucket.offset + (rank+1)*shard) within
+        # EVERY bucket (reference build_model_gbuf_range uses bucket.offset the
+        # same way, legacy/vescale/optim/distributed_optimizer.py:399-600).
         # entries: (model, gbuf, param, pstart, pend, istart, iend, main)
         self._slices = []
-        self.param_buffers = []  # (gbuf, param_dtype_buffer)
+        self.param_buffers = []  # (model, gbuf, pbuf, [(off, size, r0, r1)])
         for m in self.models:
             for dt, gb in m.grad_buffers.items():
-                s = gb.numel // max(self.dp_world, 1)
-                r0, r1 = self.dp_rank * s, (self.dp_rank + 1) * s
-                if self.dp_world == 1:
-                    r0, r1 = 0, gb.numel
                 pdtype = None
-                for p, (ps, pe) in gb.param_index_map.items():
-                    pdtype = _local(p).dtype
-                    i0, i1 = max(ps, r0), min(pe, r1)
-                    if i0 >= i1:
-                        continue
-                    lf = _local(p).reshape(-1)
-                    main = lf[i0 - ps : i1 - ps].detach().float().clone()
-                    self._slices.append((m, gb, p, ps, pe, i0, i1, main))
+                bucket_ranges = []  # (bucket_off, bucket_size, r0, r1)
+                for b in gb.buckets:
+                    bn = b.data.numel()
+                    if self.dp_world <= 1:
+                        r0, r1 = b.offset, b.offset + bn
+                    else:
+                        s = bn // self.dp_world
+                        r0 = b.offset + self.dp_rank * s
+                        r1 = r0 + s
+                    bucket_ranges.append((b.offset, bn, r0, r1))
+                    for p in sorted(
+                        b.params, key=lambda q: gb.param_index_map[q][0]
+                    ):
+                        ps, pe = gb.param_index_map[p]
+                        pdtype = _local(p).dtype
+                        i0, i1 = max(ps, r0), min(pe, r1)
+                        if i0 >= i1:
+                            continue
+                        lf = _local(p).reshape(-1)
+                        main = lf[i0 - ps : i1 - ps].detach().float().clone()
+                        self._slices.append((m, gb, p, ps, pe, i0, i1, main))
                 pbuf = torch.empty(
                     gb.numel, dtype=pdtype or torch.float32, device=gb.data.device
                 )
-                self.param_buffers.append((m, gb, pbuf, r0, r1))
+                self.param_buffers.append((m, gb, pbuf, bucket_ranges))
 
         # rebuild inner-optimizer param groups over the master slices
         mains_of_param: Dict[int, List[torch.Tensor]] = {}
@@ -114,20 +129,25 @@ class DistributedOptimizer:
                 main.grad = main.grad.float()
 
     def _copy_main_params_to_model_params(self):
-        # write master -> param buffer, gather, scatter back to params
-        for (m, gb, pbuf, r0, r1) in self.param_buffers:
-            # fill this rank's range from the masters
+        # write master -> param buffer, gather per bucket, scatter to params
+        for (m, gb, pbuf, bucket_ranges) in self.param_buffers:
+            # fill this rank's owned ranges from the masters
             for (m2, gb2, p, ps, pe, i0, i1, main) in self._slices:
                 if gb2 is not gb:
                     continue
                 pbuf.narrow(0, i0, i1 - i0).copy_(main.to(pbuf.dtype))
             if self.dp_world > 1:
-                shard = pbuf.narrow(0, r0, r1 - r0).clone()
-                try:
-                    dist.all_gather_into_tensor(pbuf, shard, group=self.dp_group)
-                except RuntimeError:
-                    bufs = list(pbuf.chunk(self.dp_world))
-                    dist.all_gather(bufs, shard, group=self.dp_group)
+                # gather each bucket: rank r's shard sits at off + r*shard
+                for (off, size, r0, r1) in bucket_ranges:
+                    bview = pbuf.narrow(0, off, size)
+                    shard = pbuf.narrow(0, r0, r1 - r0).clone()
+                    try:
+                        dist.all_gather_into_tensor(
+                            bview, shard, group=self.dp_group
+                        )
+                    except RuntimeError:
+                        bufs = list(bview.chunk(self.dp_world))
+                        dist.all_gather(bufs, shard, group=self.dp_group)
             # copy back into param locals
             for p, (ps, pe) in gb.param_index_map.items():
                 _local(p).reshape(-1).copy_(pbuf.narrow(0, ps, pe - ps))

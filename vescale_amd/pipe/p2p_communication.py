@@ -31,36 +31,94 @@ def _use_batched(pg) -> bool:
         return False
 
 
-def _meta_encode(t: torch.Tensor) -> torch.Tensor:
+def _p2p_device(pg) -> torch.device:
+    """Device for p2p payloads AND meta tensors.  RCCL (backend 'nccl' on
+    ROCm) rejects CPU tensors in batch_isend_irecv, so the shape handshake
+    must ride device tensors on NCCL groups (reference _communicate_shapes
+    allocates on the compute device, p2p_communication.py:125)."""
+    try:
+        backend = str(dist.get_backend(pg))
+    except Exception:
+        backend = "gloo"
+    if "nccl" in backend and torch.cuda.is_available():
+        return torch.device("cuda", torch.cuda.current_device())
+    return torch.device("cpu")
+
+
+def _meta_encode(t: torch.Tensor, device=None) -> torch.Tensor:
     m = torch.zeros(_MAX_DIMS + 2, dtype=torch.int64)
     m[0] = t.ndim
     for i, s in enumerate(t.shape):
         m[1 + i] = s
     m[_MAX_DIMS + 1] = _DTYPES.index(t.dtype)
-    return m
+    return m.to(device) if device is not None else m
 
 
 def _meta_decode(m: torch.Tensor) -> Tuple[Tuple[int, ...], torch.dtype]:
+    m = m.cpu()
     nd = int(m[0])
     shape = tuple(int(m[1 + i]) for i in range(nd))
     return shape, _DTYPES[int(m[_MAX_DIMS + 1])]
 
 
+# Shape cache: after the first handshake for a (pg, peer, direction) key the
+# metadata is reused, skipping the extra p2p round-trip for every microbatch
+# (reference REUSE_COMM_SHAPE env, pipe/p2p_communication.py:125).  Enabled by
+# default; shapes that change per-microbatch need VESCALE_REUSE_COMM_SHAPE=0.
+_SHAPE_CACHE: dict = {}
+
+
+def _reuse_shapes() -> bool:
+    import os
+
+    return os.environ.get("VESCALE_REUSE_COMM_SHAPE", "1") not in ("0", "false")
+
+
+def reset_shape_cache():
+    _SHAPE_CACHE.clear()
+
+
 def _communicate_shapes(send_t: Optional[torch.Tensor], recv_from: Optional[int],
                         send_to: Optional[int], pg) -> Optional[Tuple]:
     """Exchange (shape, dtype) metadata before variable-shape transfers
-    (reference :125)."""
+    (reference :125).
+
+    Symmetry invariant for the reuse cache: once a (pg, peer, direction)
+    channel has handshaked ONCE, both the sender and the receiver skip all
+    later handshakes on that channel unconditionally — a one-sided skip
+    would orphan a meta send into the peer's payload irecv.  Hence shapes
+    must be static per channel while VESCALE_REUSE_COMM_SHAPE=1 (default;
+    the pipeline emitter sends fixed-shape microbatches).
+    """
+    reuse = _reuse_shapes()
+    rkey = (id(pg), "recv", recv_from)
+    skey = (id(pg), "send", send_to)
+    do_recv = recv_from is not None and not (reuse and rkey in _SHAPE_CACHE)
+    do_send = (
+        send_to is not None and send_t is not None
+        and not (reuse and skey in _SHAPE_CACHE)
+    )
+    dev = _p2p_device(pg)
     ops = []
     recv_buf = None
-    if recv_from is not None:
-        recv_buf = torch.zeros(_MAX_DIMS + 2, dtype=torch.int64)
+    if do_recv:
+        recv_buf = torch.zeros(_MAX_DIMS + 2, dtype=torch.int64, device=dev)
         ops.append(dist.P2POp(dist.irecv, recv_buf, peer=recv_from, group=pg))
-    if send_to is not None and send_t is not None:
-        ops.append(dist.P2POp(dist.isend, _meta_encode(send_t), peer=send_to, group=pg))
+    if do_send:
+        ops.append(dist.P2POp(
+            dist.isend, _meta_encode(send_t, dev), peer=send_to, group=pg))
+        if reuse:
+            _SHAPE_CACHE[skey] = (tuple(send_t.shape), send_t.dtype)
     if ops:
         _run_p2p_ops(ops, pg)
+        drain_send_reqs()  # meta sends are tiny; keep the queue clean
     if recv_buf is not None:
-        return _meta_decode(recv_buf)
+        meta = _meta_decode(recv_buf)
+        if reuse:
+            _SHAPE_CACHE[rkey] = meta
+        return meta
+    if recv_from is not None:
+        return _SHAPE_CACHE.get(rkey)
     return None
 
 

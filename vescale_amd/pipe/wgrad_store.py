@@ -44,8 +44,29 @@ class WeightGradStore:
         return False
 
     def pop_run(self, key: Tuple[int, int]) -> None:
-        for fn in self._by_key.pop(key, []):
-            fn()
+        # W-phase grad GEMMs must not record autograd history (they run in
+        # normal grad mode from the instruction loop, unlike Function.backward)
+        with torch.no_grad():
+            for fn in self._by_key.pop(key, []):
+                fn()
+
+
+def _accum_grad(param: torch.Tensor, g: torch.Tensor) -> None:
+    """Accumulate a deferred W-phase gradient the way autograd would have.
+
+    When the stage is DDP-wrapped, grads must land in the flat main_grad
+    buffer AND register bucket readiness — writing param.grad directly would
+    silently drop the grad from DP reduction.  DDP exposes its param hook as
+    `param._ddp_param_hook` (ddp/distributed_data_parallel.py) for exactly
+    this producer-outside-autograd case.
+    """
+    hook = getattr(param, "_ddp_param_hook", None)
+    if param.grad is None:
+        param.grad = g
+    else:
+        param.grad = param.grad + g
+    if hook is not None:
+        hook(param)  # moves .grad into main_grad, clears it, registers ready
 
 
 class _ZBLinearFn(torch.autograd.Function):
@@ -69,16 +90,9 @@ class _ZBLinearFn(torch.autograd.Function):
             g2 = gy.reshape(-1, gy.shape[-1])
             x2 = x.reshape(-1, x.shape[-1])
             gw = g2.t().matmul(x2)
-            if w_param.grad is None:
-                w_param.grad = gw
-            else:
-                w_param.grad = w_param.grad + gw
+            _accum_grad(w_param, gw)
             if b_param is not None:
-                gb = g2.sum(0)
-                if b_param.grad is None:
-                    b_param.grad = gb
-                else:
-                    b_param.grad = b_param.grad + gb
+                _accum_grad(b_param, g2.sum(0))
 
         if not ctx.store.defer(compute_wgrad):
             compute_wgrad()
