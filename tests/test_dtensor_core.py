@@ -477,3 +477,81 @@ def test_slice_select_backward():
     from tests.common import spawn
 
     spawn(2, _t_slice_select_backward)
+
+
+def _t_amp_found_inf(rank, ws):
+    """AMP grad-scaler inf scan on sharded DTensor grads: an inf on ONE
+    rank's shard must set found_inf on ALL ranks (Partial("max") semantics,
+    reference vescale/dtensor/_dispatch.py:60-117), and finite shards must
+    be unscaled in place."""
+    import torch.distributed as dist
+    from vescale_amd.dtensor import Shard, distribute_tensor, init_device_mesh
+
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("DP",))
+    g1 = distribute_tensor(torch.full((4, 8), 8.0), mesh, [Shard(0)])
+    g2 = distribute_tensor(torch.full((6,), 2.0), mesh, [Shard(0)])
+    if rank == 1:
+        g1._local_tensor[0, 0] = float("inf")
+    found_inf = torch.zeros(())
+    inv_scale = torch.full((), 0.5)
+    torch._amp_foreach_non_finite_check_and_unscale_(
+        [g1, g2], found_inf, inv_scale
+    )
+    assert float(found_inf) == 1.0, f"rank {rank}: found_inf not reduced"
+    assert torch.allclose(g2._local_tensor, torch.full_like(g2._local_tensor, 1.0))
+    if rank == 0:
+        assert torch.allclose(g1._local_tensor, torch.full_like(g1._local_tensor, 4.0))
+
+
+def test_amp_found_inf_reduce():
+    spawn(2, _t_amp_found_inf)
+
+
+def _t_amp_gradscaler_flow(rank, ws):
+    """Full torch.amp.GradScaler flow over a TP-sharded linear: scale ->
+    backward -> unscale_ -> step skipped when an inf is injected on one
+    rank, taken when grads are finite."""
+    import torch.nn as nn
+    from vescale_amd.dmodule import parallelize_module
+    from vescale_amd.dtensor import Replicate, Shard, init_device_mesh
+
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("TP",))
+    torch.manual_seed(0)
+    net = nn.Linear(8, 8)
+    plan = {
+        "parameter": {r"weight": [Shard(0)], r"bias": [Shard(0)]},
+        "forward": {r"": [[Replicate()]], r"output": [[Replicate()]]},
+    }
+    parallelize_module(net, mesh, plan)
+    opt = torch.optim.SGD(net.parameters(), lr=0.1)
+    scaler = torch.amp.GradScaler("cpu", init_scale=4.0)
+
+    # step 1: finite grads -> step taken
+    w_before = net.weight._local_tensor.clone()
+    loss = net(torch.randn(4, 8)).pow(2).mean()
+    scaler.scale(loss).backward()
+    scaler.unscale_(opt)
+    for p in net.parameters():
+        lg = p.grad._local_tensor if hasattr(p.grad, "_local_tensor") else p.grad
+        assert torch.isfinite(lg).all()
+    scaler.step(opt)
+    scaler.update()
+    opt.zero_grad()
+    assert not torch.allclose(net.weight._local_tensor, w_before)
+
+    # step 2: inf on rank 0's shard -> ALL ranks skip the step
+    w_before = net.weight._local_tensor.clone()
+    loss = net(torch.randn(4, 8)).pow(2).mean()
+    scaler.scale(loss).backward()
+    if rank == 0:
+        net.weight.grad._local_tensor[0, 0] = float("inf")
+    scaler.unscale_(opt)
+    scaler.step(opt)
+    scaler.update()
+    assert torch.allclose(net.weight._local_tensor, w_before), (
+        f"rank {rank} took a step on inf grads"
+    )
+
+
+def test_amp_gradscaler_end_to_end():
+    spawn(2, _t_amp_gradscaler_flow)

@@ -222,6 +222,35 @@ def _handler_foreach_norm(dispatcher, op, args, kwargs):
     return outs
 
 
+def _handler_amp_found_inf(dispatcher, op, args, kwargs):
+    """aten._amp_foreach_non_finite_check_and_unscale_: the GradScaler inf
+    scan.  Unscale each local shard in place, then combine found_inf with a
+    MAX all-reduce across every mesh dim on which any grad is non-replicate —
+    an inf seen by ONE rank must poison the step on ALL ranks (reference
+    found_inf_reduce_handler, vescale/dtensor/_dispatch.py:60-117, which
+    wraps found_inf as Partial("max") and redistributes to Replicate)."""
+    from ..dtensor import DTensor
+
+    grads, found_inf, inv_scale = args[0], args[1], args[2]
+    locals_ = [g._local_tensor if isinstance(g, DTensor) else g for g in grads]
+    fi = found_inf._local_tensor if isinstance(found_inf, DTensor) else found_inf
+    sc = inv_scale._local_tensor if isinstance(inv_scale, DTensor) else inv_scale
+    op(locals_, fi, sc)
+    mesh = next((g._spec.mesh for g in grads if isinstance(g, DTensor)), None)
+    if mesh is not None and dist.is_initialized():
+        for md in range(mesh.ndim):
+            if mesh.size(md) <= 1:
+                continue
+            needs = any(
+                isinstance(g, DTensor)
+                and not g._spec.placements[md].is_replicate()
+                for g in grads
+            )
+            if needs:
+                dist.all_reduce(fi, op=dist.ReduceOp.MAX, group=mesh.get_group(md))
+    return None
+
+
 def _handler_fused_adam(dispatcher, op, args, kwargs):
     """aten._fused_adamw_/_fused_adam_/_fused_sgd_: unwrap DTensor lists so
     the fused multi-tensor kernel runs directly on local shards (reference
@@ -398,7 +427,8 @@ def register(dispatcher):
         dispatcher.register_handler(op, _handler_fused_adam)
     if hasattr(aten, "_amp_foreach_non_finite_check_and_unscale_"):
         dispatcher.register_handler(
-            aten._amp_foreach_non_finite_check_and_unscale_, _handler_fused_adam
+            aten._amp_foreach_non_finite_check_and_unscale_,
+            _handler_amp_found_inf,
         )
     dispatcher.register_handler(aten.linalg_vector_norm.default, _handler_vector_norm)
     dispatcher.register_handler(aten.nll_loss_forward.default, _handler_nll_loss_forward)
