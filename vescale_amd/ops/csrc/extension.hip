@@ -337,7 +337,14 @@ at::Tensor gemm_tn8(at::Tensor a, at::Tensor b, int64_t mode) {
               "gemm_tn8 requires M,N % 256 == 0 and K % 64 == 0");
   auto c = at::empty({M, N}, a.options());
   dim3 grid((unsigned)(M / 256), (unsigned)(N / 256), 1);
-  auto kern = mode == 1 ? gemm8_tn_bf16_lockstep : gemm8_tn_bf16;
+  TORCH_CHECK(mode != 5 || (K / 64) % 2 == 0, "mode 5 needs K % 128 == 0");
+  auto kern = mode == 6   ? gemm8_tn_bf16_rot3ra
+              : mode == 5 ? gemm8_tn_bf16_rot3u2
+              : mode == 4 ? gemm8_tn_bf16_rot3np
+              : mode == 3 ? gemm8_tn_bf16_rot3
+              : mode == 2 ? gemm8_tn_bf16_rot
+              : mode == 1 ? gemm8_tn_bf16_lockstep
+                          : gemm8_tn_bf16;
   hipLaunchKernelGGL(kern, grid, dim3(G8_THREADS), 0, cur_stream(),
                      (const unsigned short*)a.data_ptr(),
                      (const unsigned short*)b.data_ptr(),
