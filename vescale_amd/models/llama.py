@@ -19,6 +19,7 @@ import torch.nn.functional as F
 
 from ..ops import (
     build_rope_table,
+    linear as vlinear,
     flash_attention_causal,
     fused_add_rmsnorm,
     fused_cross_entropy,
@@ -27,6 +28,15 @@ from ..ops import (
     swiglu_packed,
 )
 from .llama_tp import TPContext, copy_to_tp, reduce_from_tp
+
+
+class VLinear(nn.Linear):
+    """nn.Linear routed through ops.linear: VESCALE_GEMM=gemm8 runs the
+    in-tree 8-phase CDNA4 MFMA kernel (fwd + dgrad) instead of hipBLASLt
+    (see ops/functional.py header for the measured A/B and default)."""
+
+    def forward(self, x):
+        return vlinear(x, self.weight, self.bias)
 
 
 @dataclass
@@ -96,8 +106,8 @@ class Attention(nn.Module):
         self.hkv = cfg.n_kv_heads // self.tp.world
         d, hd = cfg.dim, cfg.head_dim
         # fused qkv projection: one GEMM instead of three (TP: local heads)
-        self.wqkv = nn.Linear(d, (self.hq + 2 * self.hkv) * hd, bias=False)
-        self.wo = nn.Linear(self.hq * hd, d, bias=False)
+        self.wqkv = VLinear(d, (self.hq + 2 * self.hkv) * hd, bias=False)
+        self.wo = VLinear(self.hq * hd, d, bias=False)
         if self.tp.world > 1:
             # TP metadata for topology-independent (cross-TP reshardable)
             # checkpointing: (dim, global_size, [(local_start, n, global_start)])
@@ -132,8 +142,8 @@ class FeedForward(nn.Module):
         assert cfg.ffn_dim % self.tp.world == 0
         self.ffn_local = cfg.ffn_dim // self.tp.world
         # w1 (gate) and w3 (up) fused into one GEMM (TP: local ffn slice)
-        self.w13 = nn.Linear(cfg.dim, 2 * self.ffn_local, bias=False)
-        self.w2 = nn.Linear(self.ffn_local, cfg.dim, bias=False)
+        self.w13 = VLinear(cfg.dim, 2 * self.ffn_local, bias=False)
+        self.w2 = VLinear(self.ffn_local, cfg.dim, bias=False)
         if self.tp.world > 1:
             fl, r = self.ffn_local, self.tp.rank
             self.w13.weight._tp_shard = (
@@ -179,7 +189,7 @@ class LlamaModel(nn.Module):
             TransformerBlock(cfg, self.tp) for _ in range(cfg.n_layers)
         )
         self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
-        self.output = nn.Linear(cfg.dim, cfg.vocab_size, bias=False)
+        self.output = VLinear(cfg.dim, cfg.vocab_size, bias=False)
         if cfg.tie_embeddings:
             self.output.weight = self.tok_embeddings.weight
         self.register_buffer(

@@ -49,4 +49,5 @@ from .functional import (  # noqa: E402,F401
     adamw_step_flat,
     l2norm_sq,
     scale_flat_,
+    linear,
 )
