@@ -109,6 +109,45 @@ def main():
     for _ in range(args.warmup):
         step()
 
+    # hipGraph the whole train step (fwd+bwd+fused AdamW): the single-GPU
+    # step is ~10% host-launch-bound (round-1 rocprof: GPU busy 90%);
+    # capturing removes per-launch dispatch cost.  Single-rank only for
+    # now — the multi-rank path keeps eager RCCL collectives.  Disable
+    # with VESCALE_GRAPH=0.  Requirements satisfied by construction:
+    # static input (synthetic fixed batch), no dropout, no host syncs in
+    # FlatAdamW (device-side clip scale).
+    graph = None
+    static_loss = None
+    if (
+        on_gpu
+        and world_size == 1
+        and os.environ.get("VESCALE_GRAPH", "1") not in ("0", "false")
+    ):
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                step()  # allocator warmup on the capture stream
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                static_loss = step()
+            g.replay()  # one untimed replay to validate
+            torch.cuda.synchronize()
+            graph = g
+        except Exception as e:  # fall back to eager, report why
+            import sys
+
+            print(f"[bench] hipGraph capture failed ({e}); eager path",
+                  file=sys.stderr)
+            graph = None
+
+    if graph is not None:
+        def step():  # noqa: F811 — replay the captured step
+            graph.replay()
+            return static_loss
+
     if world_size > 1:
         dist.barrier()
     if on_gpu:
@@ -167,6 +206,7 @@ def main():
                 "mfu_est": round(mfu, 4),
                 "mfu_peak_ref": "2.5 PF bf16 dense per MI355X",
                 "activation_checkpointing": args.activation_checkpointing,
+                "hipgraph": graph is not None,
             },
         }
         print(json.dumps(out))
