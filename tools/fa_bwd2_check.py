@@ -14,7 +14,8 @@ def check(B, Hq, Hkv, S, tag):
     dy = torch.randn(B, Hq, S, 128, device="cuda", dtype=torch.bfloat16)
     sc = 1.0 / math.sqrt(128)
     o, lse = C.fa_fwd(q.detach(), k.detach(), v.detach(), sc)
-    dq, dk, dv = C.fa_bwd2(q.detach(), k.detach(), v.detach(), o, dy, lse, sc)
+    fused = os.environ.get("VESCALE_FA_FUSED_DVDK", "1") == "1"
+    dq, dk, dv = C.fa_bwd2(q.detach(), k.detach(), v.detach(), o, dy, lse, sc, fused)
     ref = F.scaled_dot_product_attention(q.float(), k.float().repeat_interleave(Hq//Hkv,1),
                                          v.float().repeat_interleave(Hq//Hkv,1), is_causal=True)
     ref.backward(dy.float())
@@ -44,8 +45,10 @@ def bench(fn, n=10):
     torch.cuda.synchronize(); t = time.perf_counter()
     for _ in range(n): fn()
     torch.cuda.synchronize(); return (time.perf_counter()-t)/n
-t2 = bench(lambda: C.fa_bwd2(q, k, v, o, dy, lse, sc))
-print(f"fa_bwd2: {t2*1e3:.2f} ms")
+t2 = bench(lambda: C.fa_bwd2(q, k, v, o, dy, lse, sc, True))
+print(f"fa_bwd2 fused dvdk: {t2*1e3:.2f} ms")
+t2u = bench(lambda: C.fa_bwd2(q, k, v, o, dy, lse, sc, False))
+print(f"fa_bwd2 split dvdk: {t2u*1e3:.2f} ms")
 t1 = bench(lambda: C.fa_bwd(q, k, v, o, dy, lse, sc))
 print(f"fa_bwd (old): {t1*1e3:.2f} ms")
 # library backward via autograd on aten path

@@ -15,7 +15,7 @@ for i in range(5):
     o1, l1 = C.fa_fwd(q, k, v, sc)
     neq = (o1 != o0).sum().item(); lneq = (l1 != l0).sum().item()
     print(f"fwd run {i}: o diff elems {neq}, lse diff {lneq}")
-g0 = C.fa_bwd2(q, k, v, o0, dy, l0, sc)
+g0 = C.fa_bwd2(q, k, v, o0, dy, l0, sc, True)
 for i in range(3):
-    g1 = C.fa_bwd2(q, k, v, o0, dy, l0, sc)
+    g1 = C.fa_bwd2(q, k, v, o0, dy, l0, sc, True)
     print(f"bwd run {i}: diffs", [ (a != b).sum().item() for a, b in zip(g0, g1) ])

@@ -381,6 +381,200 @@ fa2_dv_bf16(const unsigned short* __restrict__ q,
 }
 
 // ---------------------------------------------------------------------------
+// fa2_dvdk: fused dV + dK.  The two split kernels share the Q/dO staging,
+// the ST = Q.K^T GEMM, the lse/delta loads and a barrier pair per q-tile;
+// fusing removes that duplicated work plus one kernel launch.
+//
+// Geometry: 4 waves / 256 threads, 128-kv-row block tile (HALF the split
+// kernels' 8-wave/256-row blocks).  Reason: a wave needs two fp32
+// accumulator banks (dV, dK: 64 VGPRs each) + st/dp + operands ~ 320 regs.
+// A gfx950 wave can hold up to 512 (256 arch VGPRs + AGPR overflow) but
+// only at 1 wave/SIMD — so an 8-wave block (which forces >= 2 waves/SIMD)
+// pins the budget at 256 and spills 264-484 B/lane to scratch (measured
+// via -Rpass-analysis).  At 4 waves/block the block fits one-per-CU with
+// each wave on its own SIMD and the full 512-reg budget.
+//
+// K and V live in LDS (staged ONCE - the block's own kv-tile never
+// changes), read as B-fragments with fb_bfrag, the same pattern fa2_dq
+// uses for dO.  lse/delta travel as one wave register each (lane i holds
+// row qt0+i) broadcast per-element with ds_bpermute.  st[] is rewritten in
+// place P -> dS between the two MFMA phases so relayout/tr temporaries are
+// shared.  LDS 96 KB/block.  Replaces fa2_dv_bf16 + fa2_dk_bf16 (kept
+// above for A/B).
+// ---------------------------------------------------------------------------
+#define FD_WAVES 4
+#define FD_THREADS (FD_WAVES * 64)
+#define FD_TILE (FD_WAVES * FB_OWN)  // 128 kv rows per block
+
+// stage a 64x128 bf16 tile with 256 threads (fb_stage64 assumes 512)
+DEV void fd_stage64(const unsigned short* __restrict__ g,
+                    unsigned short* l, int tid) {
+#pragma unroll
+  for (int j = 0; j < FB_OTH * FB_D / (FD_THREADS * 8); ++j) {
+    int e = (tid + j * FD_THREADS) * 8;
+    fb_shortx8 v = *reinterpret_cast<const fb_shortx8*>(g + e);
+    *reinterpret_cast<fb_shortx8*>((char*)l + ff_kswz(e * 2)) = v;
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(FD_THREADS, 1)
+fa2_dvdk_bf16(const unsigned short* __restrict__ q,
+              const unsigned short* __restrict__ k,
+              const unsigned short* __restrict__ v,
+              const unsigned short* __restrict__ dout,
+              const float* __restrict__ lse,
+              const float* __restrict__ delta,
+              unsigned short* __restrict__ dv_part,
+              unsigned short* __restrict__ dk_part,
+              int B, int Hq, int Hkv, int S, float scale) {
+  __shared__ unsigned short lq[FB_OTH * FB_D];    // Q q-tile (swz)
+  __shared__ unsigned short ldo[FB_OTH * FB_D];   // dO q-tile (swz)
+  __shared__ unsigned short lk[FD_TILE * FB_D];   // block's K kv rows (swz)
+  __shared__ unsigned short lv[FD_TILE * FB_D];   // block's V kv rows (swz)
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int l31 = lane & 31;
+  const int half = lane >> 5;
+
+  int bx = blockIdx.x, by = blockIdx.y, bz = blockIdx.z;
+  ff_xcd_remap(bx, by, bz);
+  const int kt = bx;
+  const int h = by;
+  const int b = bz;
+  const int hkv = h / (Hq / Hkv);
+  const int64_t qbase = (((int64_t)b * Hq + h) * S) * FB_D;
+  const int64_t kbase = (((int64_t)b * Hkv + hkv) * S) * FB_D;
+  const int kvB = kt * FD_TILE;
+  const int kv0w = kvB + wave * FB_OWN;
+  const int my_kv = kv0w + l31;
+  const int myrow = wave * FB_OWN + l31;  // my kv row within the block tile
+  const float scale2 = scale * FB_LOG2E;
+
+  // stage the block's K/V kv rows once (swizzled, block-cooperative)
+  {
+    const unsigned short* kg = k + kbase + (int64_t)kvB * FB_D;
+    const unsigned short* vg = v + kbase + (int64_t)kvB * FB_D;
+#pragma unroll
+    for (int j = 0; j < FD_TILE * FB_D / (FD_THREADS * 8); ++j) {
+      int e = (tid + j * FD_THREADS) * 8;
+      fb_shortx8 kk = *reinterpret_cast<const fb_shortx8*>(kg + e);
+      *reinterpret_cast<fb_shortx8*>((char*)lk + ff_kswz(e * 2)) = kk;
+      fb_shortx8 vv = *reinterpret_cast<const fb_shortx8*>(vg + e);
+      *reinterpret_cast<fb_shortx8*>((char*)lv + ff_kswz(e * 2)) = vv;
+    }
+  }
+
+  fb_floatx16 dvacc[4], dkacc[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    dvacc[i] = (fb_floatx16)(0.f);
+    dkacc[i] = (fb_floatx16)(0.f);
+  }
+
+  const int jq0 = kvB / FB_OTH;
+  const int n_jq = S / FB_OTH;
+  const float* lseg = lse + ((int64_t)b * Hq + h) * S;
+  const float* dltg = delta + ((int64_t)b * Hq + h) * S;
+
+  for (int jq = jq0; jq < n_jq; ++jq) {
+    const int qt0 = jq * FB_OTH;
+    const bool active = (qt0 + FB_OTH - 1 >= kv0w);
+    const bool need_mask = (qt0 < kv0w + FB_OWN - 1);
+    __syncthreads();  // previous tile readers done; first iter: K/V staged
+    fd_stage64(q + qbase + (int64_t)qt0 * FB_D, lq, tid);
+    fd_stage64(dout + qbase + (int64_t)qt0 * FB_D, ldo, tid);
+    const float lse_reg = lseg[qt0 + lane] * FB_LOG2E;
+    const float dlt_reg = dltg[qt0 + lane];
+    __syncthreads();
+
+    if (active) {
+      // ST = Q.K^T and dP = dO.V^T in one pass — both D[q][kv], col = kv
+      fb_floatx16 st[2], dp[2];
+#pragma unroll
+      for (int s2 = 0; s2 < 2; ++s2) {
+        st[s2] = (fb_floatx16)(0.f);
+        dp[s2] = (fb_floatx16)(0.f);
+      }
+#pragma unroll
+      for (int s2 = 0; s2 < 2; ++s2)
+#pragma unroll
+        for (int ks = 0; ks < 8; ++ks) {
+          fb_shortx8 kfr = fb_bfrag(lk, myrow, ks, half);
+          fb_shortx8 qfr = fb_afrag(lq, s2, ks, l31, half);
+          st[s2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfr, kfr, st[s2], 0, 0, 0);
+          fb_shortx8 vfr = fb_bfrag(lv, myrow, ks, half);
+          fb_shortx8 dofr = fb_afrag(ldo, s2, ks, l31, half);
+          dp[s2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofr, vfr, dp[s2], 0, 0, 0);
+        }
+      // P = exp2(st*scale2 - lse2[q row]) in place (mask q < kv)
+#pragma unroll
+      for (int s2 = 0; s2 < 2; ++s2)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int qrow = s2 * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+          float l2 = __int_as_float(
+              __builtin_amdgcn_ds_bpermute(qrow * 4, __float_as_int(lse_reg)));
+          float x = st[s2][r] * scale2 - l2;
+          if (need_mask && qt0 + qrow < my_kv) x = -INFINITY;
+          st[s2][r] = __builtin_amdgcn_exp2f(x);
+        }
+      // dV phase: dVacc[d][kv] += dO^T . P
+      {
+        fb_shortx8 pb[4];
+        fb_relayout(st, pb, half);  // B-frags [kv][q-contraction]
+#pragma unroll
+        for (int ds = 0; ds < 4; ++ds) {
+          fb_shortx4 t[4][2];
+          FB_TR_READS(ldo, t, ds);
+#pragma unroll
+          for (int ks = 0; ks < 4; ++ks)
+            dvacc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                fb_cat(t[ks][0], t[ks][1]), pb[ks], dvacc[ds], 0, 0, 0);
+        }
+      }
+      // dS = P * (dP - delta[q row]) in place, then dK phase
+#pragma unroll
+      for (int s2 = 0; s2 < 2; ++s2)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int qrow = s2 * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+          float dl = __int_as_float(
+              __builtin_amdgcn_ds_bpermute(qrow * 4, __float_as_int(dlt_reg)));
+          st[s2][r] = st[s2][r] * (dp[s2][r] - dl);
+        }
+      {
+        fb_shortx8 pb[4];
+        fb_relayout(st, pb, half);  // B-frags [kv][q-contraction]
+        // dKacc[d][kv] += Q^T . dS
+#pragma unroll
+        for (int ds = 0; ds < 4; ++ds) {
+          fb_shortx4 t[4][2];
+          FB_TR_READS(lq, t, ds);
+#pragma unroll
+          for (int ks = 0; ks < 4; ++ks)
+            dkacc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                fb_cat(t[ks][0], t[ks][1]), pb[ks], dkacc[ds], 0, 0, 0);
+        }
+      }
+    }
+  }
+
+  // epilogues: dV[kv][d] = acc ; dK[kv][d] = scale * acc
+  unsigned short* ogv = dv_part + qbase + (int64_t)my_kv * FB_D;
+  unsigned short* ogk = dk_part + qbase + (int64_t)my_kv * FB_D;
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int d = ds * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+      ogv[d] = f32_to_bf16(dvacc[ds][r]);
+      ogk[d] = f32_to_bf16(dkacc[ds][r] * scale);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // fa2_dk: dK[kv][d] = scale * sum_q dS Q ; same grid as fa2_dv
 // ---------------------------------------------------------------------------
 extern "C" __global__ void __launch_bounds__(FB_THREADS, 2)

@@ -564,7 +564,7 @@ std::vector<at::Tensor> fa_fwd_ablate(at::Tensor q, at::Tensor k, at::Tensor v,
 // ------------------------------ flash-attention backward v2 ---------------
 std::vector<at::Tensor> fa_bwd2(at::Tensor q, at::Tensor k, at::Tensor v,
                                 at::Tensor o, at::Tensor dout, at::Tensor lse,
-                                double scale) {
+                                double scale, bool fused_dvdk) {
   // 32x32-MFMA backward (attention_bwd2.hip): preprocess + dq + dv + dk
   // (+ GQA reduction). q,o,dout: [B,Hq,S,D]; k,v: [B,Hkv,S,D]; lse ln-dom.
   check_bf16_contig(q, "q");
@@ -597,21 +597,36 @@ std::vector<at::Tensor> fa_bwd2(at::Tensor q, at::Tensor k, at::Tensor v,
                      (float)scale);
   auto dv_part = at::empty({B, Hq, S, D}, q.options());
   auto dk_part = at::empty({B, Hq, S, D}, q.options());
-  hipLaunchKernelGGL(fa2_dv_bf16, grid, dim3(FB_THREADS), 0, cur_stream(),
-                     (const unsigned short*)q.data_ptr(),
-                     (const unsigned short*)k.data_ptr(),
-                     (const unsigned short*)dout_c.data_ptr(),
-                     lse_c.data_ptr<float>(),
-                     (unsigned short*)dv_part.data_ptr(), B, Hq, Hkv, S,
-                     (float)scale);
-  hipLaunchKernelGGL(fa2_dk_bf16, grid, dim3(FB_THREADS), 0, cur_stream(),
-                     (const unsigned short*)q.data_ptr(),
-                     (const unsigned short*)k.data_ptr(),
-                     (const unsigned short*)v.data_ptr(),
-                     (const unsigned short*)dout_c.data_ptr(),
-                     lse_c.data_ptr<float>(), delta.data_ptr<float>(),
-                     (unsigned short*)dk_part.data_ptr(), B, Hq, Hkv, S,
-                     (float)scale);
+  if (fused_dvdk) {
+    // 4-wave blocks over 128-row kv tiles (see kernel comment): 2x the
+    // blocks of the split kernels' 256-row tiles.
+    dim3 gridf(S / FD_TILE, Hq, B);
+    hipLaunchKernelGGL(fa2_dvdk_bf16, gridf, dim3(FD_THREADS), 0, cur_stream(),
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)dout_c.data_ptr(),
+                       lse_c.data_ptr<float>(), delta.data_ptr<float>(),
+                       (unsigned short*)dv_part.data_ptr(),
+                       (unsigned short*)dk_part.data_ptr(), B, Hq, Hkv, S,
+                       (float)scale);
+  } else {
+    hipLaunchKernelGGL(fa2_dv_bf16, grid, dim3(FB_THREADS), 0, cur_stream(),
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)dout_c.data_ptr(),
+                       lse_c.data_ptr<float>(),
+                       (unsigned short*)dv_part.data_ptr(), B, Hq, Hkv, S,
+                       (float)scale);
+    hipLaunchKernelGGL(fa2_dk_bf16, grid, dim3(FB_THREADS), 0, cur_stream(),
+                       (const unsigned short*)q.data_ptr(),
+                       (const unsigned short*)k.data_ptr(),
+                       (const unsigned short*)v.data_ptr(),
+                       (const unsigned short*)dout_c.data_ptr(),
+                       lse_c.data_ptr<float>(), delta.data_ptr<float>(),
+                       (unsigned short*)dk_part.data_ptr(), B, Hq, Hkv, S,
+                       (float)scale);
+  }
   at::Tensor dk, dv;
   if (Hkv == Hq) {
     dk = dk_part;

@@ -387,7 +387,13 @@ class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
-        dq, dk, dv = _ext().fa_bwd2(q, k, v, out, dout, lse, ctx.scale)
+        # default OFF: the fused dv+dk kernel is numerically exact but
+        # measured SLOWER (whole bwd 17.9 ms vs 11.2 ms at B4 H32/8 S8192,
+        # gpurun_out/fa_fused_check.log) — its 2-accumulator register set
+        # only fits at 1 wave/SIMD (4-wave blocks), and the lost latency
+        # hiding outweighs the deduplicated ST GEMM + staging.  Kept for A/B.
+        fused = os.environ.get("VESCALE_FA_FUSED_DVDK", "0") == "1"
+        dq, dk, dv = _ext().fa_bwd2(q, k, v, out, dout, lse, ctx.scale, fused)
         return dq, dk, dv, None
 
 
