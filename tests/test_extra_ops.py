@@ -1,0 +1,154 @@
+"""Extra aten coverage (vescale_amd/dtensor/ops/extra_ops.py): the ops the
+reference's README promises DTensor support for
+(legacy/vescale/dtensor/README.md:56-74) — parity vs single-device eager on
+ws=2/gloo, plus an enumeration test that every README op resolves to a
+registered rule/handler/bypass (or a justified core-table path).
+"""
+import pytest
+import torch
+import torch.nn.functional as F
+
+from tests.common import spawn
+
+from vescale_amd import (
+    Replicate,
+    Shard,
+    distribute_tensor,
+    init_device_mesh,
+)
+
+
+def _t_sort_bucketize(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(5)
+    x = torch.randn(8, 6)
+    rv, ri = torch.sort(x, dim=1)
+    for pl in ([Shard(0)], [Shard(1)], [Replicate()]):
+        dx = distribute_tensor(x, mesh, pl)
+        dv, di = torch.sort(dx, dim=1)
+        assert torch.equal(dv.full_tensor(), rv), f"sort values {pl}"
+        assert torch.equal(di.full_tensor(), ri), f"sort indices {pl}"
+    bounds = torch.tensor([-1.0, 0.0, 1.0])
+    ref = torch.bucketize(x, bounds)
+    dx = distribute_tensor(x, mesh, [Shard(0)])
+    db = distribute_tensor(bounds, mesh, [Replicate()])
+    out = torch.bucketize(dx, db)
+    assert out._spec.placements == (Shard(0),)  # stays sharded: element-local
+    assert torch.equal(out.full_tensor(), ref)
+    ref2 = torch.searchsorted(bounds, x)
+    out2 = torch.searchsorted(db, dx)
+    assert torch.equal(out2.full_tensor(), ref2)
+
+
+def test_sort_bucketize():
+    spawn(2, _t_sort_bucketize)
+
+
+def _t_one_hot(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    idx = torch.tensor([0, 2, 1, 3, 2, 0])
+    ref = F.one_hot(idx, num_classes=4)
+    d = distribute_tensor(idx, mesh, [Shard(0)])
+    out = F.one_hot(d, num_classes=4)
+    assert out._spec.placements == (Shard(0),)
+    assert torch.equal(out.full_tensor(), ref)
+    # defaulted num_classes must use the GLOBAL max (ranks see different maxes)
+    out2 = F.one_hot(d)
+    assert out2.full_tensor().shape[-1] == 4
+    assert torch.equal(out2.full_tensor(), ref)
+
+
+def test_one_hot():
+    spawn(2, _t_one_hot)
+
+
+def _t_index_write(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(7)
+    x = torch.randn(8, 4)
+    idx = torch.tensor([1, 5])
+    src = torch.randn(2, 4)
+    ref = x.clone()
+    ref[idx] = src
+    d = distribute_tensor(x.clone(), mesh, [Shard(0)])
+    di = distribute_tensor(idx, mesh, [Replicate()])
+    ds = distribute_tensor(src, mesh, [Replicate()])
+    d[di] = ds  # index_put_ through the dispatcher
+    assert torch.equal(d.full_tensor(), ref)
+
+    ref2 = x.clone().index_add_(0, idx, src)
+    d2 = distribute_tensor(x.clone(), mesh, [Shard(0)])
+    d2.index_add_(0, di, ds)
+    assert torch.allclose(d2.full_tensor(), ref2, atol=1e-6)
+
+    # out-of-place index_put
+    d3 = distribute_tensor(x, mesh, [Shard(0)])
+    out = torch.index_put(d3, (di,), ds)
+    assert torch.equal(out.full_tensor(), ref)
+
+
+def test_index_write():
+    spawn(2, _t_index_write)
+
+
+def _t_unique_expand(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    x = torch.tensor([3, 1, 2, 3, 1, 0, 2, 2])
+    d = distribute_tensor(x, mesh, [Shard(0)])
+    vals, counts = torch.unique(d, return_counts=True)
+    rv, rc = torch.unique(x, return_counts=True)
+    assert torch.equal(vals.full_tensor(), rv)
+    assert torch.equal(counts.full_tensor(), rc)
+
+    a = torch.randn(1, 4)
+    b = torch.randn(6, 4)
+    da = distribute_tensor(a, mesh, [Replicate()])
+    db = distribute_tensor(b, mesh, [Shard(0)])
+    out = da.expand_as(db)
+    assert torch.equal(out.full_tensor(), a.expand_as(b))
+
+
+def test_unique_expand():
+    spawn(2, _t_unique_expand)
+
+
+def test_readme_op_list_registered():
+    """Every op the reference README enumerates resolves to SOME handling
+    path on our dispatcher: a rule, an eager handler, a bypass, or one of
+    the core strategy tables."""
+    from vescale_amd.dtensor.dispatch import get_dispatcher
+
+    disp = get_dispatcher()
+    aten = torch.ops.aten
+    readme_ops = {
+        "argmax": aten.argmax.default,
+        "argmin": aten.argmin.default,
+        "topk": aten.topk.default,
+        "_unique2": aten._unique2.default,
+        "scatter_src": aten.scatter.src,
+        "scatter_value": aten.scatter.value,
+        "select": aten.select.int,
+        "alias": aten.alias.default,
+        "index_put_": aten.index_put_.default,
+        "index_put": aten.index_put.default,
+        "index_add_": aten.index_add_.default,
+        "sdpa_flash": aten._scaled_dot_product_flash_attention.default,
+        "expand_as": aten.expand_as.default,
+        "one_hot": aten.one_hot.default,
+        "where": aten.where.self,
+        # additional breadth beyond the README
+        "sort": aten.sort.default,
+        "bucketize": aten.bucketize.Tensor,
+        "searchsorted": aten.searchsorted.Tensor,
+    }
+    missing = []
+    for name, op in readme_ops.items():
+        keys = (op, op.overloadpacket)
+        handled = any(
+            k in table
+            for k in keys
+            for table in (disp._rules, disp._handlers, disp._bypass)
+        )
+        if not handled:
+            missing.append(name)
+    assert not missing, f"README ops without a registered path: {missing}"

@@ -7,6 +7,7 @@ def register_all(dispatcher):
         attention,
         conv_ops,
         embedding,
+        extra_ops,
         math_ops,
         matrix,
         pointwise,
@@ -24,3 +25,4 @@ def register_all(dispatcher):
     conv_ops.register(dispatcher)
     random_ops.register(dispatcher)
     attention.register(dispatcher)
+    extra_ops.register(dispatcher)
