@@ -1,5 +1,10 @@
 from .pipe_stage import PipeModule, construct_pipeline_stage
 from .pipe_emmiter import ScheduleEngine
+from .pipe_parser import (
+    hf_symbolic_trace,
+    parse_huggingface_model,
+    split_graph_by_parameters,
+)
 from .p2p_communication import (
     recv_backward,
     recv_forward,
@@ -11,6 +16,9 @@ __all__ = [
     "PipeModule",
     "construct_pipeline_stage",
     "ScheduleEngine",
+    "hf_symbolic_trace",
+    "parse_huggingface_model",
+    "split_graph_by_parameters",
     "send_forward",
     "recv_forward",
     "send_backward",
