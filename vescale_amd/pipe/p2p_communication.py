@@ -111,7 +111,12 @@ def _communicate_shapes(send_t: Optional[torch.Tensor], recv_from: Optional[int]
             _SHAPE_CACHE[skey] = (tuple(send_t.shape), send_t.dtype)
     if ops:
         _run_p2p_ops(ops, pg)
-        drain_send_reqs()  # meta sends are tiny; keep the queue clean
+        # NOTE: no drain here.  Draining inside the handshake turns an async
+        # send stream into a sync point and deadlocks dependency-valid
+        # schedules whose two ranks sit at different instructions (seen with
+        # ZB-V: rank A drains its queued payload sends while rank B waits
+        # for A's next meta).  Meta buffers stay alive on the send queue
+        # until the end-of-schedule drain like any payload.
     if recv_buf is not None:
         meta = _meta_decode(recv_buf)
         if reuse:

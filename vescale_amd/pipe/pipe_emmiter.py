@@ -51,6 +51,22 @@ class ScheduleEngine:
         # interleaved: the chunk chain wraps P-1 -> 0
         self.wrap_prev = self.stage_to_rank(self.P - 1)
         self.wrap_next = self.stage_to_rank(0)
+        # chunk placement topology: "loop" (interleaved wrap) or "v" (ZB-V:
+        # chunk 0 descends ranks, chunk 1 ascends; zbv.py module docstring)
+        self.topology = (
+            "v"
+            if plan.schedule_type == PipelineScheduleType.ZERO_BUBBLE and self.V == 2
+            else "loop"
+        )
+        # ZB-V: one communicator per (direction, chunk) stream.  A rank pair
+        # can carry a forward stream AND a backward-grad stream of identical
+        # shapes concurrently; RCCL/gloo match p2p by posting order per
+        # (src, dst, communicator) — no tags on NCCL — so sharing one group
+        # can cross-match a forward payload into a grad buffer.  Per-stream
+        # groups make every channel single-stream and order-safe.
+        self._zbv_pgs = None
+        if self.topology == "v" and dist.is_initialized():
+            self._zbv_pgs = {k: dist.new_group() for k in ("f0", "f1", "b0", "b1")}
 
         # per-(chunk, mb) state
         self._inputs: Dict[Tuple[int, int], Optional[torch.Tensor]] = {}
@@ -85,23 +101,46 @@ class ScheduleEngine:
         if st == PipelineScheduleType.INTERLEAVED_1F1B:
             return interleaved_1f1b_schedule(self.s, self.P, n_mb, self.V)
         if st == PipelineScheduleType.ZERO_BUBBLE:
-            if self.V != 1:
-                raise ValueError(
-                    "zero-bubble schedule requires virtual_chunks == 1 "
-                    "(ZB-V with virtual chunks is future work)"
-                )
-            from .zero_bubble import zero_bubble_schedule
+            if self.V == 1:
+                from .zero_bubble import zero_bubble_schedule
 
-            return zero_bubble_schedule(self.s, self.P, n_mb)
+                return zero_bubble_schedule(self.s, self.P, n_mb)
+            if self.V == 2:
+                from . import zero_bubble  # noqa: F401 — registers BWD_B/W
+                from .zbv import zbv_schedule
+
+                return zbv_schedule(self.s, self.P, n_mb)
+            raise ValueError(
+                "zero-bubble supports virtual_chunks 1 (ZB-H1) or 2 (ZB-V)"
+            )
         raise ValueError(st)
 
+    # global endpoints depend on the chunk topology --------------------
+    def _is_first_global(self, ck: int) -> bool:
+        return self.s == 0 and ck == 0
+
+    def _is_last_global(self, ck: int) -> bool:
+        if self.topology == "v":
+            return self.s == 0 and ck == 1  # the V ascends back to rank 0
+        return self.s == self.P - 1 and ck == self.V - 1
+
     # peer ranks for a chunk's in/out edges -----------------------------
-    def _in_peer(self, ck: int) -> Optional[int]:
+    # "local" = same-rank chunk handoff (ZB-V rank P-1), handled by the
+    # executor without touching the p2p layer.
+    def _in_peer(self, ck: int):
+        if self.topology == "v":
+            if ck == 0:
+                return self.prev_rank if self.s > 0 else None
+            return "local" if self.s == self.P - 1 else self.stage_to_rank(self.s + 1)
         if self.s == 0:
             return self.wrap_prev if ck > 0 else None
         return self.prev_rank
 
-    def _out_peer(self, ck: int) -> Optional[int]:
+    def _out_peer(self, ck: int):
+        if self.topology == "v":
+            if ck == 0:
+                return "local" if self.s == self.P - 1 else self.next_rank
+            return self.stage_to_rank(self.s - 1) if self.s > 0 else None
         if self.s == self.P - 1:
             return self.wrap_next if ck < self.V - 1 else None
         return self.next_rank
@@ -133,15 +172,24 @@ class ScheduleEngine:
             kind, m, ck = ins.kind, ins.microbatch, ins.chunk
             if kind == "RECV_FWD":
                 with ndtimeit(ndm.RECV_FORWARD):
-                    t = p2p.recv_forward(self._in_peer(ck), self.pg, device=self.device)
+                    t = p2p.recv_forward(
+                        self._in_peer(ck), self._pg_for(kind, ck), device=self.device
+                    )
                 self._inputs[(ck, m)] = t.requires_grad_(True)
             elif kind == "FWD":
                 with ndtimeit(ndm.FORWARD_COMPUTE):
                     self._fwd(ck, m, xs, ys, scale)
             elif kind == "SEND_FWD":
-                p2p.send_forward(
-                    self._outputs[(ck, m)].detach(), self._out_peer(ck), self.pg
-                )
+                peer = self._out_peer(ck)
+                if peer == "local":
+                    # ZB-V rank P-1: chunk 0 output feeds own chunk 1
+                    self._inputs[(ck + 1, m)] = (
+                        self._outputs[(ck, m)].detach().requires_grad_(True)
+                    )
+                else:
+                    p2p.send_forward(
+                        self._outputs[(ck, m)].detach(), peer, self._pg_for(kind, ck)
+                    )
             elif kind == "SEND_FWD_RECV_BWD":
                 g = p2p.send_forward_recv_backward(
                     self._outputs[(ck, m)].detach(), self._out_peer(ck), self.pg,
@@ -150,14 +198,19 @@ class ScheduleEngine:
                 self._recv_grads[(ck, ins.microbatch2)] = g
             elif kind == "RECV_BWD":
                 self._recv_grads[(ck, m)] = p2p.recv_backward(
-                    self._out_peer(ck), self.pg, device=self.device
+                    self._out_peer(ck), self._pg_for(kind, ck), device=self.device
                 )
             elif kind == "BWD":
                 with ndtimeit(ndm.BACKWARD_COMPUTE):
                     self._bwd(ck, m)
             elif kind == "SEND_BWD":
                 g = self._pop_input_grad(ck, m)
-                p2p.send_backward(g, self._in_peer(ck), self.pg)
+                peer = self._in_peer(ck)
+                if peer == "local":
+                    # ZB-V rank P-1: chunk 1 input grad feeds own chunk 0
+                    self._recv_grads[(ck - 1, m)] = g
+                else:
+                    p2p.send_backward(g, peer, self._pg_for(kind, ck))
             elif kind == "SEND_BWD_RECV_FWD":
                 g = self._pop_input_grad(ck, m)
                 t = p2p.send_backward_recv_forward(
@@ -176,8 +229,8 @@ class ScheduleEngine:
 
     # ------------------------------------------------------------------
     def _fwd(self, ck, m, xs, ys, scale):
-        is_first_global = self.s == 0 and ck == 0
-        is_last_global = self.s == self.P - 1 and ck == self.V - 1
+        is_first_global = self._is_first_global(ck)
+        is_last_global = self._is_last_global(ck)
         if is_first_global:
             inp = xs[m].to(self.device)
             self._inputs[(ck, m)] = None  # no upstream grad
@@ -196,12 +249,18 @@ class ScheduleEngine:
 
     def _bwd(self, ck, m):
         out = self._outputs.pop((ck, m))
-        is_last_global = self.s == self.P - 1 and ck == self.V - 1
+        is_last_global = self._is_last_global(ck)
         if is_last_global:
             out.backward()
         else:
             g = self._recv_grads.pop((ck, m))
             torch.autograd.backward(out, grad_tensors=g)
+
+    def _pg_for(self, kind: str, ck: int):
+        if self._zbv_pgs is None:
+            return self.pg
+        key = f"f{ck}" if kind in ("RECV_FWD", "SEND_FWD") else f"b{ck}"
+        return self._zbv_pgs[key]
 
     def _pop_input_grad(self, ck, m):
         inp = self._inputs.pop((ck, m))

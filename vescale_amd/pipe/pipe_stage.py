@@ -15,7 +15,7 @@ import torch
 import torch.distributed as dist
 import torch.nn as nn
 
-from ..plan import PipelineParallelPlan, PipelineSplitMethodType
+from ..plan import PipelineParallelPlan, PipelineScheduleType, PipelineSplitMethodType
 
 
 class PipeModule(nn.Module):
@@ -130,8 +130,18 @@ def construct_pipeline_stage(
         parts = [list(module_list[bounds[i] : bounds[i + 1]]) for i in range(n_parts)]
     else:
         parts = uniform_split(module_list, n_parts)
+    zbv = (
+        plan.schedule_type == PipelineScheduleType.ZERO_BUBBLE
+        and plan.virtual_chunks == 2
+    )
     chunks = []
     for ck in range(plan.virtual_chunks):
-        part = parts[ck * plan.num_stages + stage_id]
+        if zbv:
+            # V placement (zbv.py): chunk 0 descends the ranks, chunk 1
+            # ascends — rank s hosts parts s and 2P-1-s
+            idx = stage_id if ck == 0 else n_parts - 1 - stage_id
+        else:
+            idx = ck * plan.num_stages + stage_id
+        part = parts[idx]
         chunks.append(nn.Sequential(*part) if len(part) != 1 else part[0])
     return PipeModule(chunks, stage_id, plan.num_stages)

@@ -52,7 +52,7 @@ def _zb_init(engine, ins):
 def _bwd_b(engine, ins):
     ck, m = ins.chunk, ins.microbatch
     out = engine._outputs.pop((ck, m))
-    is_last = engine.s == engine.P - 1 and ck == engine.V - 1
+    is_last = engine._is_last_global(ck)
     store: WeightGradStore = engine._wgrad_store
     store.begin()
     try:
