@@ -555,3 +555,54 @@ def _t_amp_gradscaler_flow(rank, ws):
 
 def test_amp_gradscaler_end_to_end():
     spawn(2, _t_amp_gradscaler_flow)
+
+
+# ---------------------------------------------------------------------------
+# (_StridedRaggedShard, Shard) deep composition — 2D FSDPxTP as DTensor
+# placements (reference placement_types.py:228 + docs/texts/raggedshard.md,
+# re-specified clean: SRS composes AFTER the inner Shard, flattening the
+# TP-local chunk; see api.distribute_tensor + the redistribute order_key)
+# ---------------------------------------------------------------------------
+def _t_srs_shard_compose(rank, ws):
+    from vescale_amd.dtensor import _StridedRaggedShard
+
+    mesh = init_device_mesh("cpu", (2, 2), mesh_dim_names=("dp", "tp"))
+    coord = mesh.get_coordinate()
+    dp, tp = coord[0], coord[1]
+    w = torch.arange(48, dtype=torch.float32).reshape(8, 6)
+    srs = _StridedRaggedShard(dims=(0,), local_units=(1, 1), split_factor=2)
+    pl = [srs, Shard(0)]
+
+    d = distribute_tensor(w, mesh, pl)
+    # expected: TP chunk first (Shard(0) over 2 -> 4 rows), then flat halves
+    tp_chunk = w[tp * 4 : (tp + 1) * 4]
+    expect = tp_chunk.reshape(-1)[dp * 12 : (dp + 1) * 12]
+    assert torch.equal(d._local_tensor, expect), (coord, d._local_tensor)
+
+    # full gather back
+    assert torch.equal(d.full_tensor(), w)
+
+    # peel only the ragged layer: (SRS, S) -> (R, S) is the FSDP unshard
+    unshard = d.redistribute(placements=[Replicate(), Shard(0)])
+    assert torch.equal(unshard._local_tensor, tp_chunk)
+
+    # and re-shard: (R, S) -> (SRS, S)
+    reshard = unshard.redistribute(placements=pl)
+    assert torch.equal(reshard._local_tensor, expect)
+
+    # gradient path: (Partial, S) -> (SRS, S) reduces over dp then splits
+    g = distribute_tensor(w, mesh, [Partial(), Shard(0)])
+    gs = g.redistribute(placements=pl)
+    assert torch.equal(gs._local_tensor, expect)
+
+    # uneven units: dp0 holds 1 unit, dp1 holds 2 (of 3) — ragged proper
+    srs_u = _StridedRaggedShard(dims=(0, 1), local_units=(1, 2), split_factor=2)
+    d2 = distribute_tensor(w, mesh, [srs_u, Shard(0)])
+    flat = tp_chunk.reshape(-1)
+    expect2 = flat[:8] if dp == 0 else flat[8:]
+    assert torch.equal(d2._local_tensor, expect2)
+    assert torch.equal(d2.full_tensor(), w)
+
+
+def test_srs_shard_compose():
+    spawn(4, _t_srs_shard_compose)

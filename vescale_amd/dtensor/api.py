@@ -16,6 +16,7 @@ from .device_mesh import DeviceMesh
 from .dtensor import DTensor
 from .placement_types import (
     InterleavedShard,
+    _StridedRaggedShard,
     Partial,
     Placement,
     RaggedShard,
@@ -57,7 +58,16 @@ def distribute_tensor(
     coord = device_mesh.get_coordinate()
     local = tensor
     if coord is not None:
-        for md, p in enumerate(placements):
+        # _StridedRaggedShard composes AFTER every other mesh dim: it ragged-
+        # flattens the already-sharded (e.g. TP-local) chunk, the 2D FSDPxTP
+        # composition (reference placement_types.py:228, re-specified clean:
+        # docs in placement_types._StridedRaggedShard).
+        mds = sorted(
+            range(len(placements)),
+            key=lambda i: (isinstance(placements[i], _StridedRaggedShard), i),
+        )
+        for md in mds:
+            p = placements[md]
             w = device_mesh.size(md)
             if isinstance(p, RaggedShard):
                 local = p.split_tensor(local, w)[coord[md]]

@@ -28,6 +28,7 @@ from . import _collective_utils as cc
 from ._dtensor_spec import DTensorSpec
 from .device_mesh import DeviceMesh
 from .placement_types import (
+    _StridedRaggedShard,
     InterleavedShard,
     Partial,
     Placement,
@@ -78,27 +79,50 @@ def redistribute_local_tensor(
     # UNSAFE while an inner mesh dim (> md) still shards d — gathering or
     # slicing there would interleave chunk-of-chunk layouts.  When no safe
     # transition exists, unshard the innermost conflicting dim first.
-    def touched_dims(p: Placement, q: Placement) -> set:
-        s = set()
-        for x in (p, q):
-            if isinstance(x, RaggedShard):
-                s.update(x.dims)
-            elif isinstance(x, (Shard, InterleavedShard)):
-                s.add(x.dim)
-        return s
+    all_dims = set(range(len(current_spec.shape)))
 
-    def shard_dims(p: Placement) -> set:
+    def pdims(p: Placement) -> set:
+        """Tensor dims a placement's layer entangles.  A strided-ragged
+        local is a flat view of the inner chunk, so it entangles ALL dims."""
+        if isinstance(p, _StridedRaggedShard):
+            return set(all_dims)
         if isinstance(p, RaggedShard):
             return set(p.dims)
         if isinstance(p, (Shard, InterleavedShard)):
             return {p.dim}
         return set()
 
+    def lkey(md: int, p: Placement):
+        # physical application order of a layer: mesh-dim order, except
+        # _StridedRaggedShard which composes AFTER every other mesh dim
+        # (it flattens the inner chunk) — see api.distribute_tensor
+        return (1 if isinstance(p, _StridedRaggedShard) else 0, md)
+
     def is_safe(md: int) -> bool:
-        need = touched_dims(cur[md], tgt[md])
-        for k in range(md + 1, mesh.ndim):
-            if shard_dims(cur[k]) & need:
+        """A transition peels cur[md] then applies tgt[md].  Peeling is
+        outermost-first: no overlapping existing layer may sit above
+        cur[md].  Applying goes above existing layers only (never tuck a
+        new layer under an outer one), and an overlapping pending target
+        layer that is MORE inner must be applied first."""
+        pd = pdims(cur[md])
+        for k in range(mesh.ndim):
+            if k == md:
+                continue
+            if pd and pdims(cur[k]) & pd and lkey(k, cur[k]) > lkey(md, cur[md]):
                 return False
+        ad = pdims(tgt[md])
+        if ad:
+            for k in range(mesh.ndim):
+                if k == md:
+                    continue
+                if pdims(cur[k]) & ad and lkey(k, cur[k]) > lkey(md, tgt[md]):
+                    return False
+                if (
+                    cur[k] != tgt[k]
+                    and pdims(tgt[k]) & ad
+                    and lkey(k, tgt[k]) < lkey(md, tgt[md])
+                ):
+                    return False
         return True
 
     new_local = local
@@ -122,12 +146,15 @@ def redistribute_local_tensor(
             blocked = [md for md in range(mesh.ndim) if cur[md] != tgt[md]]
             victim = None
             for md in blocked:
-                need = touched_dims(cur[md], tgt[md])
-                for k in range(mesh.ndim - 1, md, -1):
-                    if shard_dims(cur[k]) & need:
-                        victim = k
-                        break
-                if victim is not None:
+                need = pdims(cur[md]) | pdims(tgt[md])
+                conflicts = [
+                    k
+                    for k in range(mesh.ndim)
+                    if k != md and pdims(cur[k]) & need
+                ]
+                if conflicts:
+                    # peel the outermost conflicting existing layer
+                    victim = max(conflicts, key=lambda k: lkey(k, cur[k]))
                     break
             assert victim is not None, f"redistribute deadlock: {cur} -> {tgt}"
             new_local = _transition(
