@@ -365,3 +365,57 @@ def test_disable_redistribute_flag():
     from tests.common import spawn
 
     spawn(2, _t_disable_redistribute)
+
+
+def test_ndtimeline_p2p_peer_spans():
+    """ndtimeit_p2p records per-peer spans (reference p2p decorators,
+    p2p_communication.py:624-847): distinct timers per peer, extra carries
+    the peer id, and MetricSummaryHandler splits stats per peer."""
+    import time as _time
+
+    from vescale_amd.ndtimeline import NDTimerManager, ndtimeit_p2p
+    from vescale_amd.ndtimeline.handlers import MetricSummaryHandler
+    from vescale_amd.ndtimeline.timer import NDMetricLevel
+
+    mgr = NDTimerManager(NDMetricLevel.INFO)
+    sh = MetricSummaryHandler()
+    mgr.handlers.append(sh)
+    NDTimerManager.activate(mgr)
+    try:
+        for peer in (1, 2, 1):
+            with ndtimeit_p2p("recv-forward", peer):
+                _time.sleep(0.002)
+        mgr.flush()
+        mgr.wait()
+        by_key = {}
+        for s in mgr.spans:
+            by_key.setdefault((s.metric, s.extra.get("peer")), []).append(s)
+        assert len(by_key[("recv-forward", 1)]) == 2
+        assert len(by_key[("recv-forward", 2)]) == 1
+        assert all(s.dur_us >= 1500 for ss in by_key.values() for s in ss)
+        summ = sh.summary()
+        assert summ["recv-forward|peer=1"]["count"] == 2
+        assert summ["recv-forward|peer=2"]["count"] == 1
+        assert summ["recv-forward|peer=1"]["p99_us"] >= 1500
+    finally:
+        mgr.shutdown()
+        NDTimerManager._instance = None
+
+
+def test_ndtimeline_topology_inference():
+    """calculate_topo maps every rank to its mesh coordinates (reference
+    api.py:359 _calculate_topo)."""
+    import torch
+
+    from vescale_amd.ndtimeline import calculate_topo
+
+    class FakeMesh:
+        mesh = torch.arange(8).reshape(2, 2, 2)
+        mesh_dim_names = ("pp", "dp", "tp")
+
+    t = calculate_topo(FakeMesh())
+    assert t["dims"] == ["pp", "dp", "tp"]
+    assert t["shape"] == [2, 2, 2]
+    assert t["rank_coords"][0] == {"pp": 0, "dp": 0, "tp": 0}
+    assert t["rank_coords"][5] == {"pp": 1, "dp": 0, "tp": 1}
+    assert len(t["rank_coords"]) == 8

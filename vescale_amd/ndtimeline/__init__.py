@@ -1,4 +1,4 @@
-from .api import flush, init_ndtimers, wait
+from .api import attach_topology, calculate_topo, flush, init_ndtimers, wait
 from .timer import (
     DeviceTimer,
     GlobalReferenceTime,
@@ -13,6 +13,8 @@ from . import predefined
 
 __all__ = [
     "init_ndtimers",
+    "calculate_topo",
+    "attach_topology",
     "flush",
     "wait",
     "DeviceTimer",

@@ -7,7 +7,12 @@ from typing import List, Optional
 
 import torch.distributed as dist
 
-from .handlers import ChromeTraceHandler, LocalRawHandler, LoggingHandler
+from .handlers import (
+    ChromeTraceHandler,
+    LocalRawHandler,
+    LoggingHandler,
+    MetricSummaryHandler,
+)
 from .timer import GlobalReferenceTime, NDMetricLevel, NDTimerManager
 
 _chrome: Optional[ChromeTraceHandler] = None
@@ -19,6 +24,7 @@ def init_ndtimers(
     chrome_trace_path: Optional[str] = "ndtimeline_trace.json",
     raw_path: Optional[str] = None,
     log: bool = False,
+    summary: bool = False,
 ) -> NDTimerManager:
     global _chrome
     # VESCALE_NDTIMELINE_LOG_LEVEL (reference env): override the metric
@@ -42,9 +48,50 @@ def init_ndtimers(
         mgr.handlers.append(LocalRawHandler(raw_path))
     if log:
         mgr.handlers.append(LoggingHandler())
+    if summary:
+        mgr.summary_handler = MetricSummaryHandler()
+        mgr.handlers.append(mgr.summary_handler)
     NDTimerManager.activate(mgr)
     atexit.register(lambda: (mgr.shutdown(), _chrome.dump() if _chrome else None))
     return mgr
+
+
+def calculate_topo(mesh) -> dict:
+    """Infer the parallel topology for the timeline view (reference
+    api.py:359 _calculate_topo): per-rank mesh coordinates + dim names so
+    the trace UI can group rank timelines by (pp, dp, tp, ...) role.
+    Accepts our DeviceMesh (or anything with .mesh tensor + dim names)."""
+    import torch
+
+    m = mesh.mesh if hasattr(mesh, "mesh") else torch.as_tensor(mesh)
+    names = list(getattr(mesh, "mesh_dim_names", None) or
+                 [f"dim{i}" for i in range(m.ndim)])
+    topo = {}
+    flat = m.reshape(-1)
+    import itertools
+
+    for coords in itertools.product(*[range(x) for x in m.shape]):
+        rank = int(m[coords])
+        topo[rank] = {names[i]: coords[i] for i in range(m.ndim)}
+    return {"dims": names, "shape": list(m.shape), "rank_coords": topo}
+
+
+def attach_topology(mesh) -> None:
+    """Label the Chrome trace's per-rank process rows with their mesh
+    coordinates (metadata events, ph='M')."""
+    t = calculate_topo(mesh)
+    if _chrome is None:
+        return
+    for rank, coords in t["rank_coords"].items():
+        label = "/".join(f"{k}{v}" for k, v in coords.items())
+        _chrome.events.append(
+            {
+                "name": "process_name",
+                "ph": "M",
+                "pid": f"rank{rank}",
+                "args": {"name": f"rank{rank} [{label}]"},
+            }
+        )
 
 
 def flush(step: Optional[int] = None):

@@ -63,3 +63,38 @@ class LocalRawHandler:
                     )
                     + "\n"
                 )
+
+
+class MetricSummaryHandler:
+    """Streaming per-metric aggregation (reference handlers/local_timer
+    aggregation): count / total / mean / p50 / p99 over span durations,
+    split per peer when spans carry one."""
+
+    def __init__(self):
+        self.durs: Dict[str, List[float]] = {}
+
+    def __call__(self, spans: List[Span]):
+        for s in spans:
+            key = s.metric
+            peer = (s.extra or {}).get("peer")
+            if peer is not None:
+                key = f"{s.metric}|peer={peer}"
+            self.durs.setdefault(key, []).append(s.dur_us)
+
+    @staticmethod
+    def _pct(xs: List[float], q: float) -> float:
+        i = min(len(xs) - 1, max(0, int(round(q * (len(xs) - 1)))))
+        return xs[i]
+
+    def summary(self) -> Dict[str, dict]:
+        out = {}
+        for k, xs in self.durs.items():
+            ys = sorted(xs)
+            out[k] = {
+                "count": len(ys),
+                "total_us": sum(ys),
+                "mean_us": sum(ys) / len(ys),
+                "p50_us": self._pct(ys, 0.5),
+                "p99_us": self._pct(ys, 0.99),
+            }
+        return out

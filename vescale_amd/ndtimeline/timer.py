@@ -77,9 +77,11 @@ class DeviceTimer:
     On GPU, elapsed time is measured with HIP events (async, no host
     sync until flush); on CPU, perf_counter."""
 
-    def __init__(self, metric: str, level: NDMetricLevel = NDMetricLevel.INFO):
+    def __init__(self, metric: str, level: NDMetricLevel = NDMetricLevel.INFO,
+                 extra: Optional[dict] = None):
         self.metric = metric
         self.level = level
+        self.extra = extra or {}
         self._use_cuda = torch.cuda.is_available()
         self._pool: List[Tuple] = []
         self._inflight: List[Tuple] = []
@@ -118,7 +120,7 @@ class DeviceTimer:
                 self._pool.append((ev0, ev1))
             else:
                 _, _, wall, step, dur = rec
-            out.append(Span(self.metric, wall, dur, rank, step))
+            out.append(Span(self.metric, wall, dur, rank, step, dict(self.extra)))
         self._inflight.clear()
         return out
 
@@ -149,11 +151,13 @@ class NDTimerManager:
     def activate(cls, mgr: "NDTimerManager"):
         cls._instance = mgr
 
-    def timer(self, metric: str) -> DeviceTimer:
-        t = self.timers.get(metric)
+    def timer(self, metric: str, *, key: Optional[str] = None,
+              extra: Optional[dict] = None) -> DeviceTimer:
+        k = key or metric
+        t = self.timers.get(k)
         if t is None:
-            t = DeviceTimer(metric)
-            self.timers[metric] = t
+            t = DeviceTimer(metric, extra=extra)
+            self.timers[k] = t
         return t
 
     def _worker(self):
@@ -253,12 +257,25 @@ class ndtimeit_stream(ndtimeit):
         return False
 
 
-def ndtimeit_p2p(metric: str, peer: int):
-    """P2P-op timing with peer annotation (reference timer.py:730)."""
+class ndtimeit_p2p(ndtimeit):
+    """P2P-op timing with per-peer spans (reference p2p ndtimeit_p2p
+    decorators, p2p_communication.py:624-847): each (metric, peer) pair
+    gets its own timer so concurrent streams to different peers never mix
+    event pairs, and every span carries extra={"peer": N} for the
+    timeline/topology view."""
 
-    class _Ctx(ndtimeit):
-        def __exit__(self, *exc):
-            r = super().__exit__(*exc)
-            return r
+    def __init__(self, metric: str, peer):
+        super().__init__(metric)
+        self.peer = peer
 
-    return _Ctx(metric)
+    def __enter__(self):
+        mgr = NDTimerManager.current()
+        self.t = None
+        if mgr and mgr.enabled:
+            self.t = mgr.timer(
+                self.metric,
+                key=f"{self.metric}|peer={self.peer}",
+                extra={"peer": self.peer},
+            )
+            self.t.start()
+        return self

@@ -12,7 +12,7 @@ from typing import Callable, Dict, List, Optional, Sequence, Tuple
 import torch
 import torch.distributed as dist
 
-from ..ndtimeline import ndtimeit, predefined as ndm
+from ..ndtimeline import ndtimeit, ndtimeit_p2p, predefined as ndm
 from ..plan import PipelineParallelPlan, PipelineScheduleType
 from . import p2p_communication as p2p
 from .instruction import (
@@ -171,7 +171,7 @@ class ScheduleEngine:
         for ins in self.build_schedule(n_microbatches):
             kind, m, ck = ins.kind, ins.microbatch, ins.chunk
             if kind == "RECV_FWD":
-                with ndtimeit(ndm.RECV_FORWARD):
+                with ndtimeit_p2p(ndm.RECV_FORWARD, self._in_peer(ck)):
                     t = p2p.recv_forward(
                         self._in_peer(ck), self._pg_for(kind, ck), device=self.device
                     )
@@ -197,9 +197,10 @@ class ScheduleEngine:
                 )
                 self._recv_grads[(ck, ins.microbatch2)] = g
             elif kind == "RECV_BWD":
-                self._recv_grads[(ck, m)] = p2p.recv_backward(
-                    self._out_peer(ck), self._pg_for(kind, ck), device=self.device
-                )
+                with ndtimeit_p2p(ndm.RECV_BACKWARD, self._out_peer(ck)):
+                    self._recv_grads[(ck, m)] = p2p.recv_backward(
+                        self._out_peer(ck), self._pg_for(kind, ck), device=self.device
+                    )
             elif kind == "BWD":
                 with ndtimeit(ndm.BACKWARD_COMPUTE):
                     self._bwd(ck, m)
