@@ -10,6 +10,12 @@ from .calculate_chunk_size import (
     ring_chunk_geometry,
     topo_get_algo_info,
 )
+from .graph_dump import (
+    TopoGraph,
+    parse_graph_dump,
+    predict_time_us,
+    select_algo_proto,
+)
 from .topo import double_binary_trees, ring_order
 from .mesh_collectives import (
     emu_all_gather,
@@ -29,6 +35,10 @@ __all__ = [
     "compute_last_chunk_size",
     "ring_chunk_geometry",
     "double_binary_trees",
+    "TopoGraph",
+    "parse_graph_dump",
+    "predict_time_us",
+    "select_algo_proto",
     "emu_all_gather",
     "emu_reduce_scatter",
     "emu_all_to_all",

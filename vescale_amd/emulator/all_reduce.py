@@ -18,11 +18,17 @@ from .topo import double_binary_trees, ring_order, tree_children
 
 
 def run_ring_all_reduce(
-    buffers: List[torch.Tensor], chunk_bytes: int = 0
+    buffers: List[torch.Tensor], chunk_bytes: int = 0,
+    order: "List[int]" = None,
 ) -> List[torch.Tensor]:
     """Ring allreduce: reduce-scatter pass then all-gather pass.
     buffers[r] is rank r's input; returns the reduced buffers (all equal,
     each element reduced in ring order starting from its chunk owner).
+
+    `order` (from a parsed NCCL_GRAPH_DUMP_FILE ring channel,
+    graph_dump.TopoGraph.ring()) replaces the identity ring with the
+    machine's actual ring order — the reduction order then matches the
+    hardware's bitwise.
 
     chunk_bytes > 0 enables the CHUNKED geometry the real collective uses
     (emulator/calculate_chunk_size.py): the buffer is processed in loops
@@ -67,14 +73,16 @@ def run_ring_all_reduce(
             o += s
         return [segs]
 
+    ring = list(order) if order is not None else ring_order(W)
+    assert sorted(ring) == list(range(W)), f"bad ring order {ring}"
     for segs in loop_segments():
         for c, (o, s) in enumerate(segs):
             if s == 0:
                 continue
-            # chunk c starts at rank (c+1)%W and travels the ring
-            cur = flats[(c + 1) % W][o : o + s].clone()
+            # chunk c starts at ring position c+1 and travels the ring
+            cur = flats[ring[(c + 1) % W]][o : o + s].clone()
             for step in range(1, W):
-                r = (c + 1 + step) % W
+                r = ring[(c + 1 + step) % W]
                 cur = cur + flats[r][o : o + s]
             for r in range(W):
                 acc[r][o : o + s] = cur
