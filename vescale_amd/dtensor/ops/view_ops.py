@@ -434,7 +434,13 @@ def _handler_view(dispatcher, op, args, kwargs):
             g_in = spec.shape[d_in]
             l_in = x._local_tensor.shape[d_in]
             local_target[p.dim] = target[p.dim] * l_in // g_in
-    local = op(x._local_tensor, local_target, *args[2:], **kwargs)
+    try:
+        local = op(x._local_tensor, local_target, *args[2:], **kwargs)
+    except RuntimeError:
+        # view on a non-contiguous local (e.g. HF attention's
+        # transpose().reshape() path): reshape semantics — copy when the
+        # strides don't permit a view
+        local = x._local_tensor.reshape(local_target)
     osp = out_spec(mesh, out_placements, target, spec.dtype)
     return DTensor(local, osp, requires_grad=local.requires_grad)
 
