@@ -77,3 +77,42 @@ def _t_hf_tp_parity(rank, ws):
 def test_hf_llama_tp_parity():
     pytest.importorskip("transformers")
     spawn(2, _t_hf_tp_parity)
+
+
+def _t_hf_mixtral_tp_parity(rank, ws):
+    from transformers import MixtralConfig, MixtralForCausalLM
+
+    from vescale_amd.dmodule import parallelize_module
+    from vescale_amd.dtensor import Replicate, init_device_mesh
+    from vescale_amd.models.hf_mixtral_plan import hf_mixtral_tp_plan
+
+    cfg = MixtralConfig(
+        hidden_size=64,
+        intermediate_size=128,
+        num_hidden_layers=2,
+        num_attention_heads=4,
+        num_key_value_heads=2,
+        vocab_size=256,
+        num_local_experts=4,
+        num_experts_per_tok=2,
+        attn_implementation="eager",
+    )
+    torch.manual_seed(7)
+    ref = MixtralForCausalLM(cfg)
+    ids = torch.randint(0, 256, (2, 8))
+    with torch.no_grad():
+        ref_logits = ref(input_ids=ids).logits
+
+    torch.manual_seed(7)
+    m = MixtralForCausalLM(cfg)
+    mesh = init_device_mesh("cpu", (ws,))
+    m = parallelize_module(m, mesh, hf_mixtral_tp_plan())
+    out = m(input_ids=ids).logits
+    if hasattr(out, "redistribute"):
+        out = out.redistribute(placements=[Replicate()])._local_tensor
+    assert torch.allclose(out, ref_logits, atol=2e-5), (out - ref_logits).abs().max()
+
+
+def test_hf_mixtral_tp_parity():
+    pytest.importorskip("transformers")
+    spawn(2, _t_hf_mixtral_tp_parity)
