@@ -366,3 +366,32 @@ def _t_zbv(rank, ws):
 @pytest.mark.parametrize("ws", [2, 4])
 def test_zbv_accuracy_alignment(ws):
     spawn(ws, _t_zbv)
+
+
+def _t_check_nan(rank, ws):
+    """VESCALE_CHECK_NAN: a NaN injected into a stage's output is caught at
+    the receiving boundary (reference p2p check_nan)."""
+    import os
+
+    import vescale_amd.pipe.p2p_communication as p2p
+
+    os.environ["VESCALE_CHECK_NAN"] = "1"
+    try:
+        if rank == 0:
+            t = torch.ones(4)
+            t[2] = float("nan")
+            p2p.send_forward(t, 1)
+            p2p.drain_send_reqs()
+        else:
+            try:
+                p2p.recv_forward(0)
+            except FloatingPointError as e:
+                assert "non-finite" in str(e)
+            else:
+                raise AssertionError("NaN not detected at the p2p boundary")
+    finally:
+        os.environ.pop("VESCALE_CHECK_NAN", None)
+
+
+def test_p2p_check_nan():
+    spawn(2, _t_check_nan)

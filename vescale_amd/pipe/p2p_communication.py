@@ -242,12 +242,30 @@ def _communicate(
 
 
 # ------------------------- public 8-call API -------------------------------
+def _check_nan(t, what: str):
+    """VESCALE_CHECK_NAN=1: raise on NaN/Inf in p2p payloads (reference
+    p2p_communication.py:252 check_nan) — catches divergence at the stage
+    boundary it crossed instead of steps later in the loss."""
+    import os
+
+    if t is None or not os.environ.get("VESCALE_CHECK_NAN"):
+        return t
+    bad = (~torch.isfinite(t.detach())).sum().item()
+    if bad:
+        rank = dist.get_rank() if dist.is_initialized() else 0
+        raise FloatingPointError(
+            f"[rank {rank}] {what}: {bad} non-finite elements in p2p tensor "
+            f"shape={tuple(t.shape)} dtype={t.dtype}"
+        )
+    return t
+
+
 def recv_forward(prev_rank, pg=None, shape=None, dtype=None, device=None):
     """Receive activations from the previous stage; counterpart of the
     shape handshake in send_forward."""
     t, _ = _communicate(recv_prev=True, prev_rank=prev_rank, pg=pg,
                         recv_shape=shape, recv_dtype=dtype, device=device)
-    return t
+    return _check_nan(t, "recv_forward")
 
 
 def send_forward(t, next_rank, pg=None, handshake=True):
@@ -259,7 +277,7 @@ def send_forward(t, next_rank, pg=None, handshake=True):
 def recv_backward(next_rank, pg=None, shape=None, dtype=None, device=None):
     _, g = _communicate(recv_next=True, next_rank=next_rank, pg=pg,
                         recv_shape=shape, recv_dtype=dtype, device=device)
-    return g
+    return _check_nan(g, "recv_backward")
 
 
 def send_backward(g, prev_rank, pg=None, handshake=True):
