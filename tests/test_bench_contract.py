@@ -31,3 +31,27 @@ def test_bench_torchrun_cpu_smoke():
     assert d["n_gpus"] == 2
     assert d["scaling"] == "weak"
     assert d["value"] > 0
+
+
+def test_bench_torchrun_cpu_8rank_smoke():
+    """8-rank gloo smoke of the EXACT launch shape the driver uses for the
+    round-end SCALE run — tiny model, 1 step."""
+    env = dict(os.environ)
+    env["VESCALE_BENCH_BACKEND"] = "gloo"
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "8",
+            "--master-addr", "127.0.0.1", "--master-port", "29656",
+            "bench.py", "--model", "llama_tiny", "--steps", "1",
+            "--warmup", "1", "--batch", "1", "--seq", "64",
+        ],
+        capture_output=True, text=True, timeout=600, env=env,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 8
+    assert d["config"]["parallelism"] == "fsdp8"
+    assert d["value"] > 0
