@@ -152,3 +152,60 @@ def test_readme_op_list_registered():
         if not handled:
             missing.append(name)
     assert not missing, f"README ops without a registered path: {missing}"
+
+
+def _t_breadth_sweep(rank, ws):
+    """Round-2 breadth sweep ops: factories, reductions with non-Partial
+    combines, scatter_, linear re-dispatch, tail handlers."""
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(3)
+    x = torch.randn(6, 4)
+    d = distribute_tensor(x, mesh, [Shard(0)])
+
+    # new_* factories -> Replicate with explicit global shape
+    nz = d.new_zeros((3, 5))
+    assert nz._spec.placements == (Replicate(),) and tuple(nz.shape) == (3, 5)
+    nf = d.new_full((2, 2), 7.0)
+    assert torch.equal(nf.full_tensor(), torch.full((2, 2), 7.0))
+
+    # prod / all / var / count_nonzero: gather-first reductions
+    assert torch.allclose(d.prod().full_tensor(), x.prod())
+    assert bool((d > -100).all().full_tensor()) is True
+    assert torch.allclose(d.var().full_tensor(), x.var(), atol=1e-6)
+    assert int(torch.count_nonzero(d).full_tensor()) == int(torch.count_nonzero(x))
+
+    # scatter_ with global indices (in-place writeback)
+    ref = x.clone()
+    idx = torch.tensor([[0], [2], [3], [1], [3], [2]])
+    src = torch.randn(6, 1)
+    ref.scatter_(1, idx, src)
+    d2 = distribute_tensor(x.clone(), mesh, [Shard(0)])
+    di = distribute_tensor(idx, mesh, [Replicate()])
+    ds = distribute_tensor(src, mesh, [Replicate()])
+    d2.scatter_(1, di, ds)
+    assert torch.equal(d2.full_tensor(), ref)
+
+    # linear re-dispatches through the matmul rules (colwise TP)
+    w = torch.randn(8, 4)
+    b = torch.randn(8)
+    dw = distribute_tensor(w, mesh, [Shard(0)])
+    db = distribute_tensor(b, mesh, [Shard(0)])
+    out = torch.nn.functional.linear(d.redistribute(placements=[Replicate()]), dw, db)
+    assert torch.allclose(out.full_tensor(), torch.nn.functional.linear(x, w, b), atol=1e-5)
+
+    # tail handlers
+    assert torch.equal(torch.nonzero(d).full_tensor(), torch.nonzero(x))
+    assert torch.equal(d.repeat(2, 1).full_tensor(), x.repeat(2, 1))
+    pad = torch.nn.functional.pad(d, (1, 1), value=0.5)
+    assert torch.equal(pad.full_tensor(), torch.nn.functional.pad(x, (1, 1), value=0.5))
+    assert torch.equal(torch.argsort(d, dim=1).full_tensor(), torch.argsort(x, dim=1))
+
+    # new pointwise registrations smoke (a sample)
+    for fn in (torch.erfc, torch.expm1, torch.log1p, torch.asinh, torch.sinc):
+        val = fn(d.abs() + 0.1)
+        refv = fn(x.abs() + 0.1)
+        assert torch.allclose(val.full_tensor(), refv, atol=1e-6), fn
+
+
+def test_breadth_sweep():
+    spawn(2, _t_breadth_sweep)

@@ -98,6 +98,15 @@ def sdpa_cpu_bwd_rule(schema: OpSchema) -> OutputSharding:
 
 
 def register(dispatcher):
+    if hasattr(aten, "_scaled_dot_product_efficient_attention"):
+        # same sharding semantics as the flash variant (head/batch sharded)
+        dispatcher.register_rule(
+            aten._scaled_dot_product_efficient_attention.default, sdpa_flash_rule
+        )
+        dispatcher.register_rule(
+            aten._scaled_dot_product_efficient_attention_backward.default,
+            sdpa_flash_bwd_rule,
+        )
     if hasattr(aten, "_scaled_dot_product_flash_attention"):
         dispatcher.register_rule(
             aten._scaled_dot_product_flash_attention.default, sdpa_flash_rule

@@ -219,6 +219,28 @@ def _handler_one_hot(dispatcher, op, args, kwargs):
     return DTensor(res, sp, requires_grad=False)
 
 
+def new_factory_rule(schema: OpSchema) -> OutputSharding:
+    """new_zeros/new_ones/new_full/new_empty(_strided): fresh tensor of an
+    explicit GLOBAL shape on the input's mesh — Replicate output (the new
+    shape has no relation to the input's sharding)."""
+    s = schema.specs[0]
+    shape = tuple(schema.args_schema[1]) if len(schema.args_schema) > 1 else ()
+    dtype = schema.kwargs_schema.get("dtype") or s.dtype
+    rep = [Replicate()] * s.mesh.ndim
+    return OutputSharding(out_spec(s.mesh, rep, shape, dtype), None)
+
+
+def _handler_linear(dispatcher, op, args, kwargs):
+    """aten.linear on DTensors: re-dispatch as x @ w.T (+ b) so the matmul
+    rule tables drive sharding (colwise/rowwise TP propagate)."""
+    x, w = args[0], args[1]
+    b = args[2] if len(args) > 2 else None
+    out = x @ w.t()
+    if b is not None:
+        out = out + b
+    return out
+
+
 def register(dispatcher):
     for ov in (aten.sort.default, aten.sort.stable):
         dispatcher.register_rule(ov, sort_rule)
@@ -237,3 +259,30 @@ def register(dispatcher):
         dispatcher.register_handler(ov, _handler_index_write)
     dispatcher.register_handler(aten._unique2.default, _handler_replicate_compute)
     dispatcher.register_handler(aten.expand_as.default, _handler_replicate_compute)
+
+    # breadth sweep ---------------------------------
+    for ov in (aten.new_zeros.default, aten.new_ones.default,
+               aten.new_full.default, aten.new_empty.default,
+               aten.new_empty_strided.default):
+        dispatcher.register_rule(ov, new_factory_rule)
+    # in-place scatter with GLOBAL indices: gather-apply-writeback like
+    # index_put_ (scatter.src/value out-of-place rules live in tensor_ops)
+    for ov in (aten.scatter_.src, aten.scatter_.value):
+        dispatcher.register_handler(ov, _handler_index_write)
+    dispatcher.register_handler(aten.linear.default, _handler_linear)
+    # data-dependent shapes / conservative tail ops: gather to Replicate
+    for ov in (
+        aten.nonzero.default,
+        aten.repeat.default,
+        aten.constant_pad_nd.default,
+        aten.slice_backward.default,
+        aten.select_backward.default,
+        aten.slice_scatter.default,
+        aten.mse_loss.default,
+        aten.mse_loss_backward.default,
+        aten.nll_loss2d_forward.default,
+        aten.nll_loss2d_backward.default,
+        aten.embedding_renorm_.default,
+        aten.argsort.default,
+    ):
+        dispatcher.register_handler(ov, _handler_replicate_compute)

@@ -335,8 +335,30 @@ def cumsum_rule(schema: OpSchema) -> OutputSharding:
     return OutputSharding(osp, [tuple(targets)])
 
 
+def prod_rule(schema: OpSchema) -> OutputSharding:
+    # prod's Partial combine is a PRODUCT, which our Partial does not model:
+    # gather sharded/partial dims instead (breadth op, not a hot path)
+    s = schema.specs[0]
+    rep = tuple(Replicate() for _ in range(s.mesh.ndim))
+    out = _reduction_rule(schema, "sum")
+    out.input_targets = [rep]
+    sp = out.output_spec
+    out.output_spec = out_spec(s.mesh, rep, tuple(sp.shape), sp.dtype)
+    return out
+
+
+def all_rule(schema: OpSchema) -> OutputSharding:
+    # logical-AND reduce: same gather-first treatment as prod
+    return prod_rule(schema)
+
+
 def register(dispatcher):
     dispatcher.register_rule(aten.sum, sum_rule)
+    dispatcher.register_rule(aten.prod, prod_rule)
+    dispatcher.register_rule(aten.all, all_rule)
+    dispatcher.register_rule(aten.any, all_rule)
+    dispatcher.register_rule(aten.var, prod_rule)  # var over a sharded dim is not Partial-combinable: gather first
+    dispatcher.register_rule(aten.count_nonzero, prod_rule)
     dispatcher.register_rule(aten.mean, mean_rule)
     dispatcher.register_rule(aten.amax.default, amax_rule)
     dispatcher.register_rule(aten.amin.default, amin_rule)

@@ -38,6 +38,32 @@ POINTWISE_OPS = [
     aten.leaky_relu_backward, aten.hardtanh_backward, aten.logit_backward,
     # casts / copies with same layout
     aten._to_copy, aten.copy_, aten.to,
+    # breadth sweep (reference registers these too —
+    # all plain element-local ops, same pointwise rule)
+    aten.acos, aten.acos_, aten.acosh, aten.acosh_, aten.asin, aten.asin_,
+    aten.asinh, aten.asinh_, aten.atan_, aten.atan2_, aten.atanh,
+    aten.atanh_, aten.bitwise_left_shift, aten.bitwise_left_shift_,
+    aten.bitwise_not_, aten.bitwise_right_shift, aten.bitwise_right_shift_,
+    aten.bitwise_xor_, aten.clip, aten.clip_, aten.conj_physical,
+    aten.conj_physical_, aten.copysign, aten.copysign_, aten.cosh_,
+    aten.deg2rad, aten.deg2rad_, aten.digamma, aten.digamma_, aten.erfc,
+    aten.erfc_, aten.erfinv, aten.erfinv_, aten.exp2, aten.exp2_,
+    aten.expm1_, aten.float_power, aten.float_power_, aten.floor_divide_,
+    aten.fmod_, aten.frac_, aten.hypot_, aten.i0, aten.i0_, aten.igamma,
+    aten.igamma_, aten.igammac, aten.igammac_, aten.isneginf, aten.isposinf,
+    aten.ldexp, aten.ldexp_, aten.lgamma, aten.lgamma_, aten.log10_,
+    aten.log1p_, aten.log2_, aten.logaddexp, aten.logaddexp2,
+    aten.logical_and_, aten.logical_not_, aten.logical_or_,
+    aten.logical_xor, aten.logical_xor_, aten.logit_, aten.mvlgamma,
+    aten.mvlgamma_, aten.nextafter, aten.nextafter_, aten.positive,
+    aten.rad2deg, aten.rad2deg_, aten.reciprocal_, aten.remainder_,
+    aten.round_, aten.sgn_, aten.sign_, aten.signbit, aten.sinc, aten.sinc_,
+    aten.sinh_, aten.tan_, aten.true_divide, aten.trunc_, aten.xlogy,
+    aten.xlogy_, aten._conj,
+    aten.__lshift__, aten.__rshift__, aten.__ilshift__, aten.__irshift__,
+    # mse loss fwd/bwd are element-local given matching shapes (reduction
+    # handled by the math-ops mean/sum rules when decomposed)
+    aten.mse_loss_backward,
 ]
 
 
