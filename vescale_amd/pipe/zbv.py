@@ -27,10 +27,13 @@ Dependencies simulated (cat in {F, B, W}):
     B(0,s,m): B(0,s+1,m) | B(1,P-1,m) local  [s<P-1 | s==P-1]
   W(c,s,m): B(c,s,m)
 
-Not modeled (honest deviations from the reference's CostGraph): measured
-per-phase costs, comm latency, and the activation-memory bound — uniform
-costs keep the builder deterministic and dependency-exact; the bubble
-structure (W fills idle) is preserved.
+Not modeled (honest deviations from the reference's CostGraph): comm
+latency and the activation-memory bound.  Per-phase costs ARE a
+parameter of build_zbv_timetable (feed measured means); the default unit
+costs keep the builder deterministic and dependency-exact, and the
+bubble structure (W fills idle) is preserved either way.  All ranks must
+build with the SAME costs — the per-(direction, chunk) communicators
+keep each channel single-stream so matching stays order-safe.
 """
 from __future__ import annotations
 
@@ -41,8 +44,15 @@ from .instruction import Instr
 F, B, W = 0, 1, 2
 
 
-def build_zbv_timetable(P: int, n_mb: int) -> List[List[Tuple[int, int, int]]]:
-    """Per-rank ordered op lists [(cat, chunk, mb), ...] for the V topology."""
+def build_zbv_timetable(
+    P: int, n_mb: int, costs: Tuple[float, float, float] = (1.0, 1.0, 1.0)
+) -> List[List[Tuple[int, int, int]]]:
+    """Per-rank ordered op lists [(cat, chunk, mb), ...] for the V topology.
+
+    costs = (f, b, w) phase durations — pass measured per-phase times (e.g.
+    from ndtimeline's MetricSummaryHandler means) to tighten the simulated
+    packing; the default unit costs preserve dependency order exactly and
+    only affect which bubbles W phases land in."""
     end: Dict[Tuple[int, int, int, int], float] = {}
     order: List[List[Tuple[int, int, int]]] = [[] for _ in range(P)]
     cnt = [[0] * 6 for _ in range(P)]  # per-rank next mb, index cat*2+chunk
@@ -88,7 +98,7 @@ def build_zbv_timetable(P: int, n_mb: int) -> List[List[Tuple[int, int, int]]]:
                     best = (key, s, cat, ck, m)
         assert best is not None, "ZBV timetable: no ready op (cyclic deps?)"
         (start, _prio, _s), s, cat, ck, m = best
-        fin = start + 1.0
+        fin = start + costs[cat]
         t[s] = fin
         end[(cat, ck, s, m)] = fin
         cnt[s][cat * 2 + ck] = m + 1
