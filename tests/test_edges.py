@@ -149,3 +149,27 @@ def _t_bypass_precedence(rank, ws):
 
 def test_bypass_precedence():
     spawn(2, _t_bypass_precedence)
+
+
+# ---------------------------------------------------------------------------
+# VeDeviceMesh strategy-coordinate queries (reference devicemesh_api/api.py)
+# ---------------------------------------------------------------------------
+def _t_vedevicemesh_queries(rank, ws):
+    from vescale_amd.devicemesh_api import VESCALE_DEVICE_MESH as V
+
+    V.init_device_mesh("cpu", (2, 2), mesh_dim_names=("DP", "TP"))
+    coord = V.get_strategy_coordinate()
+    assert coord == [rank // 2, rank % 2]
+    assert V.get_strategy_coordinate(3) == [1, 1]
+    assert V.lookup_rank("DP") == rank // 2
+    assert V.lookup_rank(1) == rank % 2
+    assert V.get_strategy_size("TP") == 2 and V.get_strategy_size(0) == 2
+    assert V.shape == (2, 2) and V.size() == 4
+    assert V.get_coordinate() == coord
+    tp_rows = V.get_global_tensor_parallel_meshes()
+    assert tp_rows == [[0, 1], [2, 3]]
+    assert isinstance(V.get_local_rank(), int)
+
+
+def test_vedevicemesh_queries():
+    spawn(4, _t_vedevicemesh_queries)

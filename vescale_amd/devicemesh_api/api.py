@@ -118,6 +118,72 @@ class VeDeviceMesh:
     def is_last_stage(self) -> bool:
         return self.get_pipeline_parallel_rank() == self.get_pipeline_parallel_world_size() - 1
 
+    # strategy-coordinate queries (reference api.py:188-281) --------------
+    def get_strategy_coordinate(self, local_rank: Optional[int] = None):
+        """Coordinate of `local_rank` (default: this rank) over the mesh's
+        strategy dimensions — unlike DeviceMesh.get_coordinate(), any
+        rank's coordinate can be queried (reference api.py:188)."""
+        m = self._require().mesh
+        r = dist.get_rank() if local_rank is None else local_rank
+        return [int(i) for i in (m == r).nonzero(as_tuple=True)]
+
+    def lookup_rank(self, dim) -> int:
+        """This rank's strategy index along `dim` (name or position)."""
+        coord = self.get_strategy_coordinate()
+        if isinstance(dim, str):
+            return coord[self._dim_names.index(dim)]
+        return coord[dim]
+
+    def get_strategy_size(self, dim) -> int:
+        """Size of strategy dimension `dim` (name or position)."""
+        if isinstance(dim, str):
+            return self._require().size(self._dim_names.index(dim))
+        return self._require().size(dim)
+
+    def get_local_rank(self) -> int:
+        """Rank within this machine (node-local device index)."""
+        import torch
+
+        per_node = (
+            torch.cuda.device_count() if torch.cuda.is_available() else 8
+        )
+        return dist.get_rank() % max(per_node, 1)
+
+    def get_coordinate(self):
+        return self._require().get_coordinate()
+
+    def size(self, dim: Optional[int] = None) -> int:
+        return self._require().size(dim)
+
+    @property
+    def shape(self):
+        return self._require().shape
+
+    def __getitem__(self, name: str) -> DeviceMesh:
+        return self._require()[name]
+
+    def get_global_tensor_parallel_meshes(self):
+        """Rank rows of every TP submesh (reference api.py:361): the global
+        view a scheduler uses to enumerate TP groups."""
+        idx = self._dim_index(self.TP_DIM_NAMES)
+        assert idx is not None
+        m = self._require().mesh
+        perm = [d for d in range(m.ndim) if d != idx] + [idx]
+        return [row.tolist() for row in m.permute(perm).reshape(-1, m.size(idx))]
+
+    def get_global_pipeline_parallel_meshes(self):
+        """Rank rows of every PP submesh (reference api.py:333)."""
+        idx = self._dim_index(self.PP_DIM_NAMES)
+        assert idx is not None
+        m = self._require().mesh
+        perm = [d for d in range(m.ndim) if d != idx] + [idx]
+        return [row.tolist() for row in m.permute(perm).reshape(-1, m.size(idx))]
+
+    def _require(self) -> DeviceMesh:
+        mesh = self.get()
+        assert mesh is not None, "Must initialize global DeviceMesh first!"
+        return mesh
+
     # stage peers --------------------------------------------------------
     def get_global_rank_of_stage(self, stage: int) -> int:
         """Global rank holding `stage` at this rank's (dp, tp) coordinate."""
