@@ -28,6 +28,7 @@ from .mem_pool import (  # noqa: F401
     copy_gpu_tensor_to_cpu_pinned_mem_pool,
     release_cpu_tensor,
 )
+from .mem_server import MemCheckpointServer, fetch_checkpoint  # noqa: F401
 from .planner import VeScaleLoadPlanner, VeScaleSavePlanner
 from .ragged_boxes import break_ragged_box
 
@@ -37,6 +38,8 @@ __all__ = [
     "VeScaleSavePlanner",
     "VeScaleLoadPlanner",
     "PinnedStoragePool",
+    "MemCheckpointServer",
+    "fetch_checkpoint",
     "copy_gpu_tensor_to_cpu_pinned_mem_pool",
     "break_ragged_box",
 ]
