@@ -170,3 +170,47 @@ def test_xcd_bijective_remap(nwg):
         assert 0 <= g < nwg
         seen.add(g)
     assert len(seen) == nwg
+
+
+def _t_redistribute_planner_soak(rank, ws):
+    """Property soak of the order-aware redistribute planner (round-2 lkey
+    logic): any (cur -> tgt) pair over {R, S(0), S(1), Partial, SRS∘S}
+    on a 2x2 mesh must terminate and produce value-correct full tensors."""
+    import itertools
+
+    import torch
+
+    from vescale_amd.dtensor import (
+        Partial,
+        Replicate,
+        Shard,
+        _StridedRaggedShard,
+        distribute_tensor,
+        init_device_mesh,
+    )
+
+    mesh = init_device_mesh("cpu", (2, 2))
+    w = torch.arange(64, dtype=torch.float32).reshape(8, 8)
+    srs = _StridedRaggedShard(dims=(0,), local_units=(1, 1), split_factor=2)
+    cands = [
+        [Replicate(), Replicate()],
+        [Replicate(), Shard(0)],
+        [Shard(0), Replicate()],
+        [Shard(0), Shard(1)],
+        [Shard(1), Shard(0)],
+        [srs, Shard(0)],
+        [srs, Shard(1)],
+    ]
+    for cur, tgt in itertools.product(cands, cands):
+        d = distribute_tensor(w, mesh, cur)
+        r = d.redistribute(placements=tgt)
+        assert torch.equal(r.full_tensor(), w), (cur, tgt)
+        # and back again
+        rr = r.redistribute(placements=cur)
+        assert torch.equal(rr.full_tensor(), w), (tgt, cur)
+
+
+def test_redistribute_planner_soak():
+    from tests.common import spawn
+
+    spawn(4, _t_redistribute_planner_soak)

@@ -123,6 +123,16 @@ def redistribute_local_tensor(
                     and lkey(k, tgt[k]) < lkey(md, tgt[md])
                 ):
                     return False
+                if (
+                    cur[k] != tgt[k]
+                    and pdims(cur[k]) & ad
+                    and lkey(k, cur[k]) < lkey(md, tgt[md])
+                ):
+                    # my new layer would sit INNER to an existing outer
+                    # layer that still has to be peeled — peeling it later
+                    # would be blocked by me (planner-loop guard, caught by
+                    # the property soak: [S(0),R] -> [SRS, S(0)])
+                    return False
         return True
 
     new_local = local
@@ -147,14 +157,14 @@ def redistribute_local_tensor(
             victim = None
             for md in blocked:
                 need = pdims(cur[md]) | pdims(tgt[md])
-                conflicts = [
-                    k
-                    for k in range(mesh.ndim)
-                    if k != md and pdims(cur[k]) & need
-                ]
-                if conflicts:
-                    # peel the outermost conflicting existing layer
-                    victim = max(conflicts, key=lambda k: lkey(k, cur[k]))
+                # candidates INCLUDE md itself: when md's own layer is the
+                # outermost (e.g. SRS) the right first move is to peel IT,
+                # not an inner layer beneath it (peeling under a live
+                # strided-ragged flat view scrambles values — caught by the
+                # planner property soak)
+                cands = [k for k in range(mesh.ndim) if pdims(cur[k]) & need]
+                if cands:
+                    victim = max(cands, key=lambda k: lkey(k, cur[k]))
                     break
             assert victim is not None, f"redistribute deadlock: {cur} -> {tgt}"
             new_local = _transition(
