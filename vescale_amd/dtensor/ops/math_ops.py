@@ -348,8 +348,12 @@ def prod_rule(schema: OpSchema) -> OutputSharding:
 
 
 def all_rule(schema: OpSchema) -> OutputSharding:
-    # logical-AND reduce: same gather-first treatment as prod
-    return prod_rule(schema)
+    # logical-AND/OR reduce: same gather-first treatment as prod, but the
+    # result dtype is bool
+    out = prod_rule(schema)
+    sp = out.output_spec
+    out.output_spec = out_spec(sp.mesh, sp.placements, tuple(sp.shape), torch.bool)
+    return out
 
 
 def register(dispatcher):
