@@ -190,7 +190,7 @@ def _t_2d_tp_dp(rank, ws):
             r"fc2.bias": [Replicate()],
         },
         "forward": {
-            r"": [[Replicate()]],
+            "input": [[Replicate()]],
             r"fc2.output": [[Replicate()]],
         },
     }

@@ -27,7 +27,7 @@ MLP_PLAN = {
         r"fc2.weight": [Shard(1)],
     },
     "forward": {
-        r"": [[Replicate()]],          # root input replicate
+        "input": [[Replicate()]],          # root input replicate
         r"fc2.output": [[Replicate()]],  # allreduce partial at the end
     },
 }

@@ -520,7 +520,7 @@ def _t_amp_gradscaler_flow(rank, ws):
     net = nn.Linear(8, 8)
     plan = {
         "parameter": {r"weight": [Shard(0)], r"bias": [Shard(0)]},
-        "forward": {r"": [[Replicate()]], r"output": [[Replicate()]]},
+        "forward": {"input": [[Replicate()]], "output": [[Replicate()]]},
     }
     parallelize_module(net, mesh, plan)
     opt = torch.optim.SGD(net.parameters(), lr=0.1)

@@ -58,6 +58,16 @@ def install_forward_hooks(root: nn.Module, mesh: DeviceMesh, fwd_plan: Dict[str,
             output_plans[strip_suffix(k, "output")] = v
         elif k.endswith(".weight_placement"):
             weight_plans[strip_suffix(k, "weight_placement")] = v
+        elif k in ("input", "output"):
+            # root-module IO: the fqn of the root is "", so the bare
+            # suffix is the natural spelling
+            (input_plans if k == "input" else output_plans)[""] = v
+        else:
+            raise ValueError(
+                f"forward-plan key {k!r} must end in .input / .output / "
+                f".weight_placement (root IO: 'input' / 'output') — a bare "
+                f"module fqn would be silently ignored"
+            )
 
     def match(table, fqn):
         for pattern, v in table.items():

@@ -202,6 +202,19 @@ class _ToLocal(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad_output: torch.Tensor):
         spec = ctx.spec
+        if isinstance(grad_output, DTensor):
+            # mixed plain/DTensor ops can hand a DTensor grad to the plain
+            # side of a to_local boundary; reduce any Partial dims first
+            # (skipping this re-wrapped an UNREDUCED partial and corrupted
+            # every upstream weight grad — caught by the 4D PPxDPxTP test)
+            g = grad_output
+            tgt = tuple(
+                Replicate() if isinstance(p, Partial) else p
+                for p in g._spec.placements
+            )
+            if tgt != g._spec.placements:
+                g = g.redistribute(placements=tgt)
+            grad_output = g._local_tensor
         grad_placements = tuple(
             Replicate() if isinstance(p, Partial) else p for p in spec.placements
         )
