@@ -416,7 +416,7 @@ class _ToLocal(nn.Module):
         return out.to_local() if hasattr(out, "to_local") else out
 
 
-def _t_pp_dp_tp_4d(rank, ws):
+def _t_pp_dp_tp_4d(rank, ws, sched="1f1b"):
     import torch.distributed as dist
     from vescale_amd.ddp import DistributedDataParallel as DDP
     from vescale_amd.dmodule import parallelize_module
@@ -464,7 +464,8 @@ def _t_pp_dp_tp_4d(rank, ws):
     blocks = [_ToLocal(parallelize_module(m, tp_mesh, tp_plan)) for m in mods]
     plan = PipelineParallelPlan(
         num_stages=2,
-        schedule_type=PipelineScheduleType.SIMPLE_1F1B,
+        virtual_chunks=2 if sched == "zero_bubble_v" else 1,
+        schedule_type=PipelineScheduleType(sched),
         split_method=PipelineSplitMethodType.UNIFORM,
     )
     stage = construct_pipeline_stage(blocks, plan, pp_rank)
@@ -490,14 +491,18 @@ def _t_pp_dp_tp_4d(rank, ws):
                 b.inner.finish_grad_sync()
         ddp.finish_grad_sync()
         opt.step()
-        if pp_rank == 1:
+        # loss lands on the last global chunk's rank: pp_rank P-1 for
+        # 1F1B, pp_rank 0 for the V placement
+        loss_pp = 0 if sched == "zero_bubble_v" else 1
+        if pp_rank == loss_pp:
             l = loss.detach().clone() / n_mb
             dist.all_reduce(l, group=dp_group)
             losses.append(float(l) / 2)
-    if pp_rank == 1:
+    if pp_rank == (0 if sched == "zero_bubble_v" else 1):
         for a, b in zip(losses, ref_losses):
             assert abs(a - b) < 1e-4, (losses, ref_losses)
 
 
-def test_pp_dp_tp_4d():
-    spawn(8, _t_pp_dp_tp_4d)
+@pytest.mark.parametrize("sched", ["1f1b", "zero_bubble_v"])
+def test_pp_dp_tp_4d(sched):
+    spawn(8, _t_pp_dp_tp_4d, sched)
