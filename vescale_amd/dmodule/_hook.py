@@ -85,11 +85,20 @@ def install_forward_hooks(root: nn.Module, mesh: DeviceMesh, fwd_plan: Dict[str,
             mod.register_forward_hook(_make_post_hook(op, mesh))
 
 
+def _dynamo_opaque(fn):
+    """DTensor boundaries are opaque to torch.compile (eager-first compat,
+    reference patch #9's scope): the hook's redistribute/from_local runs
+    eagerly via a graph break instead of dynamo tracing DeviceMesh/PG
+    state (whose guards are unpicklable)."""
+    disable = getattr(getattr(torch, "compiler", None), "disable", None)
+    return disable(fn) if disable is not None else fn
+
+
 def _make_pre_hook(plan_list, mesh):
     def hook(mod, args):
         return _convert_seq(args, plan_list, mesh)
 
-    return hook
+    return _dynamo_opaque(hook)
 
 
 def _make_post_hook(plan_list, mesh):
@@ -101,4 +110,4 @@ def _make_post_hook(plan_list, mesh):
             return type(output)(_convert_seq(output, plan_list, mesh))
         return output
 
-    return hook
+    return _dynamo_opaque(hook)

@@ -217,6 +217,15 @@ class DeviceMesh:
     def __repr__(self) -> str:
         return f"DeviceMesh({self.device_type}, {self.mesh.tolist()}, names={self.mesh_dim_names})"
 
+    # a mesh is a process-global handle to live communicators; copying one
+    # is meaningless and deepcopy would die pickling ProcessGroups (dynamo
+    # guard construction deepcopies DTensor specs -> mesh)
+    def __copy__(self):
+        return self
+
+    def __deepcopy__(self, memo):
+        return self
+
 
 def init_device_mesh(
     device_type: str,

@@ -77,6 +77,13 @@ class DTensor(torch.Tensor):
 
         return get_dispatcher().dispatch(func, args, kwargs or {})
 
+    # eager-first torch.compile compat (reference patch #9's scope): dynamo
+    # must never trace INTO DTensor construction — a wrapper-subclass
+    # mid-__new__ breaks __tensor_flatten__-based fakeification.  Opaque
+    # construction = graph break; the compiled region covers the local
+    # compute between DTensor boundaries.  (Applied after class creation
+    # below; no-op when torch.compiler is absent.)
+
     # ------------------------------------------------------------------
     @classmethod
     def _from_local_spec(
@@ -221,3 +228,8 @@ class _ToLocal(torch.autograd.Function):
         tm = TensorMeta(spec.tensor_meta.shape, spec.tensor_meta.stride, grad_output.dtype)
         gspec = DTensorSpec(spec.mesh, grad_placements, tm)
         return DTensor(grad_output, gspec, requires_grad=grad_output.requires_grad)
+
+
+_dynamo_disable = getattr(getattr(torch, "compiler", None), "disable", None)
+if _dynamo_disable is not None:  # torch.compile present: opaque construction
+    DTensor.__new__ = staticmethod(_dynamo_disable(DTensor.__new__))
