@@ -123,3 +123,42 @@ def _t_factory_mode(rank, ws):
 
 def test_factory_mode():
     spawn(2, _t_factory_mode)
+
+
+def _t_obj_return(rank, ws):
+    """DModules returning non-tensor structures (dicts / dataclass-like)
+    pass through the output hooks untouched except tensor leaves
+    (reference dmodule/test_obj_return.py)."""
+    from vescale_amd.dmodule import parallelize_module
+    from vescale_amd.dtensor import init_device_mesh
+
+    class Net(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = nn.Linear(8, 8)
+
+        def forward(self, x):
+            h = self.fc(x)
+            return {"hidden": h, "meta": {"n": 3}, "both": (h, "tag")}
+
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(2)
+    net = Net()
+    torch.manual_seed(2)
+    ref = Net()
+    plan = {
+        "parameter": {r"fc.weight": [Shard(0)], r"fc.bias": [Shard(0)]},
+        "forward": {"input": [[Replicate()]], r"fc.output": [[Replicate()]]},
+    }
+    net = parallelize_module(net, mesh, plan)
+    x = torch.randn(4, 8)
+    out = net(x)
+    assert out["meta"] == {"n": 3}
+    assert out["both"][1] == "tag"
+    h = out["hidden"]
+    h = h.to_local() if hasattr(h, "to_local") else h
+    assert torch.allclose(h, ref(x)["hidden"], atol=1e-6)
+
+
+def test_obj_return():
+    spawn(2, _t_obj_return)

@@ -506,3 +506,61 @@ def _t_pp_dp_tp_4d(rank, ws, sched="1f1b"):
 @pytest.mark.parametrize("sched", ["1f1b", "zero_bubble_v"])
 def test_pp_dp_tp_4d(sched):
     spawn(8, _t_pp_dp_tp_4d, sched)
+
+
+# ---------------------------------------------------------------------------
+# user-defined schedules through the instruction registry (reference
+# instruction/test_userdefine_schedule.py + test_pipe_instruction_register)
+# ---------------------------------------------------------------------------
+def _t_user_schedule(rank, ws):
+    from vescale_amd.engine import PipeEngine
+    from vescale_amd.pipe.instruction import (
+        VESCALE_INSTRUCTION_REGISTRY,
+        Instr,
+        register_instruction,
+    )
+    from vescale_amd.pipe.pipe_emmiter import ScheduleEngine
+    from vescale_amd.pipe.pipe_stage import construct_pipeline_stage
+
+    n_mb, bs, d = 2, 4, 16
+    mods = _make_modules()
+    plan = PipelineParallelPlan(
+        num_stages=ws,
+        schedule_type=PipelineScheduleType.GPIPE,
+        split_method=PipelineSplitMethodType.UNIFORM,
+    )
+    stage = construct_pipeline_stage(mods, plan, rank)
+
+    hits = {"n": 0}
+
+    @register_instruction("USER_MARK")
+    def _user_mark(engine, ins):
+        hits["n"] += 1
+
+    # a custom schedule: GPipe with USER_MARK instructions interleaved
+    orig = ScheduleEngine._build_schedule_impl
+
+    def custom(self, n_microbatches):
+        sched = orig(self, n_microbatches)
+        out = []
+        for ins in sched:
+            out.append(ins)
+            if ins.kind == "FWD":
+                out.append(Instr("USER_MARK", ins.microbatch))
+        return out
+
+    ScheduleEngine._build_schedule_impl = custom
+    try:
+        engine = PipeEngine(stage, plan, loss_fn=_loss_fn, device=torch.device("cpu"))
+        torch.manual_seed(23)
+        x = torch.randn(bs, d)
+        y = torch.randn(bs, d)
+        engine.forward_backward((x, y), n_mb)
+        assert hits["n"] == n_mb, hits  # one mark per FWD on this stage
+    finally:
+        ScheduleEngine._build_schedule_impl = orig
+        VESCALE_INSTRUCTION_REGISTRY.pop("USER_MARK", None)
+
+
+def test_user_defined_schedule():
+    spawn(2, _t_user_schedule)
