@@ -24,6 +24,7 @@ from ..placement_types import (
     Replicate,
     Shard,
     TensorMeta,
+    _StridedRaggedShard,
 )
 from .common import contiguous_stride, out_spec
 
@@ -241,6 +242,47 @@ def _handler_linear(dispatcher, op, args, kwargs):
     return out
 
 
+def _handler_trilu(dispatcher, op, args, kwargs):
+    """tril/triu are POSITION-dependent, not pointwise (caught by the op
+    parity sweep: plain local tril on a Shard(0) input masked as if local
+    row 0 were global row 0).  A shard of the last two dims still computes
+    LOCALLY by shifting the diagonal by the shard's global offset:
+    row-shard r0: keep col <= (r0+i) + d  ==  local diagonal d + r0;
+    col-shard c0: keep (c0+j) - i <= d    ==  local diagonal d - c0.
+    Other layered placements gather first."""
+    from ..dtensor import DTensor
+
+    x = args[0]
+    d = args[1] if len(args) > 1 else kwargs.get("diagonal", 0)
+    if not isinstance(x, DTensor):
+        return op(*args, **kwargs)
+    spec = x._spec
+    nd = len(spec.shape)
+    row_dim, col_dim = nd - 2, nd - 1
+    shift = 0
+    ok_local = True
+    for md, p in enumerate(spec.placements):
+        if isinstance(p, _StridedRaggedShard) or isinstance(p, (RaggedShard, InterleavedShard)):
+            ok_local = False
+        elif isinstance(p, Shard):
+            w = spec.mesh.size(md)
+            my = spec.mesh.get_coordinate()[md]
+            if p.dim % nd == row_dim:
+                shift += Shard.chunk_offset(spec.shape[row_dim], w, my)
+            elif p.dim % nd == col_dim:
+                shift -= Shard.chunk_offset(spec.shape[col_dim], w, my)
+    if not ok_local:
+        return _handler_replicate_compute(dispatcher, op, args, kwargs)
+    res = op(x._local_tensor, d + shift)
+    tm = TensorMeta(spec.shape, spec.tensor_meta.stride, res.dtype)
+    out = DTensor(res, DTensorSpec(spec.mesh, spec.placements, tm),
+                  requires_grad=res.requires_grad)
+    if op._schema.name.endswith("_"):
+        x._local_tensor.copy_(res)
+        return x
+    return out
+
+
 def register(dispatcher):
     for ov in (aten.sort.default, aten.sort.stable):
         dispatcher.register_rule(ov, sort_rule)
@@ -257,6 +299,9 @@ def register(dispatcher):
         aten.index_add_.default,
     ):
         dispatcher.register_handler(ov, _handler_index_write)
+    for ov in (aten.tril.default, aten.triu.default, aten.tril_.default,
+               aten.triu_.default):
+        dispatcher.register_handler(ov, _handler_trilu)
     dispatcher.register_handler(aten._unique2.default, _handler_replicate_compute)
     dispatcher.register_handler(aten.expand_as.default, _handler_replicate_compute)
 

@@ -30,7 +30,7 @@ POINTWISE_OPS = [
     aten.tanh_, aten.threshold, aten.trunc, aten.square, aten.square_,
     aten.logit, aten.polar, aten.hypot, aten.nan_to_num, aten.nan_to_num_,
     aten.masked_fill, aten.masked_fill_, aten.where, aten.fill, aten.fill_,
-    aten.zero_, aten.tril, aten.triu, aten.tril_, aten.triu_,
+    aten.zero_,
     aten.clamp_min, aten.sgn, aten.exponential_,
     # backward pointwise
     aten.gelu_backward, aten.silu_backward, aten.sigmoid_backward,
