@@ -150,3 +150,49 @@ def _t_grad_sweep(rank, ws):
 
 def test_grad_parity_sweep():
     spawn(2, _t_grad_sweep)
+
+
+def _t_ragged_interleaved_sweep(rank, ws):
+    """Elementwise/norm ops on RaggedShard and InterleavedShard inputs
+    (reference test/dtensor/ragged_shard/test_{elementwise,norm}.py and
+    legacy shard/test_interleaved_shard.py)."""
+    from vescale_amd import InterleavedShard, RaggedShard
+
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(17)
+    x = torch.randn(8, 6)
+    cases = [
+        [RaggedShard((0,), (1, 1))],
+        [RaggedShard((0, 1), (1, 3))],   # uneven flat units
+        [InterleavedShard(0, 2)],
+    ]
+    fns = [
+        ("relu", torch.relu),
+        ("scale_add", lambda t: t * 2 + 1),
+        ("sum", lambda t: t.sum()),
+        ("vec_norm", lambda t: torch.linalg.vector_norm(t)),
+        ("sq_mean", lambda t: t.pow(2).mean()),
+    ]
+    fails = []
+    for pl in cases:
+        d = distribute_tensor(x, mesh, pl)
+        for name, fn in fns:
+            try:
+                out = fn(d)
+                full = out.full_tensor() if hasattr(out, "full_tensor") else out
+                ref = fn(x)
+                if not torch.allclose(full, ref, atol=1e-5):
+                    fails.append((name, pl, "value"))
+            except Exception as e:
+                fails.append((name, pl, f"{type(e).__name__}: {str(e)[:120]}"))
+        # round trip
+        try:
+            if not torch.allclose(d.full_tensor(), x):
+                fails.append(("full_tensor", pl, "value"))
+        except Exception as e:
+            fails.append(("full_tensor", pl, f"{type(e).__name__}"))
+    assert not fails, fails
+
+
+def test_ragged_interleaved_sweep():
+    spawn(2, _t_ragged_interleaved_sweep)

@@ -69,9 +69,15 @@ def _reduction_rule(schema: OpSchema, reduce_kind: str) -> OutputSharding:
     for md, p in enumerate(s.placements):
         if isinstance(p, RaggedShard):
             # reducing a ragged tensor over everything -> Partial
-            if all_dims:
-                out_pl.append(Partial("sum" if reduce_kind in ("sum", "mean") else reduce_kind))
+            if all_dims and reduce_kind != "mean":
+                out_pl.append(Partial(reduce_kind))
+            elif all_dims and len(set(p.local_units)) == 1:
+                # equal units: mean of equal-sized local means == global mean
+                # (Partial("sum") here summed the local means — caught by
+                # the ragged op sweep)
+                out_pl.append(Partial("avg"))
             else:
+                # uneven mean (or dim-reduction): no Partial combine exists
                 targets[0][md] = Replicate()
                 out_pl.append(Replicate())
             continue
