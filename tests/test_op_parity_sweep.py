@@ -196,3 +196,47 @@ def _t_ragged_interleaved_sweep(rank, ws):
 
 def test_ragged_interleaved_sweep():
     spawn(2, _t_ragged_interleaved_sweep)
+
+
+def _t_2d_mesh_sweep(rank, ws):
+    """2D (2x2) mesh: ops with placements across BOTH mesh dims, catching
+    cross-dim rule bugs the 1D sweeps cannot."""
+    mesh = init_device_mesh("cpu", (2, 2))
+    torch.manual_seed(19)
+    x = torch.randn(8, 8)
+    pls = [
+        [Shard(0), Shard(1)],
+        [Shard(1), Shard(0)],
+        [Shard(0), Replicate()],
+        [Replicate(), Shard(1)],
+        [Shard(0), Shard(0)],   # chunk-of-chunk on one dim
+    ]
+    fns = [
+        ("relu", torch.relu),
+        ("softmax", lambda t: F.softmax(t, -1)),
+        ("sum_all", lambda t: t.sum()),
+        ("sum_d0", lambda t: t.sum(0)),
+        ("mean_all", lambda t: t.mean()),
+        ("transpose", lambda t: t.transpose(0, 1).contiguous()),
+        ("matmul_self", lambda t: t @ t.t()),
+        ("scale", lambda t: t * 0.5 + 2),
+    ]
+    fails = []
+    for pl in pls:
+        d = distribute_tensor(x, mesh, pl)
+        if not torch.allclose(d.full_tensor(), x):
+            fails.append(("full_tensor", pl, "roundtrip"))
+            continue
+        for name, fn in fns:
+            try:
+                out = fn(d)
+                full = out.full_tensor() if hasattr(out, "full_tensor") else out
+                if not torch.allclose(full, fn(x), atol=1e-5):
+                    fails.append((name, pl, "value"))
+            except Exception as e:
+                fails.append((name, pl, f"{type(e).__name__}: {str(e)[:120]}"))
+    assert not fails, f"{len(fails)}:\n" + "\n".join(map(str, fails[:15]))
+
+
+def test_2d_mesh_sweep():
+    spawn(4, _t_2d_mesh_sweep)
