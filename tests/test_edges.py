@@ -173,3 +173,28 @@ def _t_vedevicemesh_queries(rank, ws):
 
 def test_vedevicemesh_queries():
     spawn(4, _t_vedevicemesh_queries)
+
+
+def _t_cache_bounded(rank, ws):
+    """The dispatch schema caches stay bounded under dynamic shapes
+    (variable-seq inference would otherwise grow them without limit —
+    reference worries about the same hot loop, _dispatch.py:253-258)."""
+    from vescale_amd.dtensor.dispatch import get_dispatcher
+
+    disp = get_dispatcher()
+    disp._cache_cap = 64  # tighten for the test
+    mesh = init_device_mesh("cpu", (ws,))
+    try:
+        for n in range(1, 200):
+            d = distribute_tensor(torch.ones(2, n), mesh, [Shard(0)])
+            _ = (d * 2).full_tensor()
+            assert len(disp._cache) <= 64
+            assert len(disp._fast_cache) <= 4 * 64  # mirrors at most per entry
+    finally:
+        disp._cache_cap = 16384
+        disp._cache.clear()
+        disp._fast_cache.clear()
+
+
+def test_dispatch_cache_bounded():
+    spawn(1, _t_cache_bounded)

@@ -61,6 +61,7 @@ class OpDispatcher:
         self._bypass: Dict[Any, Callable] = {}
         self._cache: Dict[Any, OutputSharding] = {}
         self._fast_cache: Dict[Any, OutputSharding] = {}
+        self._cache_cap = 16384  # bound for dynamic-shape workloads
         self._rng_tracker = None
         self._random_ops = set()
         self._pre_patches: list = []
@@ -242,6 +243,11 @@ class OpDispatcher:
         if sharding is None:
             sharding = self._propagate(schema)
             if cache_key is not None:
+                if len(self._cache) >= self._cache_cap:
+                    # dynamic shapes (e.g. variable-seq inference) would
+                    # grow the schema cache without bound; refill is cheap
+                    self._cache.clear()
+                    self._fast_cache.clear()
                 self._cache[cache_key] = sharding
         # mirror into the fast cache when the signature is flat
         try:
