@@ -240,3 +240,45 @@ def _t_2d_mesh_sweep(rank, ws):
 
 def test_2d_mesh_sweep():
     spawn(4, _t_2d_mesh_sweep)
+
+
+VIEW_FNS = [
+    ("permute", lambda t: t.permute(1, 0).contiguous()),
+    ("movedim", lambda t: t.movedim(0, 1).contiguous()),
+    ("select0", lambda t: t.select(0, 3)),
+    ("select1", lambda t: t.select(1, 5)),
+    ("squeeze_chain", lambda t: t.unsqueeze(0).squeeze(0)),
+    ("view_merge", lambda t: t.reshape(48)),
+    ("view_split", lambda t: t.reshape(6, 2, 4)),
+    ("view_back", lambda t: t.reshape(2, 3, 8).reshape(6, 8)),
+    ("expand_bcast", lambda t: (t.unsqueeze(0).expand(3, 6, 8) + 0).sum(0)),
+    ("cat_d0", lambda t: torch.cat([t, t], 0)),
+    ("cat_d1", lambda t: torch.cat([t, t], 1)),
+    ("slice_step", lambda t: t[::2]),
+    ("flip", lambda t: t.flip(1)),
+    ("roll", lambda t: t.roll(2, dims=1)),
+    ("repeat_interleave", lambda t: t.repeat_interleave(2, dim=1)),
+]
+
+
+def _t_view_sweep(rank, ws):
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(23)
+    x = torch.randn(6, 8)
+    fails = []
+    for name, fn in VIEW_FNS:
+        ref = fn(x)
+        for pl in ([Shard(0)], [Shard(1)], [Replicate()]):
+            d = distribute_tensor(x, mesh, pl)
+            try:
+                out = fn(d)
+                full = out.full_tensor() if hasattr(out, "full_tensor") else out
+                if not torch.allclose(full, ref, atol=1e-5):
+                    fails.append((name, pl, "value"))
+            except Exception as e:
+                fails.append((name, pl, f"{type(e).__name__}: {str(e)[:120]}"))
+    assert not fails, f"{len(fails)}:\n" + "\n".join(map(str, fails[:15]))
+
+
+def test_view_sweep():
+    spawn(2, _t_view_sweep)
