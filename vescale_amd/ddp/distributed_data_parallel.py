@@ -10,7 +10,7 @@ plain modules.
 from __future__ import annotations
 
 from contextlib import contextmanager
-from typing import Dict, List, Optional, Union
+from typing import Dict, List, Union
 
 import torch
 import torch.distributed as dist

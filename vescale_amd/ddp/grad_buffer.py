@@ -8,8 +8,7 @@ latency-bound.
 """
 from __future__ import annotations
 
-import math
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
 import torch.distributed as dist

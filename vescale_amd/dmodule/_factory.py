@@ -6,7 +6,6 @@ DModule.prepare_factory, _dmodule.py:389).
 """
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 from torch.overrides import TorchFunctionMode

@@ -9,7 +9,7 @@ A None placement entry leaves the arg untouched.
 from __future__ import annotations
 
 import re
-from typing import Any, Dict, List, Optional, Sequence
+from typing import Any, Dict
 
 import torch
 import torch.nn as nn

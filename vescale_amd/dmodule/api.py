@@ -18,8 +18,8 @@ from __future__ import annotations
 
 import re
 import os
-from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional, Sequence, Union
+from dataclasses import dataclass
+from typing import Any, Dict, Sequence, Union
 
 import torch
 import torch.nn as nn

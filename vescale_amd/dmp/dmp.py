@@ -7,7 +7,7 @@ PlanGenerator, plan overriding policy).
 from __future__ import annotations
 
 import re
-from typing import Callable, Dict, Optional, Tuple
+from typing import Callable, Dict, Optional
 
 import torch.nn as nn
 

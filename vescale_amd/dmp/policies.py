@@ -7,7 +7,7 @@ Shard(1), LayerNorm SP pass-through, embedding, lm head.
 """
 from __future__ import annotations
 
-from typing import Callable, Dict, List, Optional, Tuple
+from typing import Callable, Dict, List, Tuple
 
 import torch.nn as nn
 

@@ -4,8 +4,8 @@ Parity concept: legacy/vescale/dtensor/op_schema.py + sharding_prop tables.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple, Union
+from dataclasses import dataclass
+from typing import Any, Dict, List, Optional, Sequence, Tuple, Union
 
 import torch
 

@@ -10,7 +10,7 @@ in the PP layout).
 """
 from __future__ import annotations
 
-from typing import Optional, Sequence, Tuple
+from typing import Optional
 
 import torch
 import torch.distributed as dist

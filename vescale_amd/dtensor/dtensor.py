@@ -8,7 +8,7 @@ __torch_dispatch__ -> OpDispatcher (dispatch.py).
 """
 from __future__ import annotations
 
-from typing import Any, Optional, Sequence, Tuple
+from typing import Optional, Sequence, Tuple
 
 import torch
 import torch.distributed as dist

@@ -11,7 +11,6 @@ win: 2 x [N] messages instead of [N, V/w]).
 from __future__ import annotations
 
 import contextlib
-from typing import Optional
 
 import torch
 import torch.distributed as dist

@@ -2,7 +2,7 @@
 dtensor/ops/math_ops.py, vescale/dtensor/_ops/_math_ops.py)."""
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import List, Optional
 
 import torch
 

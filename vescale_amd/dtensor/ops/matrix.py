@@ -9,7 +9,7 @@ DModule hooks then place.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 

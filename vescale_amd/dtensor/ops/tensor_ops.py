@@ -5,7 +5,6 @@ the fused-optimizer unwrap at vescale/dtensor/_dispatch.py:118,269-271.
 """
 from __future__ import annotations
 
-from typing import List
 
 import torch
 import torch.distributed as dist

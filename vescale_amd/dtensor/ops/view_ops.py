@@ -2,7 +2,7 @@
 vescale_view_ops.py, tensor_ops.py slice/cat/stack sections)."""
 from __future__ import annotations
 
-from typing import List, Optional, Sequence, Tuple
+from typing import List, Optional, Sequence
 
 import torch
 

@@ -20,7 +20,7 @@ rather than a port.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence, Tuple
+from typing import List, Sequence
 
 import torch
 

@@ -6,7 +6,7 @@ torch.distributed calls to loop over ranks).
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 import torch
 

@@ -5,7 +5,7 @@ forward_backward, parameters, sync_shared_params).
 """
 from __future__ import annotations
 
-from typing import Callable, List, Optional, Sequence, Tuple
+from typing import Callable, Optional, Tuple
 
 import torch
 import torch.distributed as dist

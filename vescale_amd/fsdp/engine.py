@@ -25,7 +25,7 @@ open-sourced, so this engine is a from-scratch design):
 from __future__ import annotations
 
 import logging
-from typing import Callable, Dict, List, Optional, Sequence, Tuple, Type, Union
+from typing import Dict, List, Optional, Sequence, Tuple
 
 import torch
 import torch.distributed as dist

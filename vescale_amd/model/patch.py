@@ -8,7 +8,6 @@ vocab-sharded Shard(0).
 """
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 import torch.distributed as dist

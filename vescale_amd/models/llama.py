@@ -9,7 +9,6 @@ Parity role: reference examples' HF Llama2 / open_llama configs
 """
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 from typing import Optional
 

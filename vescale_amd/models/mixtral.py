@@ -7,7 +7,7 @@ vescale_amd.moe.parallelize_experts (token all-to-all over xGMI).
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.distributed as dist

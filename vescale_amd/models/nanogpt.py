@@ -6,7 +6,6 @@ this is the TP/SP (DModule) test vehicle, as in the reference.
 """
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 from typing import Optional
 

@@ -6,7 +6,7 @@ natural fit: tokens fan out over all 7 p2p links concurrently.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import Optional, Sequence
 
 import torch
 import torch.distributed as dist

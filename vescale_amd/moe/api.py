@@ -7,7 +7,6 @@ per-(rank, expert) counts exchanged on a small int tensor first.
 """
 from __future__ import annotations
 
-import re
 from typing import Optional
 
 import torch

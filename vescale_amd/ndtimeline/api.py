@@ -3,7 +3,7 @@ init_ndtimers / flush / wait)."""
 from __future__ import annotations
 
 import atexit
-from typing import List, Optional
+from typing import Optional
 
 import torch.distributed as dist
 

@@ -3,7 +3,7 @@ implementations used (a) on CPU boxes for tests, (b) as the fp32 numerics
 oracle the GPU tests compare against (SURVEY.md §4 test strategy)."""
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import os
 

@@ -4,7 +4,7 @@ Parity: legacy/vescale/optim/base_optimizer.py:116.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence, Union
+from typing import List, Sequence, Union
 
 import torch
 import torch.nn as nn

@@ -6,7 +6,7 @@ groups, in-place scale.
 """
 from __future__ import annotations
 
-from typing import Iterable, List, Optional
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist

@@ -7,8 +7,8 @@ registry executed by the ScheduleEngine.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
-from typing import Callable, Dict, List, Optional
+from dataclasses import dataclass
+from typing import Callable, Dict, List
 
 VESCALE_INSTRUCTION_REGISTRY: Dict[str, Callable] = {}
 

@@ -11,7 +11,7 @@ order — batch_isend_irecv guarantees this by fusing into one RCCL group.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence, Tuple
+from typing import List, Optional, Tuple
 
 import torch
 import torch.distributed as dist

@@ -8,7 +8,7 @@ models that are not a flat module list.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence, Type
+from typing import Dict, List, Optional, Sequence
 
 import torch
 import torch.fx as fx

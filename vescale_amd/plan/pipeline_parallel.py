@@ -3,7 +3,7 @@ legacy/vescale/plan/pipeline_parallel.py:28)."""
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Any, Dict, List, Optional, Sequence
+from typing import Any, List, Optional, Sequence
 
 from .spec import (
     ModeType,
