@@ -317,11 +317,11 @@ def test_zbv_timetable_properties():
         assert first_w < last_b, "no W/B interleaving: schedule is GPipe-like"
 
 
-def _t_zbv(rank, ws):
+def _t_zbv(rank, ws, n_mb=4):
     from vescale_amd.engine import PipeEngine
     from vescale_amd.pipe.pipe_stage import construct_pipeline_stage
 
-    n_mb, bs, d = 4, 8, 16
+    bs, d = 8, 16
     mods = _make_modules()
     plan = PipelineParallelPlan(
         num_stages=ws,
@@ -363,9 +363,10 @@ def _t_zbv(rank, ws):
             )
 
 
-@pytest.mark.parametrize("ws", [2, 4])
-def test_zbv_accuracy_alignment(ws):
-    spawn(ws, _t_zbv)
+@pytest.mark.parametrize("ws,n_mb", [(2, 4), (4, 4), (2, 1), (4, 2)])
+def test_zbv_accuracy_alignment(ws, n_mb):
+    # (2,1) and (4,2): fewer microbatches than stages — the bubbliest edge
+    spawn(ws, _t_zbv, n_mb)
 
 
 def _t_check_nan(rank, ws):
