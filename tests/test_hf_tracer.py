@@ -142,6 +142,38 @@ def test_hf_split_backward_flows():
     _run_in_subprocess("_check_backward_flows")
 
 
+def _check_attention_mask_input():
+    import torch
+
+    from vescale_amd.pipe import hf_symbolic_trace, parse_huggingface_model
+
+    m = _mk_model()
+    ids = torch.randint(0, 256, (2, 8))
+    mask = torch.ones(2, 8, dtype=torch.long)
+    mask[:, :2] = 0  # left padding
+    ref = m(input_ids=ids, attention_mask=mask).logits
+    gm = hf_symbolic_trace(m, input_names=("input_ids", "attention_mask"))
+    out = gm(ids, mask)
+    out = out[0] if isinstance(out, (tuple, list)) else out
+    assert torch.allclose(out, ref, atol=1e-6)
+    # regression: the config flowing into the mask builder must be the REAL
+    # config object (a GraphModule get_attr copy loses _attn_implementation
+    # and the builder silently returned no mask)
+    stages = parse_huggingface_model(m, 2, input_names=("input_ids", "attention_mask"))
+    cur = (ids, mask)
+    for s in stages:
+        cur = s(*cur)
+        if not isinstance(cur, (tuple, list)):
+            cur = (cur,)
+    assert torch.allclose(cur[0], ref, atol=1e-6)
+    print("CHECK_OK")
+
+
+def test_hf_attention_mask_input():
+    pytest.importorskip("transformers")
+    _run_in_subprocess("_check_attention_mask_input")
+
+
 def test_trace_leaves_model_unpatched():
     pytest.importorskip("transformers")
     _run_in_subprocess("_check_patches_restored")

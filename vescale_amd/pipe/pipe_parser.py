@@ -140,16 +140,47 @@ class _PatchedGraphLeaves:
 
     @staticmethod
     def _wrap(orig):
+        import functools
+
         from torch.fx.proxy import Proxy
 
         def wrapped(*args, **kwargs):
             flat = list(args) + list(kwargs.values())
             prox = [a for a in flat if isinstance(a, Proxy)]
-            if prox:
-                return prox[0].tracer.create_proxy(
-                    "call_function", orig, args, kwargs
-                )
-            return orig(*args, **kwargs)
+            if not prox:
+                return orig(*args, **kwargs)
+            tracer = prox[0].tracer
+
+            def resolve(a):
+                """Non-tensor attribute proxies (e.g. `self.config`) are
+                bound by VALUE into the recorded call: GraphModule's
+                get_attr copy of a config object loses private state
+                (transformers' _attn_implementation came back None and the
+                mask builder silently returned no mask)."""
+                if isinstance(a, Proxy) and a.node.op == "get_attr":
+                    obj = tracer.root
+                    try:
+                        for part in str(a.node.target).split("."):
+                            obj = getattr(obj, part)
+                    except AttributeError:
+                        return a
+                    if not isinstance(obj, torch.Tensor):
+                        return obj
+                return a
+
+            args = tuple(resolve(a) for a in args)
+            kwargs = {k: resolve(v) for k, v in kwargs.items()}
+            const_kwargs = {
+                k: v for k, v in kwargs.items() if not isinstance(v, Proxy)
+                and not isinstance(v, torch.Tensor)
+            }
+            live_kwargs = {k: v for k, v in kwargs.items() if k not in const_kwargs}
+            if const_kwargs:
+                target = functools.partial(orig, **const_kwargs)
+                functools.update_wrapper(target, orig)
+            else:
+                target = orig
+            return tracer.create_proxy("call_function", target, args, live_kwargs)
 
         wrapped.__name__ = getattr(orig, "__name__", "wrapped")
         return wrapped
