@@ -14,7 +14,7 @@ import torch
 from tests.common import spawn
 
 
-def _t_hf_tp_parity(rank, ws):
+def _t_hf_tp_parity(rank, ws, attn="eager"):
     from transformers import LlamaConfig, LlamaForCausalLM
 
     from vescale_amd.dmodule import parallelize_module
@@ -28,7 +28,7 @@ def _t_hf_tp_parity(rank, ws):
         num_attention_heads=4,
         num_key_value_heads=2,
         vocab_size=256,
-        attn_implementation="eager",
+        attn_implementation=attn,
     )
     torch.manual_seed(7)
     ref = LlamaForCausalLM(cfg)
@@ -74,9 +74,11 @@ def _t_hf_tp_parity(rank, ws):
     assert torch.allclose(qg, rqg, atol=5e-5), (qg - rqg).abs().max()
 
 
-def test_hf_llama_tp_parity():
+@pytest.mark.parametrize("attn", ["eager", "sdpa"])
+def test_hf_llama_tp_parity(attn):
+    # sdpa exercises the head-sharded _scaled_dot_product rules end-to-end
     pytest.importorskip("transformers")
-    spawn(2, _t_hf_tp_parity)
+    spawn(2, _t_hf_tp_parity, attn)
 
 
 def _t_hf_mixtral_tp_parity(rank, ws):
