@@ -458,3 +458,193 @@ gemm8_tn_bf16_rot3ra(const unsigned short* __restrict__ A,
                      unsigned short* __restrict__ C, int M, int N, int K) {
   g8_body_rot<3, false, false, true>(A, B, C, M, N, K);
 }
+
+
+// ---------------------------------------------------------------------------
+// rot8: 8-phase / 2-K-tile iteration with a 1-slot-per-phase rotation and
+// TWO counted waits per iteration (vs rot3's two per tile).
+// MEASURED: correct + race-stable but SLOWER than rot3np (917-1010 TF vs
+// 1090-1208 on model shapes): halving the wait COUNT doesn't pay because
+// each wait here is nearly a full drain (vmcnt(2) — FIFO forces it, the
+// needed slots are only 2-4 positions old).  Going deeper needs the
+// needed slots pushed earlier, which the 2-dbuf region lifetimes forbid
+// within the 128 KiB LDS budget.  Kept as mode 7 for A/B evidence.
+//
+// Steady-state slot table during tiles (T, T+1) (T = 2i), phases p0..p7
+// (p0-3 compute tile T from buf0, p4-7 tile T+1 from buf1):
+//   p0: A(T+1)q1 -> buf1   [A(T-1)q1 died prev p7; read p6,p7]
+//   p1: B(T+1)h0 -> buf1   [B(T-1) died prev p7;   read p4..p7]
+//   p2: B(T+1)h1 -> buf1
+//   p3: A(T+2)q0 -> buf0   [A(T)q0 died p1;  read next p0,p1]
+//   p4: A(T+2)q1 -> buf0   [A(T)q1 died p3;  read next p2,p3]
+//   p5: B(T+2)h0 -> buf0   [B(T) died p3;    read next p0..p3]
+//   p6: B(T+2)h1 -> buf0
+//   p7: A(T+3)q0 -> buf1   [A(T+1)q0 died p5; read next p4,p5]
+// Waits (FIFO vmcnt; slots issue 2 global_load_lds each):
+//   p0 start: vmcnt(2) — newest allowed = prev p7's slot; guarantees
+//     everything through prev p6 landed: tile T fully staged (A q0 prev
+//     p3, A q1 prev p4, B prev p5/p6).
+//   p4 start: vmcnt(2) — newest allowed = p3's slot; guarantees p0..p2
+//     landed: tile T+1 fully staged (A q0 prev p7, A q1 p0, B p1/p2).
+// Each slot thus has >= 4 phases (~64 MFMAs) of latency cover before its
+// wait, with up to 5 slots in flight mid-iteration.  Requires ntile even
+// and >= 4 (host falls back to rot3np otherwise).
+// ---------------------------------------------------------------------------
+template <int SWZ>
+DEV void g8_body_rot8(const unsigned short* __restrict__ A,
+                      const unsigned short* __restrict__ B,
+                      unsigned short* __restrict__ C, int M, int N, int K) {
+  __shared__ unsigned short lA[2][2][128 * G8_BK];
+  __shared__ unsigned short lB[2][2][128 * G8_BK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+  const int bh = wn >> 1;
+
+  int gx = gridDim.x, nwg = gx * (int)gridDim.y;
+  int f = blockIdx.x + gx * blockIdx.y;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = f & 7, off = f >> 3;
+    f = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  const int bm = (f % gx) * G8_BM;
+  const int bn = (f / gx) * G8_BN;
+  const unsigned short* Ag = A + (int64_t)bm * K;
+  const unsigned short* Bg = B + (int64_t)bn * K;
+
+  g8_floatx4 acc[G8_MFRAG][G8_NFRAG];
+#pragma unroll
+  for (int i = 0; i < G8_MFRAG; ++i)
+#pragma unroll
+    for (int j = 0; j < G8_NFRAG; ++j) acc[i][j] = (g8_floatx4)(0.f);
+
+  const int ntile = K / G8_BK;
+
+  // prologue: tiles 0 (buf0) and 1 (buf1) fully staged + drained — the
+  // loop's waits then gate the rotation from iteration 1 on.
+  {
+    const unsigned short* A1 = Ag + G8_BK;
+    const unsigned short* B1 = Bg + G8_BK;
+    g8_prefetch_half<SWZ>(Ag, K, lA[0][0], tid);
+    g8_prefetch_half<SWZ>(Ag + (int64_t)128 * K, K, lA[0][1], tid);
+    g8_prefetch_half<SWZ>(Bg, K, lB[0][0], tid);
+    g8_prefetch_half<SWZ>(Bg + (int64_t)128 * K, K, lB[0][1], tid);
+    g8_prefetch_half<SWZ>(A1, K, lA[1][0], tid);
+    g8_prefetch_half<SWZ>(A1 + (int64_t)128 * K, K, lA[1][1], tid);
+    g8_prefetch_half<SWZ>(B1, K, lB[1][0], tid);
+    g8_prefetch_half<SWZ>(B1 + (int64_t)128 * K, K, lB[1][1], tid);
+    asm volatile("s_waitcnt vmcnt(0)");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  for (int T = 0; T < ntile; T += 2) {
+#pragma unroll
+    for (int gp = 0; gp < 8; ++gp) {
+      const int buf = gp >> 2;          // tile T (buf0) then T+1 (buf1)
+      const int p = gp & 3;
+      const int kc = p & 1;
+      const int mh2 = p >> 1;
+      const int kt = T + buf;
+
+      // counted waits: the iteration's two drain points
+      if (gp == 0 && T >= 2) asm volatile("s_waitcnt vmcnt(2)");
+      if (gp == 4 && T >= 2) asm volatile("s_waitcnt vmcnt(2)");
+      asm volatile("" ::: "memory");
+
+      // this phase's operand frags
+      g8_shortx8 af[4], bf[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        af[i] = g8_frag<SWZ>(lA[buf][wm], mh2 * 4 + i, kc, lane);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[j] = g8_frag<SWZ>(lB[buf][bh], wn * 4 - bh * 8 + j, kc, lane);
+
+      // slot table (see header)
+      switch (gp) {
+        case 0:
+          // T==0: tile 1 was fully staged by the prologue
+          if (T >= 2 && T + 1 < ntile) {
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 1) * G8_BK, K, lA[1][0], 1, tid);
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 1) * G8_BK + (int64_t)128 * K, K, lA[1][1], 1, tid);
+          }
+          break;
+        case 1:
+          if (T >= 2 && T + 1 < ntile) {
+            g8_prefetch_half<SWZ>(Bg + (int64_t)(T + 1) * G8_BK, K, lB[1][0], tid);
+          }
+          break;
+        case 2:
+          if (T >= 2 && T + 1 < ntile) {
+            g8_prefetch_half<SWZ>(Bg + (int64_t)(T + 1) * G8_BK + (int64_t)128 * K, K, lB[1][1], tid);
+          }
+          break;
+        case 3:
+          if (T + 2 < ntile) {
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 2) * G8_BK, K, lA[0][0], 0, tid);
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 2) * G8_BK + (int64_t)128 * K, K, lA[0][1], 0, tid);
+          }
+          break;
+        case 4:
+          if (T + 2 < ntile) {
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 2) * G8_BK, K, lA[0][0], 1, tid);
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 2) * G8_BK + (int64_t)128 * K, K, lA[0][1], 1, tid);
+          }
+          break;
+        case 5:
+          if (T + 2 < ntile)
+            g8_prefetch_half<SWZ>(Bg + (int64_t)(T + 2) * G8_BK, K, lB[0][0], tid);
+          break;
+        case 6:
+          if (T + 2 < ntile)
+            g8_prefetch_half<SWZ>(Bg + (int64_t)(T + 2) * G8_BK + (int64_t)128 * K, K, lB[0][1], tid);
+          break;
+        case 7:
+          if (T + 3 < ntile) {
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 3) * G8_BK, K, lA[1][0], 0, tid);
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 3) * G8_BK + (int64_t)128 * K, K, lA[1][1], 0, tid);
+          }
+          break;
+      }
+
+      // lockstep barrier pair around the MFMA burst
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)");
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[mh2 * 4 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[mh2 * 4 + i][j], 0, 0, 0);
+      asm volatile("" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      (void)kt;
+    }
+  }
+
+  const int c0 = lane & 15;
+  const int r0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < G8_MFRAG; ++i) {
+#pragma unroll
+    for (int j = 0; j < G8_NFRAG; ++j) {
+      int row = bm + wm * 128 + i * 16 + r0;
+      int col = bn + wn * 64 + j * 16 + c0;
+      unsigned short* cg = C + (int64_t)row * N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        cg[(int64_t)r * N] = f32_to_bf16(acc[i][j][r]);
+    }
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(G8_THREADS, 1)
+gemm8_tn_bf16_rot8(const unsigned short* __restrict__ A,
+                   const unsigned short* __restrict__ B,
+                   unsigned short* __restrict__ C, int M, int N, int K) {
+  g8_body_rot8<3>(A, B, C, M, N, K);
+}
