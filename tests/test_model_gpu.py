@@ -121,3 +121,4 @@ def test_dtensor_dispatch_on_gpu():
         assert torch.allclose(full.float(), ref.float(), atol=1e-2), i
     d2 = distribute_tensor(x, mesh, [Replicate()])
     assert torch.equal(d2.full_tensor(), x)
+    dist.destroy_process_group()
