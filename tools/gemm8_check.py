@@ -10,7 +10,7 @@ def check(M, N, K):
     b = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) / 8
     ref = a.float() @ b.float().t()
     ok = True
-    for mode, name in ((0, "burst"), (4, "rot3np"), (7, "rot8")):
+    for mode, name in ((4, "rot3np"), (7, "rot8"), (8, "rot9")):
         # race screen: the rot schedule must be multi-run stable (m152)
         for it in range(3):
             c = C.gemm_tn8(a, b, mode)
@@ -35,9 +35,9 @@ def bench(M, N, K, n=20):
     b = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
     bt = b.t()
     def rot3np(): C.gemm_tn8(a, b, 4)
-    def rot8(): C.gemm_tn8(a, b, 7)
+    def rot9(): C.gemm_tn8(a, b, 8)
     def lib(): torch.matmul(a, bt)
-    for fn, name in ((rot3np, "rot3np"), (rot8, "rot8"), (lib, "hipblaslt")):
+    for fn, name in ((rot3np, "rot3np"), (rot9, "rot9"), (lib, "hipblaslt")):
         for _ in range(3): fn()
         torch.cuda.synchronize(); t = time.perf_counter()
         for _ in range(n): fn()

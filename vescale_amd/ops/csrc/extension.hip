@@ -338,9 +338,10 @@ at::Tensor gemm_tn8(at::Tensor a, at::Tensor b, int64_t mode) {
   auto c = at::empty({M, N}, a.options());
   dim3 grid((unsigned)(M / 256), (unsigned)(N / 256), 1);
   TORCH_CHECK(mode != 5 || (K / 64) % 2 == 0, "mode 5 needs K % 128 == 0");
-  TORCH_CHECK(mode != 7 || ((K / 64) % 2 == 0 && K / 64 >= 4),
-              "mode 7 (rot8) needs an even K/64 >= 4");
-  auto kern = mode == 7   ? gemm8_tn_bf16_rot8
+  TORCH_CHECK((mode != 7 && mode != 8) || ((K / 64) % 2 == 0 && K / 64 >= 4),
+              "modes 7/8 need an even K/64 >= 4");
+  auto kern = mode == 8   ? gemm8_tn_bf16_rot9
+              : mode == 7 ? gemm8_tn_bf16_rot8
               : mode == 6 ? gemm8_tn_bf16_rot3ra
               : mode == 5 ? gemm8_tn_bf16_rot3u2
               : mode == 4 ? gemm8_tn_bf16_rot3np

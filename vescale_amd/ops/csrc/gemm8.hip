@@ -648,3 +648,196 @@ gemm8_tn_bf16_rot8(const unsigned short* __restrict__ A,
                    unsigned short* __restrict__ C, int M, int N, int K) {
   g8_body_rot8<3>(A, B, C, M, N, K);
 }
+
+
+// ---------------------------------------------------------------------------
+// rot9: the deep 8-phase schedule UNLOCKED by triple-buffering B.
+// MEASURED: correct + race-stable, 893-979 TF vs rot3np's 1027-1137 on
+// the same box — even with the guide's exact wait discipline (vmcnt(6)
+// at phases 0/4, 3 slots in flight, every producer 4-7 phases ahead) the
+// schedule macro-structure is NOT the remaining lever.  Together with
+// rot8 this isolates the rot3np->guide gap to instruction-level
+// ds_read/MFMA interleave (m196's hand-ordered stream), not waits,
+// depth, or buffering.  Kept as mode 8 for A/B.
+//
+// rot8's lesson: with 2 dbufs for both operands, FIFO vmcnt forces
+// near-full drains (consumers sit 2-4 slots behind producers).  B is the
+// constraint — its regions only die at a tile's LAST phase.  Giving B a
+// third buffer (LDS 64 KiB A + 96 KiB B = 160 KiB, the exact CU capacity)
+// makes every B slot ancient by its first read, and the A rotation then
+// supports the guide template's exact wait discipline: vmcnt(6) at
+// phases 0 and 4 only — 3 slots in flight across each wait, every slot
+// issued 4-7 phases before its consumer.
+//
+// Steady slot table, iteration (T, T+1), phases gp0..gp7
+// (gp0-3 compute tile T: A from lA[T&1], B from lB[T%3]):
+//   gp0: A(T+1)q1   [over A(T-1)q1, died prev gp7]   read gp6,7
+//   gp1: B(T+2)h0   [over B(T-1),   died prev gp7]   read next gp0-3
+//   gp2: B(T+2)h1
+//   gp3: A(T+2)q0   [over A(T)q0,   died gp1]        read next gp0,1
+//   gp4: A(T+2)q1   [over A(T)q1,   died gp3]        read next gp2,3
+//   gp5: B(T+3)h0   [over B(T),     died gp3]        read next gp4-7
+//   gp6: B(T+3)h1
+//   gp7: A(T+3)q0   [over A(T+1)q0, died gp5]        read next gp4,5
+// Waits (2 global_load_lds per slot, FIFO):
+//   vmcnt(6)@gp0: newest 3 slots (prev gp5,6,7) may be outstanding ->
+//     prev gp3 (A(T)q0) and prev gp4 (A(T)q1) are LANDED; B(T) was staged
+//     a full iteration earlier.
+//   vmcnt(6)@gp4: newest 3 (gp1,2,3) outstanding -> gp0 (A(T+1)q1) and
+//     prev gp7 (A(T+1)q0) landed; B(T+1) ancient.
+// Requires even ntile >= 2 (host mode guard).
+// ---------------------------------------------------------------------------
+template <int SWZ>
+DEV void g8_body_rot9(const unsigned short* __restrict__ A,
+                      const unsigned short* __restrict__ B,
+                      unsigned short* __restrict__ C, int M, int N, int K) {
+  __shared__ unsigned short lA[2][2][128 * G8_BK];
+  __shared__ unsigned short lB[3][2][128 * G8_BK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+  const int bh = wn >> 1;
+
+  int gx = gridDim.x, nwg = gx * (int)gridDim.y;
+  int f = blockIdx.x + gx * blockIdx.y;
+  {
+    int q = nwg >> 3, r = nwg & 7;
+    int xcd = f & 7, off = f >> 3;
+    f = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + off;
+  }
+  const int bm = (f % gx) * G8_BM;
+  const int bn = (f / gx) * G8_BN;
+  const unsigned short* Ag = A + (int64_t)bm * K;
+  const unsigned short* Bg = B + (int64_t)bn * K;
+
+  g8_floatx4 acc[G8_MFRAG][G8_NFRAG];
+#pragma unroll
+  for (int i = 0; i < G8_MFRAG; ++i)
+#pragma unroll
+    for (int j = 0; j < G8_NFRAG; ++j) acc[i][j] = (g8_floatx4)(0.f);
+
+  const int ntile = K / G8_BK;
+
+  // prologue: tiles 0,1 fully staged (A bufs 0,1 + B bufs 0,1), drained
+  {
+    const unsigned short* A1 = Ag + G8_BK;
+    const unsigned short* B1 = Bg + G8_BK;
+    g8_prefetch_half<SWZ>(Ag, K, lA[0][0], tid);
+    g8_prefetch_half<SWZ>(Ag + (int64_t)128 * K, K, lA[0][1], tid);
+    g8_prefetch_half<SWZ>(Bg, K, lB[0][0], tid);
+    g8_prefetch_half<SWZ>(Bg + (int64_t)128 * K, K, lB[0][1], tid);
+    g8_prefetch_half<SWZ>(A1, K, lA[1][0], tid);
+    g8_prefetch_half<SWZ>(A1 + (int64_t)128 * K, K, lA[1][1], tid);
+    g8_prefetch_half<SWZ>(B1, K, lB[1][0], tid);
+    g8_prefetch_half<SWZ>(B1 + (int64_t)128 * K, K, lB[1][1], tid);
+    asm volatile("s_waitcnt vmcnt(0)");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  int b3 = 0;  // lB buffer of tile T (cycles 0,2,1,0,...: +2 mod 3)
+  for (int T = 0; T < ntile; T += 2) {
+    const int bT = b3;                    // B buf of tile T
+    const int bT1 = b3 + 1 == 3 ? 0 : b3 + 1;   // tile T+1
+    const int bT2 = bT1 + 1 == 3 ? 0 : bT1 + 1; // tiles T+2 / T+3 staging
+#pragma unroll
+    for (int gp = 0; gp < 8; ++gp) {
+      const int abuf = gp >> 2;  // lA index: tile T -> 0/1 by parity of T
+      const int p = gp & 3;
+      const int kc = p & 1;
+      const int mh2 = p >> 1;
+      const unsigned short* lAband = lA[(T + abuf) & 1][wm];
+      const unsigned short* lBband = lB[gp < 4 ? bT : bT1][bh];
+
+      if ((gp == 0 || gp == 4) && T >= 2) asm volatile("s_waitcnt vmcnt(6)");
+      asm volatile("" ::: "memory");
+
+      g8_shortx8 af[4], bf[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        af[i] = g8_frag<SWZ>(lAband, mh2 * 4 + i, kc, lane);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        bf[j] = g8_frag<SWZ>(lBband, wn * 4 - bh * 8 + j, kc, lane);
+
+      switch (gp) {
+        case 0:  // A(T+1)q1 (prologue covered T==0)
+          if (T >= 2 && T + 1 < ntile) {
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 1) * G8_BK, K, lA[(T + 1) & 1][0], 1, tid);
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 1) * G8_BK + (int64_t)128 * K, K, lA[(T + 1) & 1][1], 1, tid);
+          }
+          break;
+        case 1:
+          if (T + 2 < ntile)
+            g8_prefetch_half<SWZ>(Bg + (int64_t)(T + 2) * G8_BK, K, lB[bT2][0], tid);
+          break;
+        case 2:
+          if (T + 2 < ntile)
+            g8_prefetch_half<SWZ>(Bg + (int64_t)(T + 2) * G8_BK + (int64_t)128 * K, K, lB[bT2][1], tid);
+          break;
+        case 3:
+          if (T + 2 < ntile) {
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 2) * G8_BK, K, lA[T & 1][0], 0, tid);
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 2) * G8_BK + (int64_t)128 * K, K, lA[T & 1][1], 0, tid);
+          }
+          break;
+        case 4:
+          if (T + 2 < ntile) {
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 2) * G8_BK, K, lA[T & 1][0], 1, tid);
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 2) * G8_BK + (int64_t)128 * K, K, lA[T & 1][1], 1, tid);
+          }
+          break;
+        case 5:
+          if (T + 3 < ntile)
+            g8_prefetch_half<SWZ>(Bg + (int64_t)(T + 3) * G8_BK, K, lB[bT][0], tid);
+          break;
+        case 6:
+          if (T + 3 < ntile)
+            g8_prefetch_half<SWZ>(Bg + (int64_t)(T + 3) * G8_BK + (int64_t)128 * K, K, lB[bT][1], tid);
+          break;
+        case 7:
+          if (T + 3 < ntile) {
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 3) * G8_BK, K, lA[(T + 1) & 1][0], 0, tid);
+            g8_prefetch_quarter<SWZ>(Ag + (int64_t)(T + 3) * G8_BK + (int64_t)128 * K, K, lA[(T + 1) & 1][1], 0, tid);
+          }
+          break;
+      }
+
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)");
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[mh2 * 4 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[mh2 * 4 + i][j], 0, 0, 0);
+      asm volatile("" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+    b3 = bT2;  // advance by 2 mod 3
+  }
+
+  const int c0 = lane & 15;
+  const int r0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < G8_MFRAG; ++i) {
+#pragma unroll
+    for (int j = 0; j < G8_NFRAG; ++j) {
+      int row = bm + wm * 128 + i * 16 + r0;
+      int col = bn + wn * 64 + j * 16 + c0;
+      unsigned short* cg = C + (int64_t)row * N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        cg[(int64_t)r * N] = f32_to_bf16(acc[i][j][r]);
+    }
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(G8_THREADS, 1)
+gemm8_tn_bf16_rot9(const unsigned short* __restrict__ A,
+                   const unsigned short* __restrict__ B,
+                   unsigned short* __restrict__ C, int M, int N, int K) {
+  g8_body_rot9<3>(A, B, C, M, N, K);
+}
