@@ -81,6 +81,43 @@ def test_hf_llama_tp_parity(attn):
     spawn(2, _t_hf_tp_parity, attn)
 
 
+def _t_hf_tp_sp_parity(rank, ws):
+    """sp=True: sequence-sharded (Shard(1)) activations between decoder
+    layers — the Megatron-SP boundary — on the unmodified HF model."""
+    import torch as _t
+
+    from transformers import LlamaConfig, LlamaForCausalLM
+
+    from vescale_amd.dmodule import parallelize_module
+    from vescale_amd.dtensor import Replicate, init_device_mesh
+    from vescale_amd.models.hf_llama_plan import hf_llama_tp_plan
+
+    cfg = LlamaConfig(
+        hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+        num_attention_heads=4, num_key_value_heads=2, vocab_size=256,
+        attn_implementation="eager",
+    )
+    _t.manual_seed(7)
+    ref = LlamaForCausalLM(cfg)
+    ids = _t.randint(0, 256, (2, 8))
+    with _t.no_grad():
+        rl = ref(input_ids=ids).logits
+    _t.manual_seed(7)
+    m = LlamaForCausalLM(cfg)
+    mesh = init_device_mesh("cpu", (ws,))
+    m = parallelize_module(m, mesh, hf_llama_tp_plan(sp=True))
+    with _t.no_grad():
+        out = m(input_ids=ids).logits
+    if hasattr(out, "redistribute"):
+        out = out.redistribute(placements=[Replicate()])._local_tensor
+    assert _t.allclose(out, rl, atol=2e-5), (out - rl).abs().max()
+
+
+def test_hf_llama_tp_sp_parity():
+    pytest.importorskip("transformers")
+    spawn(2, _t_hf_tp_sp_parity)
+
+
 def _t_hf_mixtral_tp_parity(rank, ws):
     from transformers import MixtralConfig, MixtralForCausalLM
 
