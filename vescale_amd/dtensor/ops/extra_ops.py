@@ -187,6 +187,10 @@ def _handler_index_write(dispatcher, op, args, kwargs):
     out = res if isinstance(res, torch.Tensor) else full_self
     if inplace and isinstance(self_dt, DTensor):
         return _writeback_inplace(self_dt, out)
+    if inplace:
+        # plain self with DTensor index/source args: op already mutated
+        # full_self (== self); in-place contract returns self itself
+        return self_dt
     mesh = self_dt._spec.mesh if isinstance(self_dt, DTensor) else _first_mesh(args, kwargs)
     return _wrap_replicate(out, mesh)
 
