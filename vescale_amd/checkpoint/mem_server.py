@@ -114,7 +114,9 @@ def fetch_checkpoint(url: str, dest_dir: str, timeout: float = 60.0) -> int:
         manifest = json.loads(r.read())
     total = 0
     for rel, size in manifest["files"].items():
-        out = os.path.join(dest_dir, rel)
+        out = os.path.normpath(os.path.join(dest_dir, rel))
+        if not out.startswith(os.path.normpath(dest_dir) + os.sep):
+            raise ValueError(f"refusing path outside dest_dir: {rel!r}")
         os.makedirs(os.path.dirname(out) or dest_dir, exist_ok=True)
         with urllib.request.urlopen(f"{url}/{rel}", timeout=timeout) as r:
             data = r.read()
