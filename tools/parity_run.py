@@ -75,7 +75,7 @@ def run(out_path):
     opt = FlatAdamW(eng, lr=1e-3, grad_clip=1.0, weight_decay=0.0)
     g = torch.Generator().manual_seed(1234)
     losses = []
-    for step in range(30):
+    for step in range(int(os.environ.get("PARITY_STEPS", "30"))):
         x = torch.randint(0, cfg.vocab_size, (4, seq), generator=g).to(dev)
         y = torch.roll(x, -1, dims=1)
         loss = eng(x, y)
@@ -89,7 +89,7 @@ def run(out_path):
             "losses": losses,
             "config": {"dim": cfg.dim, "layers": cfg.n_layers, "heads": cfg.n_heads,
                        "kv_heads": cfg.n_kv_heads, "head_dim": cfg.head_dim,
-                       "seq": seq, "batch": 4, "steps": 30},
+                       "seq": seq, "batch": 4, "steps": int(os.environ.get("PARITY_STEPS", "30"))},
             "kernels": kernels,
         },
         open(out_path, "w"),
