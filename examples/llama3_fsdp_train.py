@@ -43,7 +43,8 @@ def main():
     if args.ndtimeline:
         from vescale_amd.ndtimeline import init_ndtimers
 
-        init_ndtimers(chrome_trace_path="ndtimeline_trace.json")
+        mgr = init_ndtimers(chrome_trace_path="ndtimeline_trace.json",
+                            summary=True)
 
     cfg = getattr(M, args.model)() if args.model != "llama_tiny" else M.llama_tiny()
     if not on_gpu:
@@ -73,6 +74,21 @@ def main():
         ckpt.save(args.ckpt_dir, {"model": eng})
         if rank == 0:
             print("checkpoint saved to", args.ckpt_dir)
+    if args.ndtimeline:
+        from vescale_amd.ndtimeline import flush, wait
+
+        flush(args.steps)
+        wait()
+        sh = getattr(mgr, "summary_handler", None)
+        if sh is not None and rank == 0:
+            import json as _json
+
+            top = sorted(sh.summary().items(),
+                         key=lambda kv: -kv[1]["total_us"])[:8]
+            print("ndtimeline summary (top by total time):")
+            for k, v in top:
+                print(f"  {k:<28} n={v['count']:<5} mean={v['mean_us']:.0f}us "
+                      f"p99={v['p99_us']:.0f}us total={v['total_us']/1e3:.1f}ms")
     if world > 1:
         dist.destroy_process_group()
 
