@@ -57,7 +57,7 @@ def main():
         loss.backward()
         opt.step()
         if rank == 0 and step % 2 == 0:
-            print(f"[ws{ws}] step {step} loss {float(loss):.4f}", flush=True)
+            print(f"[ws{ws}] step {step} loss {float(loss.detach()):.4f}", flush=True)
 
     if args.save:
         ckpt.save(args.save, {"model": eng, "optimizer": opt})
@@ -73,7 +73,7 @@ def main():
         loss.backward()
         opt.step()
         if rank == 0:
-            print(f"post-load step loss {float(loss):.4f}", flush=True)
+            print(f"post-load step loss {float(loss.detach()):.4f}", flush=True)
     dist.barrier()
     dist.destroy_process_group()
 

@@ -69,7 +69,7 @@ def main():
         if rank == 0 and step % 10 == 0:
             el = time.perf_counter() - t0
             toks = args.batch * args.seq * world * (step + 1)
-            print(f"step {step} loss {float(loss):.4f} tok/s {toks/el:,.0f}")
+            print(f"step {step} loss {float(loss.detach()):.4f} tok/s {toks/el:,.0f}")
     if args.ckpt_dir:
         ckpt.save(args.ckpt_dir, {"model": eng})
         if rank == 0:

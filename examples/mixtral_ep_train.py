@@ -85,7 +85,7 @@ def main():
                     print(f"step {step}: re-placed experts "
                           f"{moe.allocator.placement}", flush=True)
         if rank == 0 and step % 5 == 0:
-            print(f"step {step} loss {float(loss):.4f}")
+            print(f"step {step} loss {float(loss.detach()):.4f}")
     dist.destroy_process_group()
 
 
