@@ -26,4 +26,28 @@ from .dtensor import (  # noqa: F401
     to_local,
 )
 
+# top-level API parity with the reference's `import vescale` surface
+# (legacy/vescale/__init__.py): the whole framework is reachable from the
+# package root.
+from . import checkpoint  # noqa: F401,E402
+from .ddp.distributed_data_parallel import DistributedDataParallel  # noqa: F401,E402
+from .dmodule.api import (  # noqa: F401,E402
+    PlacementsInterface,
+    is_dmodule,
+    parallelize_module,
+)
+from .dmp import (  # noqa: F401,E402
+    auto_parallelize_module,
+    get_plan_overriding_policy,
+    set_plan_overriding_policy,
+)
+from .initialize.deferred_init import (  # noqa: F401,E402
+    deferred_init,
+    is_deferred,
+    materialize_dparameter,
+    materialize_dtensor,
+)
+from .optim.base_optimizer import BasicOptimizer, BasicOptimizerHook  # noqa: F401,E402
+from .optim.distributed_optimizer import DistributedOptimizer  # noqa: F401,E402
+
 __version__ = "0.1.0"

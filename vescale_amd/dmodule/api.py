@@ -105,3 +105,9 @@ def parallelize_module(
     module._is_dmodule = True
     module._device_mesh = device_mesh
     return module
+
+
+def is_dmodule(module: nn.Module) -> bool:
+    """True if `module` was parallelized by parallelize_module
+    (reference dmodule/api.py is_dmodule)."""
+    return bool(getattr(module, "_is_dmodule", False))

@@ -12,6 +12,19 @@ import torch.nn as nn
 from ..dtensor import DTensor
 
 
+class BasicOptimizerHook:
+    """Grad-sync trigger hook attached to BasicOptimizer (reference
+    optim/base_optimizer.py BasicOptimizerHook): subclass and override
+    sync() to customize when/how Partial grads are reduced; the default
+    calls each registered module's finish_grad_sync()."""
+
+    def sync(self, models) -> None:
+        for m in models or []:
+            fn = getattr(m, "finish_grad_sync", None)
+            if callable(fn):
+                fn()
+
+
 class BasicOptimizer:
     def __init__(
         self,
