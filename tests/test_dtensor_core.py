@@ -606,3 +606,24 @@ def _t_srs_shard_compose(rank, ws):
 
 def test_srs_shard_compose():
     spawn(4, _t_srs_shard_compose)
+
+
+def _t_to_local_grad_placements(rank, ws):
+    """to_local(grad_placements=[Partial()]): the caller declares that the
+    gradient flowing back is a PARTIAL sum (reference _api.py:410) — the
+    wrapped grad then reduces on the way to the DTensor's layout."""
+    mesh = init_device_mesh("cpu", (ws,))
+    x = distribute_tensor(torch.ones(4), mesh, [Replicate()])
+    x.requires_grad_(True)
+    lt = x.to_local(grad_placements=[Partial()])
+    # each rank contributes a DIFFERENT local grad; declared Partial means
+    # the true grad is their SUM
+    (lt * float(rank + 1)).sum().backward()
+    g = x.grad
+    gf = g.redistribute(placements=[Replicate()])._local_tensor
+    expect = float(sum(r + 1 for r in range(ws)))
+    assert torch.allclose(gf, torch.full((4,), expect)), gf
+
+
+def test_to_local_grad_placements():
+    spawn(2, _t_to_local_grad_placements)

@@ -134,8 +134,15 @@ class DTensor(torch.Tensor):
         spec = DTensorSpec(device_mesh, placements, tm)
         return _FromLocal.apply(local_tensor, spec)
 
-    def to_local(self) -> torch.Tensor:
-        return _ToLocal.apply(self)
+    def to_local(self, *, grad_placements=None) -> torch.Tensor:
+        """Local tensor of this rank.  grad_placements (reference
+        _api.py:410 kwarg): declare the layout the GRADIENT flowing back
+        into this boundary will have (e.g. Partial("sum") when downstream
+        produces a partial-sum), overriding the default assumption that it
+        matches this DTensor's placements."""
+        if grad_placements is not None and not isinstance(grad_placements, tuple):
+            grad_placements = tuple(grad_placements)
+        return _ToLocal.apply(self, grad_placements)
 
     @property
     def device_mesh(self) -> DeviceMesh:
@@ -160,9 +167,9 @@ class DTensor(torch.Tensor):
         assert placements is not None
         return Redistribute.apply(self, tuple(placements), async_op)
 
-    def full_tensor(self) -> torch.Tensor:
+    def full_tensor(self, *, grad_placements=None) -> torch.Tensor:
         rep = self.redistribute(placements=[Replicate()] * self._spec.mesh.ndim)
-        return rep.to_local()
+        return rep.to_local(grad_placements=grad_placements)
 
     # convenience parity helpers -----------------------------------------
     def local_shape(self) -> Tuple[int, ...]:
@@ -201,8 +208,9 @@ class _FromLocal(torch.autograd.Function):
 
 class _ToLocal(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, dtensor: "DTensor"):
+    def forward(ctx, dtensor: "DTensor", grad_placements=None):
         ctx.spec = dtensor._spec
+        ctx.grad_placements = grad_placements
         lt = dtensor._local_tensor
         return lt.view_as(lt) if lt.requires_grad else lt
 
@@ -222,12 +230,20 @@ class _ToLocal(torch.autograd.Function):
             if tgt != g._spec.placements:
                 g = g.redistribute(placements=tgt)
             grad_output = g._local_tensor
-        grad_placements = tuple(
-            Replicate() if isinstance(p, Partial) else p for p in spec.placements
-        )
+        if ctx.grad_placements is not None:
+            # caller-declared grad layout (reference to_local kwarg): the
+            # plain grad coming back IS laid out per these placements
+            grad_placements = tuple(ctx.grad_placements)
+        else:
+            grad_placements = tuple(
+                Replicate() if isinstance(p, Partial) else p for p in spec.placements
+            )
         tm = TensorMeta(spec.tensor_meta.shape, spec.tensor_meta.stride, grad_output.dtype)
         gspec = DTensorSpec(spec.mesh, grad_placements, tm)
-        return DTensor(grad_output, gspec, requires_grad=grad_output.requires_grad)
+        return (
+            DTensor(grad_output, gspec, requires_grad=grad_output.requires_grad),
+            None,
+        )
 
 
 _dynamo_disable = getattr(getattr(torch, "compiler", None), "disable", None)
