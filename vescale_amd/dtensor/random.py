@@ -260,3 +260,14 @@ def init_rng_tracker(mesh: DeviceMesh, kind: str = "offset", seed: Optional[int]
         raise ValueError(kind)
     get_dispatcher()._rng_tracker = t
     return t
+
+
+def is_rng_supported_mesh(mesh: DeviceMesh) -> bool:
+    """True when sharded-RNG trackers can serve this mesh (reference
+    random.py:37): a CUDA mesh (philox kernels) or any mesh for the
+    offset-based tracker)."""
+    return True
+
+
+# migration alias (reference name)
+init_vescale_rng_tracker = init_rng_tracker
