@@ -1,5 +1,5 @@
-from .pipe_stage import PipeModule, construct_pipeline_stage
-from .pipe_emmiter import ScheduleEngine
+from .pipe_stage import PipeModule, build_shared_module_group, construct_pipeline_stage, construct_stage_modules
+from .pipe_emmiter import ScheduleEngine, validate_pipeline_schedule
 from .pipe_parser import (
     hf_symbolic_trace,
     parse_huggingface_model,
@@ -15,6 +15,9 @@ from .p2p_communication import (
 __all__ = [
     "PipeModule",
     "construct_pipeline_stage",
+    "construct_stage_modules",
+    "build_shared_module_group",
+    "validate_pipeline_schedule",
     "ScheduleEngine",
     "hf_symbolic_trace",
     "parse_huggingface_model",

@@ -269,3 +269,17 @@ class ScheduleEngine:
         g = inp.grad
         assert g is not None, f"no input grad for chunk {ck} mb {m}"
         return g
+
+
+def validate_pipeline_schedule(plan: PipelineParallelPlan) -> None:
+    """Validate schedule/virtual-chunk consistency up front (reference
+    pipe_emmiter.py:345) — same checks the engine applies lazily."""
+    st = plan.schedule_type
+    if st == PipelineScheduleType.INTERLEAVED_1F1B:
+        assert plan.virtual_chunks > 1, "interleaved 1F1B needs virtual_chunks > 1"
+    elif st == PipelineScheduleType.SIMPLE_1F1B:
+        assert plan.virtual_chunks == 1, "1F1B needs virtual_chunks == 1"
+    elif st == PipelineScheduleType.ZERO_BUBBLE:
+        assert plan.virtual_chunks in (1, 2), (
+            "zero-bubble supports virtual_chunks 1 (ZB-H1) or 2 (ZB-V)"
+        )

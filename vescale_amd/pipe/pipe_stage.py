@@ -145,3 +145,20 @@ def construct_pipeline_stage(
         part = parts[idx]
         chunks.append(nn.Sequential(*part) if len(part) != 1 else part[0])
     return PipeModule(chunks, stage_id, plan.num_stages)
+
+
+def construct_stage_modules(
+    module_list: Sequence[nn.Module],
+    plan: PipelineParallelPlan,
+    stage_id: int,
+    *,
+    split_points: Optional[List[int]] = None,
+) -> List[nn.Module]:
+    """The raw chunk modules this stage hosts (reference pipe_stage.py:249
+    construct_stage_modules) — construct_pipeline_stage without the
+    PipeModule wrapper."""
+    return list(
+        construct_pipeline_stage(
+            module_list, plan, stage_id, split_points=split_points
+        ).chunks
+    )
