@@ -1,4 +1,4 @@
-from .api import parallelize_experts, rebalance_experts
+from .api import is_experts_parallized, parallelize_experts, rebalance_experts
 from .experts_allocator import (
     BasicExpertsAllocator,
     ExpertsAllocator,

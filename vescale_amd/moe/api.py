@@ -151,3 +151,12 @@ def rebalance_experts(
             layer.experts[e] = mod
     layer.local_expert_ids = alloc.experts_of(rank)
     return True
+
+
+def is_experts_parallized(model) -> bool:
+    """True if any MoE layer in `model` has been expert-parallelized
+    (reference moe/api.py is_experts_parallized; spelling kept)."""
+    for mod in model.modules():
+        if getattr(mod, "_ep", False):
+            return True
+    return False
