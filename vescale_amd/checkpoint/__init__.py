@@ -33,6 +33,8 @@ from .planner import VeScaleLoadPlanner, VeScaleSavePlanner
 from .ragged_boxes import break_ragged_box
 
 __all__ = [
+    "VeScaleCheckpointer",
+    "CheckpointState",
     "save",
     "load",
     "VeScaleSavePlanner",
@@ -207,3 +209,23 @@ def load(
             except Exception:
                 pass  # in-place DCP load already mutated the tensors
     return checkpoint_state
+
+
+# reference-shape surface (api/vescale_checkpointer.py): a class namespace
+# over the same save/load; CheckpointState is the {"model": ..., ...} dict
+CheckpointState = Dict[str, Any]
+
+
+class VeScaleCheckpointer:
+    """Classmethod namespace over save/load (reference VeScaleCheckpointer)."""
+
+    @classmethod
+    def save(cls, path: str, checkpoint_state: CheckpointState,
+             async_checkpoint: bool = False):
+        return save(path, checkpoint_state, async_checkpoint=async_checkpoint)
+
+    @classmethod
+    def load(cls, path: str, checkpoint_state: CheckpointState,
+             broadcast_checkpoint: bool = False):
+        return load(path, checkpoint_state,
+                    broadcast_checkpoint=broadcast_checkpoint)
