@@ -30,7 +30,7 @@ def main():
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     on_gpu = torch.cuda.is_available()
-    dist.init_process_group("nccl" if on_gpu else "gloo")
+    dist.init_process_group(os.environ.get("VESCALE_BACKEND", "nccl" if on_gpu else "gloo"))
     rank = dist.get_rank()
     ws = dist.get_world_size()
     dev = torch.device("cuda", rank % max(torch.cuda.device_count(), 1)) if on_gpu else torch.device("cpu")

@@ -28,7 +28,7 @@ def main():
     args = ap.parse_args()
 
     if not dist.is_initialized():
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = os.environ.get("VESCALE_BACKEND", "nccl" if torch.cuda.is_available() else "gloo")
         dist.init_process_group(backend)
     rank = dist.get_rank()
     ws = dist.get_world_size()

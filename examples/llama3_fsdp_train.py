@@ -33,7 +33,7 @@ def main():
     on_gpu = torch.cuda.is_available()
     device = torch.device(f"cuda:{os.environ.get('LOCAL_RANK', 0)}") if on_gpu else torch.device("cpu")
     if world > 1:
-        dist.init_process_group("nccl" if on_gpu else "gloo")
+        dist.init_process_group(os.environ.get("VESCALE_BACKEND", "nccl" if on_gpu else "gloo"))
 
     from vescale_amd.dtensor import init_device_mesh
     from vescale_amd.fsdp import FSDP, FlatAdamW

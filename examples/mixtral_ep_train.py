@@ -26,7 +26,7 @@ def main():
     args = ap.parse_args()
 
     on_gpu = torch.cuda.is_available()
-    dist.init_process_group("nccl" if on_gpu else "gloo")
+    dist.init_process_group(os.environ.get("VESCALE_BACKEND", "nccl" if on_gpu else "gloo"))
     rank = dist.get_rank()
     world = dist.get_world_size()
     device = torch.device("cuda") if on_gpu else torch.device("cpu")

@@ -32,7 +32,7 @@ def main():
     ap.add_argument("--blocks", type=int, default=8)
     args = ap.parse_args()
 
-    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    backend = os.environ.get("VESCALE_BACKEND", "nccl" if torch.cuda.is_available() else "gloo")
     dist.init_process_group(backend)
     rank, ws = dist.get_rank(), dist.get_world_size()
     if torch.cuda.is_available():
