@@ -41,7 +41,14 @@ class _AllToAllSingle(torch.autograd.Function):
 
 
 def _a2a_fallback(out, inp, out_splits, in_splits, group):
-    """gloo: pairwise isend/irecv."""
+    """gloo: pairwise isend/irecv.  CUDA tensors are staged through CPU —
+    gloo point-to-point on device memory is outside its supported envelope
+    (observed to take down a node when two CUDA ranks shared a GPU)."""
+    if inp.is_cuda:
+        cpu_out = torch.empty(out.shape, dtype=out.dtype)
+        _a2a_fallback(cpu_out, inp.cpu(), out_splits, in_splits, group)
+        out.copy_(cpu_out)
+        return
     ws = dist.get_world_size(group)
     me = dist.get_rank(group)
     iss = list(in_splits) if in_splits else [inp.shape[0] // ws] * ws
