@@ -224,20 +224,41 @@ def _communicate(
         pass
 
     dev = device or (torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu"))
+    # gloo p2p on device memory is outside its supported envelope (a CUDA
+    # send over gloo was observed to take a node down); transport through
+    # CPU and copy back onto the requested device
+    gloo_cuda = not _use_batched(pg) and (
+        dev.type == "cuda"
+        or (tensor_send_prev is not None and tensor_send_prev.is_cuda)
+        or (tensor_send_next is not None and tensor_send_next.is_cuda)
+    )
+    wire_dev = torch.device("cpu") if gloo_cuda else dev
+    send_prev = tensor_send_prev
+    send_next = tensor_send_next
+    if gloo_cuda:
+        if send_prev is not None and send_prev.is_cuda:
+            send_prev = send_prev.detach().cpu()
+        if send_next is not None and send_next.is_cuda:
+            send_next = send_next.detach().cpu()
     ops: List[dist.P2POp] = []
     if recv_prev:
         assert recv_shape is not None
-        tensor_recv_prev = torch.empty(recv_shape, dtype=recv_dtype or torch.float32, device=dev)
+        tensor_recv_prev = torch.empty(recv_shape, dtype=recv_dtype or torch.float32, device=wire_dev)
         ops.append(dist.P2POp(dist.irecv, tensor_recv_prev, peer=prev_rank, group=pg))
     if recv_next:
         assert recv_shape is not None
-        tensor_recv_next = torch.empty(recv_shape, dtype=recv_dtype or torch.float32, device=dev)
+        tensor_recv_next = torch.empty(recv_shape, dtype=recv_dtype or torch.float32, device=wire_dev)
         ops.append(dist.P2POp(dist.irecv, tensor_recv_next, peer=next_rank, group=pg))
-    if tensor_send_prev is not None:
-        ops.append(dist.P2POp(dist.isend, tensor_send_prev.contiguous(), peer=prev_rank, group=pg))
-    if tensor_send_next is not None:
-        ops.append(dist.P2POp(dist.isend, tensor_send_next.contiguous(), peer=next_rank, group=pg))
+    if send_prev is not None:
+        ops.append(dist.P2POp(dist.isend, send_prev.contiguous(), peer=prev_rank, group=pg))
+    if send_next is not None:
+        ops.append(dist.P2POp(dist.isend, send_next.contiguous(), peer=next_rank, group=pg))
     _run_p2p_ops(ops, pg)
+    if gloo_cuda:
+        if tensor_recv_prev is not None:
+            tensor_recv_prev = tensor_recv_prev.to(dev)
+        if tensor_recv_next is not None:
+            tensor_recv_next = tensor_recv_next.to(dev)
     return tensor_recv_prev, tensor_recv_next
 
 
