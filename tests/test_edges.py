@@ -198,3 +198,66 @@ def _t_cache_bounded(rank, ws):
 
 def test_dispatch_cache_bounded():
     spawn(1, _t_cache_bounded)
+
+
+# ---------------------------------------------------------------------------
+# placement hash/eq contract + global equal/allclose
+# (reference legacy/test/dtensor/hash/test_hash.py and general/test_equal.py)
+# ---------------------------------------------------------------------------
+def test_placement_hash_eq_contract():
+    from vescale_amd.dtensor.placement_types import (
+        InterleavedShard,
+        Partial,
+        RaggedShard,
+        Replicate,
+        Shard,
+        _StridedRaggedShard,
+    )
+
+    # equal objects must have equal hash; copies are equal
+    pairs = [
+        (Shard(0), Shard(0)),
+        (Replicate(), Replicate()),
+        (Partial("sum"), Partial("sum")),
+        (InterleavedShard(0, 2), InterleavedShard(0, 2)),
+        (RaggedShard((0,), (2, 3)), RaggedShard((0,), (2, 3))),
+    ]
+    for a, b in pairs:
+        assert a == b and hash(a) == hash(b), (a, b)
+    # distinct parameterisations are unequal
+    assert Shard(0) != Shard(1)
+    assert Partial("sum") != Partial("max")
+    assert InterleavedShard(0, 2) != InterleavedShard(0, 4)
+    assert RaggedShard((0,), (2, 3)) != RaggedShard((0,), (3, 2))
+    # cross-class: same fields never compare equal (dispatch-cache safety)
+    assert Shard(0) != InterleavedShard(0, 2)
+    assert Shard(0) != Replicate()
+    assert RaggedShard((0,), (2, 3)) != _StridedRaggedShard((0,), (2, 3))
+    # tuples of placements (cache keys) follow from element eq/hash
+    k1 = (Shard(0), Replicate())
+    k2 = (Shard(0), Replicate())
+    assert k1 == k2 and hash(k1) == hash(k2)
+    assert hash((Shard(0),)) != hash((Shard(1),)) or (Shard(0),) != (Shard(1),)
+
+
+def _t_global_equal_allclose(rank, ws):
+    from vescale_amd.dtensor import _utils as U
+
+    mesh = init_device_mesh("cpu", (ws,))
+    g = torch.arange(16, dtype=torch.float32).reshape(4, 4)
+    d1 = distribute_tensor(g, mesh, [Shard(1)])
+    d2 = distribute_tensor(g.clone(), mesh, [Shard(1)])
+    assert U.equal(d1, d2) and U.allclose(d1, d2)
+    # different placement, same global value: allclose compares full tensors
+    d3 = distribute_tensor(g.clone(), mesh, [Shard(0)])
+    assert U.allclose(d1, d3)
+    # value perturbation below tolerance vs above
+    d4 = distribute_tensor(g + 1e-7, mesh, [Shard(1)])
+    assert U.allclose(d1, d4, atol=1e-5)
+    assert not U.equal(d1, d4)
+    d5 = distribute_tensor(g + 1.0, mesh, [Shard(1)])
+    assert not U.allclose(d1, d5)
+
+
+def test_global_equal_allclose():
+    spawn(2, _t_global_equal_allclose)
