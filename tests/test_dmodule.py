@@ -162,3 +162,76 @@ def _t_obj_return(rank, ws):
 
 def test_obj_return():
     spawn(2, _t_obj_return)
+
+
+def _t_obj_return_named(rank, ws):
+    """Dict-shaped plans: name-keyed input plans bind forward() params
+    (positional or kw), and name-keyed output plans convert fields of
+    dataclass / Mapping returns (reference dmodule/test_obj_return.py +
+    _hook.py dict-like paths)."""
+    import dataclasses
+
+    from vescale_amd.dmodule import parallelize_module
+    from vescale_amd.dtensor import init_device_mesh
+
+    @dataclasses.dataclass
+    class Out:
+        last_hidden: torch.Tensor = None
+        hidden_plus: torch.Tensor = None
+        note: str = "keep"
+
+    class Net(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = nn.Linear(8, 8)
+
+        def forward(self, x, scale=1.0):
+            h = self.fc(x) * scale
+            return Out(last_hidden=h, hidden_plus=h + 1)
+
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(3)
+    net = Net()
+    torch.manual_seed(3)
+    ref = Net()
+    plan = {
+        "parameter": {},
+        "forward": {
+            "input": {"x": [Replicate()]},
+            "output": {"last_hidden": [Replicate()], "hidden_plus": [Shard(0)]},
+        },
+    }
+    net = parallelize_module(net, mesh, plan)
+    x = torch.randn(4, 8)
+    # kwargs path exercises the signature binding too
+    out = net(x, scale=2.0)
+    assert isinstance(out.last_hidden, DTensor)
+    assert tuple(out.last_hidden.placements) == (Replicate(),)
+    assert isinstance(out.hidden_plus, DTensor)
+    assert tuple(out.hidden_plus.placements) == (Shard(0),)
+    assert out.note == "keep"
+    want = ref(x, scale=2.0)
+    assert torch.allclose(out.last_hidden.to_local(), want.last_hidden, atol=1e-6)
+    assert torch.allclose(out.hidden_plus.full_tensor(), want.hidden_plus, atol=1e-6)
+
+    # Mapping return + dict plan
+    class NetD(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = nn.Linear(8, 8)
+
+        def forward(self, x):
+            h = self.fc(x)
+            return {"hidden": h, "tag": "t"}
+
+    torch.manual_seed(4)
+    netd = parallelize_module(
+        NetD(), mesh,
+        {"parameter": {}, "forward": {"output": {"hidden": [Replicate()]}}},
+    )
+    outd = netd(x)
+    assert isinstance(outd["hidden"], DTensor) and outd["tag"] == "t"
+
+
+def test_obj_return_named():
+    spawn(2, _t_obj_return_named)

@@ -5,10 +5,18 @@ PostHookOutput).  Each hook converts positional tensor args:
   plain tensor -> DTensor.from_local(t, mesh, placements)
   DTensor      -> redistribute(placements)   [=> RCCL comm]
 A None placement entry leaves the arg untouched.
+
+Plans may also be DICTS keyed by name (reference _hook.py:82-90,158,
+219-251): for inputs the keys are forward() parameter names (positional
+or keyword), for outputs the keys are fields of a Mapping / dataclass /
+HF ModelOutput return — unnamed fields pass through untouched.
 """
 from __future__ import annotations
 
+import dataclasses
+import inspect
 import re
+from collections.abc import Mapping
 from typing import Any, Dict
 
 import torch
@@ -80,7 +88,12 @@ def install_forward_hooks(root: nn.Module, mesh: DeviceMesh, fwd_plan: Dict[str,
         ip = match(input_plans, fqn) if fqn or "" in input_plans else match(input_plans, fqn)
         op = match(output_plans, fqn)
         if ip is not None:
-            mod.register_forward_pre_hook(_make_pre_hook(ip, mesh))
+            if isinstance(ip, Mapping):
+                mod.register_forward_pre_hook(
+                    _make_named_pre_hook(ip, mesh, mod), with_kwargs=True
+                )
+            else:
+                mod.register_forward_pre_hook(_make_pre_hook(ip, mesh))
         if op is not None:
             mod.register_forward_hook(_make_post_hook(op, mesh))
 
@@ -101,8 +114,58 @@ def _make_pre_hook(plan_list, mesh):
     return _dynamo_opaque(hook)
 
 
+def _make_named_pre_hook(plan_dict, mesh, mod):
+    """Name-keyed input plan: convert forward() arguments by parameter
+    name, whether passed positionally or as keywords."""
+    try:
+        sig = inspect.signature(mod.forward)
+    except (TypeError, ValueError):
+        sig = None
+
+    def hook(mod, args, kwargs):
+        if sig is None:
+            return args, kwargs
+        try:
+            bound = sig.bind(*args, **kwargs)
+        except TypeError:
+            return args, kwargs
+        for name, pl in plan_dict.items():
+            if name in bound.arguments:
+                bound.arguments[name] = _convert(bound.arguments[name], pl, mesh)
+        return bound.args, bound.kwargs
+
+    return _dynamo_opaque(hook)
+
+
+def _convert_named(output, plan_dict, mesh):
+    """Convert named fields of a Mapping / dataclass / HF-ModelOutput
+    return; fields absent from the plan pass through untouched."""
+    if isinstance(output, Mapping):
+        new = {
+            k: _convert(v, plan_dict.get(k), mesh) if k in plan_dict else v
+            for k, v in output.items()
+        }
+        return type(output)(**new)
+    if dataclasses.is_dataclass(output) and not isinstance(output, type):
+        for f in dataclasses.fields(output):
+            if f.name in plan_dict:
+                v = _convert(getattr(output, f.name), plan_dict[f.name], mesh)
+                # HF ModelOutput subclasses are also dicts; plain frozen
+                # dataclasses need object.__setattr__
+                try:
+                    setattr(output, f.name, v)
+                except (AttributeError, dataclasses.FrozenInstanceError):
+                    object.__setattr__(output, f.name, v)
+                if isinstance(output, Mapping) and f.name in output:
+                    output[f.name] = v
+        return output
+    return output
+
+
 def _make_post_hook(plan_list, mesh):
     def hook(mod, args, output):
+        if isinstance(plan_list, Mapping):
+            return _convert_named(output, plan_list, mesh)
         if isinstance(output, torch.Tensor):
             pl = plan_list[0] if plan_list else None
             return _convert(output, pl, mesh)
