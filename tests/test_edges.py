@@ -261,3 +261,31 @@ def _t_global_equal_allclose(rank, ws):
 
 def test_global_equal_allclose():
     spawn(2, _t_global_equal_allclose)
+
+
+# ---------------------------------------------------------------------------
+# allreduce reassociation: Partial stays Partial through pointwise adds
+# (the reference needs an explicit DeferReshardMode / ".lazy" plan suffix
+# for this — legacy/vescale/dtensor/_diff.py:74 — here it is the DEFAULT
+# dispatch rule: allreduce(x)+allreduce(y) == allreduce(x+y), one comm)
+# ---------------------------------------------------------------------------
+def _t_partial_add_reassociation(rank, ws):
+    from vescale_amd.debug import CommDebugMode
+    from vescale_amd.dtensor import Partial
+
+    mesh = init_device_mesh("cpu", (ws,))
+    xa = torch.full((4,), float(rank + 1))
+    xb = torch.full((4,), float(10 * (rank + 1)))
+    a = DTensor.from_local(xa, mesh, [Partial("sum")])
+    b = DTensor.from_local(xb, mesh, [Partial("sum")])
+    with CommDebugMode() as cm:
+        c = a + b                      # stays Partial: zero comms
+        assert c.placements[0].is_partial()
+        r = c.redistribute(placements=[Replicate()])
+    assert cm.total == 1, f"expected exactly one allreduce, got {cm.get_comm_counts()}"
+    want = sum(i + 1 + 10 * (i + 1) for i in range(ws))
+    assert torch.allclose(r.to_local(), torch.full((4,), float(want)))
+
+
+def test_partial_add_reassociation():
+    spawn(2, _t_partial_add_reassociation)
