@@ -223,3 +223,72 @@ def _t_2d_tp_dp(rank, ws):
 
 def test_2d_tp_dp_zero2():
     spawn(4, _t_2d_tp_dp)
+
+
+# ---------------------------------------------------------------------------
+# tied/shared parameter under DDP + DistributedOptimizer
+# (reference legacy/test/parallel/ddp_optim/test_shared_weight.py: a weight
+# used both as embedding and as the output projection must accumulate grads
+# from both uses, be reduced ONCE, and stay tied after optimizer updates)
+# ---------------------------------------------------------------------------
+class _TiedNet(nn.Module):
+    def __init__(self, vocab=12, d=8):
+        super().__init__()
+        self.emb = nn.Embedding(vocab, d)
+        self.fc = nn.Linear(d, d)
+
+    def forward(self, idx):
+        h = torch.tanh(self.fc(self.emb(idx)))
+        return h @ self.emb.weight.t()  # tied output projection
+
+
+def _tied_data(n_steps, bs=4, seqlen=5, vocab=12):
+    g = torch.Generator().manual_seed(11)
+    return [torch.randint(0, vocab, (bs, seqlen), generator=g) for _ in range(n_steps)]
+
+
+def _tied_ref(n_steps):
+    torch.manual_seed(5)
+    net = _TiedNet()
+    opt = torch.optim.AdamW(net.parameters(), lr=1e-2)
+    losses = []
+    for idx in _tied_data(n_steps):
+        logits = net(idx)
+        loss = F.cross_entropy(logits.reshape(-1, logits.size(-1)), idx.reshape(-1))
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(loss))
+    return losses
+
+
+def _t_tied_weight(rank, ws, n_steps, use_do):
+    torch.manual_seed(5)
+    net = _TiedNet()
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("DP",))
+    ddp = DDP(net, mesh, use_distributed_optimizer=use_do)
+    inner = torch.optim.AdamW(net.parameters(), lr=1e-2)
+    opt = DistributedOptimizer(inner, [ddp]) if use_do else BasicOptimizer(inner, [ddp])
+    losses = []
+    for idx in _tied_data(n_steps):
+        shard = torch.chunk(idx, ws)[rank]
+        logits = ddp(shard)
+        loss = F.cross_entropy(logits.reshape(-1, logits.size(-1)), shard.reshape(-1))
+        loss.backward()
+        ddp.finish_grad_sync()
+        opt.step()
+        opt.zero_grad()
+        g = loss.detach().clone()
+        torch.distributed.all_reduce(g)
+        losses.append(float(g) / ws)
+    ref = _tied_ref(n_steps)
+    for a, b in zip(losses, ref):
+        assert abs(a - b) < 1e-4, (losses, ref)
+
+
+def test_ddp_tied_weight_parity():
+    spawn(2, _t_tied_weight, 4, False)
+
+
+def test_ddp_zero2_tied_weight_parity():
+    spawn(2, _t_tied_weight, 4, True)
