@@ -235,3 +235,42 @@ def _t_obj_return_named(rank, ws):
 
 def test_obj_return_named():
     spawn(2, _t_obj_return_named)
+
+
+def _t_state_dict_saveload(rank, ws):
+    """torch.save/load roundtrip of a parallelized module's state dict
+    (reference dmodule/test_saveload.py): entries are DTensors, reload
+    restores bitwise-equal locals."""
+    import tempfile
+
+    from vescale_amd.dtensor import init_device_mesh
+
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(9)
+    net = nn.Sequential(nn.Linear(8, 16), nn.Tanh(), nn.Linear(16, 8))
+    plan = {
+        "parameter": {
+            r"0.weight": [Shard(0)], r"0.bias": [Shard(0)],
+            r"2.weight": [Shard(1)], r"2.bias": [Replicate()],
+        },
+        "forward": {"input": [[Replicate()]]},
+    }
+    net = parallelize_module(net, mesh, plan)
+    sd = net.state_dict()
+    assert all(isinstance(v, DTensor) for v in sd.values()), sd.keys()
+    path = tempfile.mktemp(suffix=f".r{rank}.pt")
+    torch.save(sd, path)
+    # perturb, then restore
+    with torch.no_grad():
+        for p in net.parameters():
+            p.add_(1.0)
+    loaded = torch.load(path, weights_only=False)
+    net.load_state_dict(loaded)
+    for k, v in net.state_dict().items():
+        assert torch.equal(v._local_tensor, sd[k]._local_tensor), k
+    import os as _os
+    _os.unlink(path)
+
+
+def test_state_dict_saveload():
+    spawn(2, _t_state_dict_saveload)

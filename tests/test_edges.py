@@ -289,3 +289,23 @@ def _t_partial_add_reassociation(rank, ws):
 
 def test_partial_add_reassociation():
     spawn(2, _t_partial_add_reassociation)
+
+
+def _t_from_local_zero_copy(rank, ws):
+    """from_local must alias, not copy (reference dtensor/memory/
+    test_memory.py asserts zero new allocations): the DTensor's local
+    view shares storage with the source tensor."""
+    mesh = init_device_mesh("cpu", (ws,))
+    t = torch.randn(4, 4)
+    d = DTensor.from_local(t, mesh, [Shard(0)])
+    assert d._local_tensor.data_ptr() == t.data_ptr()
+    # and to_local of that DTensor aliases the same storage
+    l = d.to_local()
+    assert l.data_ptr() == t.data_ptr()
+    # in-place writes through the source are visible in the DTensor
+    t.fill_(3.0)
+    assert torch.all(d._local_tensor == 3.0)
+
+
+def test_from_local_zero_copy():
+    spawn(2, _t_from_local_zero_copy)

@@ -49,11 +49,19 @@ class _MeshEnv(threading.local):
             _init_process_groups=False,
         )
         child._dim_groups = [pg]
+        _LIVE_MESHES.setdefault(child._registry_key(), child)
         self.child_to_parent[child] = (parent, mesh_dim)
         return child
 
 
 _mesh_env = _MeshEnv()
+
+# live meshes by description, for pickle rebinding: a mesh travels as its
+# description (device_type + rank array + names) and re-attaches to the
+# process's live communicators on load — ProcessGroups never serialize.
+# (Reference parity: dmodule/test_saveload.py torch.saves DTensor state
+# dicts; DCP checkpointing does the same via spec metadata.)
+_LIVE_MESHES: Dict[tuple, "DeviceMesh"] = {}
 
 
 def _get_device_handle(device_type: str):
@@ -100,6 +108,7 @@ class DeviceMesh:
             )
             assert self.mesh.ndim == 1, "pg-based construction is 1-D"
             self._dim_groups = [pg]
+            _LIVE_MESHES.setdefault(self._registry_key(), self)
             self._setup_device()
             return
 
@@ -107,6 +116,7 @@ class DeviceMesh:
             self._maybe_init_default_pg()
             self._setup_device()
             self._init_dim_groups()
+            _LIVE_MESHES.setdefault(self._registry_key(), self)
 
     # ------------------------------------------------------------------
     def _maybe_init_default_pg(self) -> None:
@@ -183,10 +193,60 @@ class DeviceMesh:
         if isinstance(mesh_dim, str):
             assert self.mesh_dim_names, "mesh has no dim names"
             mesh_dim = self.mesh_dim_names.index(mesh_dim)
+        self._ensure_groups()
         return self._dim_groups[mesh_dim]
 
     def get_all_groups(self) -> List[dist.ProcessGroup]:
+        self._ensure_groups()
         return list(self._dim_groups)
+
+    # ------------------------------------------------------------------
+    # pickling: a mesh serializes as its DESCRIPTION only (communicators
+    # never pickle).  On load it rebinds to the live mesh with the same
+    # description, or lazily re-creates the dim groups on first use (all
+    # ranks load in lockstep in the save/load pattern, so group creation
+    # stays collective-ordered).
+    def _registry_key(self) -> tuple:
+        return (
+            self.device_type,
+            tuple(self.mesh.shape),
+            tuple(self.mesh.flatten().tolist()),
+            self.mesh_dim_names,
+        )
+
+    def __getstate__(self):
+        return {
+            "device_type": self.device_type,
+            "mesh": self.mesh.tolist(),
+            "mesh_dim_names": self.mesh_dim_names,
+        }
+
+    def __setstate__(self, state):
+        self.device_type = state["device_type"]
+        self.mesh = torch.tensor(state["mesh"], dtype=torch.int64)
+        if self.mesh.ndim == 0:
+            self.mesh = self.mesh.reshape(1)
+        self.mesh_dim_names = state["mesh_dim_names"]
+        self._flat_rank_map = {int(r): i for i, r in enumerate(self.mesh.flatten())}
+        live = _LIVE_MESHES.get(self._registry_key())
+        self._dim_groups = live._dim_groups if live is not None else None
+
+    def _ensure_groups(self) -> None:
+        if self._dim_groups is not None:
+            return
+        live = _LIVE_MESHES.get(self._registry_key())
+        if live is not None and live._dim_groups:
+            self._dim_groups = live._dim_groups
+            return
+        if not dist.is_initialized():
+            raise RuntimeError(
+                f"unpickled {self!r} has no live communicators and "
+                "torch.distributed is not initialized — construct the mesh "
+                "(or init_device_mesh) before using the loaded DTensors"
+            )
+        self._dim_groups = []
+        self._init_dim_groups()
+        _LIVE_MESHES.setdefault(self._registry_key(), self)
 
     def get_dim_groups(self, mesh_dim: Optional[int] = None):
         if mesh_dim is None:
