@@ -309,3 +309,29 @@ def _t_from_local_zero_copy(rank, ws):
 
 def test_from_local_zero_copy():
     spawn(2, _t_from_local_zero_copy)
+
+
+def _t_meta_device_dtensor(rank, ws):
+    """Meta-device DTensors: sharding propagation with ZERO comms and zero
+    real memory (reference dtensor/meta_device/test_meta_device.py) — the
+    basis of deferred init for models too big to materialize."""
+    from vescale_amd.debug import CommDebugMode
+    from vescale_amd.dtensor import distribute_tensor
+
+    mesh = init_device_mesh("cpu", (ws,))
+    with CommDebugMode() as cm:
+        t = torch.empty(8, 4, device="meta")
+        d = distribute_tensor(t, mesh, [Shard(0)])
+        assert d._local_tensor.device.type == "meta"
+        assert d._local_tensor.shape == (8 // ws, 4)
+        e = torch.tanh(d + d)
+        assert e.placements[0].is_shard(0)
+        w = distribute_tensor(torch.empty(4, 6, device="meta"), mesh, [Shard(0)])
+        # S(0) x S(0)-on-K -> matmul contracts: local compute stays meta
+        m = d @ w.redistribute(placements=[Replicate()])
+        assert m.shape == (8, 6) and m._local_tensor.device.type == "meta"
+    assert cm.total == 0, f"meta ops must not communicate: {cm.get_comm_counts()}"
+
+
+def test_meta_device_dtensor():
+    spawn(2, _t_meta_device_dtensor)

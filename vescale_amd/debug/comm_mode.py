@@ -42,7 +42,15 @@ class CommDebugMode:
 
             def make(nm, fn):
                 def wrapper(*a, **k):
-                    self.comm_counts[nm] += 1
+                    # meta tensors short-circuit inside the helper (no c10d
+                    # call happens) — count only real communications
+                    import torch as _t
+
+                    if not any(
+                        isinstance(x, _t.Tensor) and x.device.type == "meta"
+                        for x in a
+                    ):
+                        self.comm_counts[nm] += 1
                     return fn(*a, **k)
 
                 return wrapper

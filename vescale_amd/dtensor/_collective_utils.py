@@ -41,11 +41,18 @@ def _get_op(reduce_op: str) -> dist.ReduceOp:
     }[reduce_op]
 
 
+def _is_meta(t: torch.Tensor) -> bool:
+    """Meta tensors short-circuit every collective: c10d has no meta
+    kernels, and meta-device DTensors exist exactly to propagate shapes
+    and placements with zero comms (reference test/dtensor/meta_device)."""
+    return t.device.type == "meta"
+
+
 def mesh_all_reduce(
     tensor: torch.Tensor, mesh: DeviceMesh, reduce_op: str = "sum", mesh_dim: int = 0,
     async_op: bool = False,
 ):
-    if mesh.size(mesh_dim) == 1:
+    if mesh.size(mesh_dim) == 1 or _is_meta(tensor):
         return None
     pg = mesh.get_group(mesh_dim)
     if reduce_op == "avg" and not hasattr(dist.ReduceOp, "AVG"):
@@ -59,7 +66,7 @@ def mesh_broadcast(
     tensor: torch.Tensor, mesh: DeviceMesh, mesh_dim: int = 0, src_mesh_rank: int = 0,
     async_op: bool = False,
 ):
-    if mesh.size(mesh_dim) == 1:
+    if mesh.size(mesh_dim) == 1 or _is_meta(tensor):
         return None
     pg = mesh.get_group(mesh_dim)
     src_global = dist.get_global_rank(pg, src_mesh_rank)
@@ -82,6 +89,12 @@ def mesh_all_gather(
     ws = mesh.size(mesh_dim)
     if ws == 1:
         return tensor
+    if _is_meta(tensor):
+        total = sum(sizes) if sizes is not None else ws * tensor.shape[gather_dim]
+        shape = list(tensor.shape)
+        shape[gather_dim] = total
+        out = tensor.new_empty(shape)
+        return (out, None) if async_op else out
     pg = mesh.get_group(mesh_dim)
     if sizes is None:
         t = tensor.contiguous()
@@ -129,6 +142,12 @@ def mesh_reduce_scatter(
     ws = mesh.size(mesh_dim)
     if ws == 1:
         return tensor
+    if _is_meta(tensor):
+        shape = list(tensor.shape)
+        assert shape[scatter_dim] % ws == 0
+        shape[scatter_dim] //= ws
+        out = tensor.new_empty(shape)
+        return (out, None) if async_op else out
     pg = mesh.get_group(mesh_dim)
     t = tensor.movedim(scatter_dim, 0).contiguous() if scatter_dim != 0 else tensor.contiguous()
     assert t.shape[0] % ws == 0, f"reduce_scatter dim {t.shape[0]} % {ws} != 0"
@@ -153,6 +172,8 @@ def mesh_scatter(
     src_mesh_rank: int = 0,
     async_op: bool = False,
 ):
+    if _is_meta(output):
+        return None
     pg = mesh.get_group(mesh_dim)
     src_global = dist.get_global_rank(pg, src_mesh_rank)
     if dist.get_rank() == src_global:
@@ -175,6 +196,8 @@ def mesh_scatter_ragged(
     all_to_all_single where only the root contributes input — one RCCL call,
     and on xGMI the root's 7 links fan out concurrently.
     """
+    if _is_meta(output):
+        return output
     pg = mesh.get_group(mesh_dim)
     ws = mesh.size(mesh_dim)
     my = mesh.get_local_rank(mesh_dim) if mesh.get_coordinate() is not None else -1
@@ -206,6 +229,8 @@ def mesh_all_to_all_single(
     ws = mesh.size(mesh_dim)
     if ws == 1:
         output.copy_(input.view_as(output))
+        return None
+    if _is_meta(output):
         return None
     pg = mesh.get_group(mesh_dim)
     if _supports_a2a(pg):
@@ -244,6 +269,8 @@ def mesh_all_to_all(
     if ws == 1:
         output_list[0].copy_(input_list[0])
         return None
+    if output_list and _is_meta(output_list[0]):
+        return None
     pg = mesh.get_group(mesh_dim)
     if _supports_a2a(pg):
         return dist.all_to_all(output_list, input_list, group=pg, async_op=async_op)
@@ -278,6 +305,8 @@ def broadcast_across_mesh(
 ):
     """P2P broadcast from a rank in one mesh to ranks of another (PP spec
     exchange).  Uses the default PG."""
+    if _is_meta(tensor):
+        return tensor
     me = dist.get_rank()
     reqs = []
     if me == src_rank:
