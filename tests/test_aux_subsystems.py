@@ -492,3 +492,50 @@ def test_emulator_tuner_model():
     ts = [predict_time_us(nb, 8, ALGO_RING, PROTO_SIMPLE, topo)
           for nb in (1 << 10, 1 << 20, 64 << 20)]
     assert ts[0] < ts[1] < ts[2]
+
+
+def _t_deferred_parallelize_materialize(rank, ws):
+    """The reference's combined init flow (dmodule/test_initialize.py):
+    deferred_init -> parallelize_module (still meta, zero comms) ->
+    materialize_dmodule allocates ONLY local shards and runs init."""
+    import torch.nn as nn
+
+    from vescale_amd.dmodule import parallelize_module
+    from vescale_amd.dtensor import DTensor, Replicate, Shard, init_device_mesh
+    from vescale_amd.initialize import deferred_init, is_deferred, materialize_dmodule
+
+    def build():
+        return nn.Sequential(nn.Linear(8, 16, bias=False), nn.Tanh(), nn.Linear(16, 8, bias=False))
+
+    mesh = init_device_mesh("cpu", (ws,))
+    m = deferred_init(build)
+    assert is_deferred(m)
+    plan = {
+        "parameter": {r"0.weight": [Shard(0)], r"2.weight": [Shard(1)]},
+        "forward": {"input": [[Replicate()]]},
+    }
+    m = parallelize_module(m, mesh, plan)
+    # still meta after parallelize: no memory, no comms
+    for p in m.parameters():
+        assert isinstance(p.data, DTensor) and p.data._local_tensor.is_meta
+
+    def det_init(mod):
+        with torch.no_grad():
+            for i, p in enumerate(mod.parameters()):
+                p.fill_(0.01 * (i + 1))
+
+    m = materialize_dmodule(m, device=torch.device("cpu"), init_weights=det_init)
+    for p in m.parameters():
+        assert not p.data._local_tensor.is_meta
+        assert p.data._local_tensor.device.type == "cpu"
+    # eager twin with the same init
+    ref = build()
+    det_init(ref)
+    x = torch.randn(4, 8, generator=torch.Generator().manual_seed(0))
+    out = m(x)
+    out = out.full_tensor() if isinstance(out, DTensor) else out
+    assert torch.allclose(out, ref(x), atol=1e-6)
+
+
+def test_deferred_parallelize_materialize():
+    spawn(2, _t_deferred_parallelize_materialize)

@@ -2,6 +2,7 @@ from .deferred_init import (
     deferred_init,
     is_deferred,
     materialize_dparameter,
+    materialize_dmodule,
     materialize_dtensor,
     materialize_module,
 )
@@ -11,5 +12,6 @@ __all__ = [
     "is_deferred",
     "materialize_dtensor",
     "materialize_dparameter",
+    "materialize_dmodule",
     "materialize_module",
 ]

@@ -102,3 +102,70 @@ def materialize_module(
     elif hasattr(module, "init_weights"):
         module.init_weights()
     return module
+
+
+def materialize_dmodule(
+    module: nn.Module,
+    *,
+    device: Optional[torch.device] = None,
+    init_weights: Optional[Callable[[nn.Module], None]] = None,
+) -> nn.Module:
+    """Materialize a module that was deferred_init-ed and THEN
+    parallelized (reference dmodule/test_initialize.py flow:
+    deferred_init -> parallelize_module -> reset_parameters).
+
+    Every DTensor parameter/buffer whose local shard is still meta gets
+    its LOCAL shard allocated on `device` (no full-tensor allocation at
+    any point); plain meta tensors are allocated whole.  Weight ties are
+    preserved.  Then `init_weights(module)` runs if given, else
+    `module.init_weights()` if present, else each submodule's
+    `reset_parameters()`.
+    """
+    dev = device or (
+        torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+    )
+    replaced = {}
+
+    def _alloc(t: torch.Tensor) -> torch.Tensor:
+        if isinstance(t, DTensor):
+            if not t._local_tensor.is_meta:
+                return t
+            local = torch.empty(
+                t._local_tensor.shape, dtype=t._local_tensor.dtype, device=dev
+            )
+            return DTensor(local, t._spec, requires_grad=t.requires_grad)
+        if t.is_meta:
+            return torch.empty(t.shape, dtype=t.dtype, device=dev)
+        return t
+
+    for mod in module.modules():
+        for pname, p in list(mod.named_parameters(recurse=False)):
+            if p is None:
+                continue
+            if id(p) in replaced:
+                mod._parameters[pname] = replaced[id(p)]
+                continue
+            newd = _alloc(p.data)
+            if newd is p.data:
+                continue
+            newp = nn.Parameter(newd, requires_grad=p.requires_grad)
+            replaced[id(p)] = newp
+            mod._parameters[pname] = newp
+        for bname, b in list(mod.named_buffers(recurse=False)):
+            if b is None:
+                continue
+            nb = _alloc(b)
+            if nb is not b:
+                mod._buffers[bname] = nb
+
+    module._vescale_deferred = False
+    if init_weights is not None:
+        init_weights(module)
+    elif hasattr(module, "init_weights"):
+        module.init_weights()
+    else:
+        for mod in module.modules():
+            rp = getattr(mod, "reset_parameters", None)
+            if callable(rp):
+                rp()
+    return module
