@@ -565,3 +565,80 @@ def _t_user_schedule(rank, ws):
 
 def test_user_defined_schedule():
     spawn(2, _t_user_schedule)
+
+
+# ---------------------------------------------------------------------------
+# direct p2p API (reference pipeline/backend/test_p2p_comm.py scenarios:
+# handshake vs known-shape, fused bidirectional, dtype fidelity, shape cache)
+# ---------------------------------------------------------------------------
+def _t_p2p_direct_api(rank, ws):
+    import vescale_amd.pipe.p2p_communication as p2p
+
+    dev = torch.device("cpu")
+    # 1) handshake path (no shape known at the receiver), non-fp32 dtype
+    if rank == 0:
+        t = torch.arange(12, dtype=torch.bfloat16).reshape(3, 4) + 1
+        p2p.send_forward(t, 1)
+        p2p.drain_send_reqs()
+    else:
+        r = p2p.recv_forward(0)
+        assert r.shape == (3, 4) and r.dtype == torch.bfloat16
+        assert torch.equal(r, torch.arange(12, dtype=torch.bfloat16).reshape(3, 4) + 1)
+
+    torch.distributed.barrier()
+    # 2) known-shape path (no handshake)
+    if rank == 0:
+        t = torch.full((2, 5), 7.0)
+        p2p.send_forward(t, 1, handshake=False)
+        p2p.drain_send_reqs()
+    else:
+        r = p2p.recv_forward(0, shape=(2, 5), dtype=torch.float32, device=dev)
+        assert torch.equal(r, torch.full((2, 5), 7.0))
+
+    torch.distributed.barrier()
+    # 3) backward direction with handshake
+    if rank == 1:
+        g = torch.randn(4, 3, generator=torch.Generator().manual_seed(1))
+        p2p.send_backward(g, 0)
+        p2p.drain_send_reqs()
+    else:
+        g = p2p.recv_backward(1)
+        want = torch.randn(4, 3, generator=torch.Generator().manual_seed(1))
+        assert torch.equal(g, want)
+
+    torch.distributed.barrier()
+    # channels re-handshake from here: the reuse cache pins one shape per
+    # (pg, peer, direction) channel (static-shape invariant), and scenario
+    # 4 reuses channels from 1/3 with different shapes
+    p2p.reset_shape_cache()
+    # 4) fused bidirectional pair, as the 1F1B steady state posts them:
+    # stage i runs send_forward_recv_backward while stage i+1 runs
+    # send_backward_recv_forward — both sides fused, matching post order
+    if rank == 0:
+        t = torch.ones(2, 2)
+        g = p2p.send_forward_recv_backward(t, 1)
+        assert torch.equal(g, torch.full((2, 2), 5.0))
+    else:
+        t = p2p.send_backward_recv_forward(torch.full((2, 2), 5.0), 0)
+        assert torch.equal(t, torch.ones(2, 2))
+        p2p.drain_send_reqs()
+
+    torch.distributed.barrier()
+    # 5) shape-cache reuse: same shape resent with VESCALE_REUSE_COMM_SHAPE
+    os.environ["VESCALE_REUSE_COMM_SHAPE"] = "1"
+    p2p.reset_shape_cache()
+    try:
+        for step in range(3):
+            if rank == 0:
+                p2p.send_forward(torch.full((3, 4), float(step)), 1)
+                p2p.drain_send_reqs()
+            else:
+                r = p2p.recv_forward(0)
+                assert torch.equal(r, torch.full((3, 4), float(step)))
+    finally:
+        del os.environ["VESCALE_REUSE_COMM_SHAPE"]
+        p2p.reset_shape_cache()
+
+
+def test_p2p_direct_api():
+    spawn(2, _t_p2p_direct_api)
