@@ -642,3 +642,58 @@ def _t_p2p_direct_api(rank, ws):
 
 def test_p2p_direct_api():
     spawn(2, _t_p2p_direct_api)
+
+
+# ---------------------------------------------------------------------------
+# forward-only (eval) projection of the schedules
+# ---------------------------------------------------------------------------
+def _t_pp_forward_only(rank, ws, sched, virtual_chunks):
+    from vescale_amd.engine import PipeEngine
+    from vescale_amd.pipe.pipe_stage import construct_pipeline_stage
+
+    n_mb, bs, d = 4, 8, 16
+    mods = _make_modules()
+    plan = PipelineParallelPlan(
+        num_stages=ws,
+        virtual_chunks=virtual_chunks,
+        schedule_type=PipelineScheduleType(sched),
+        split_method=PipelineSplitMethodType.UNIFORM,
+    )
+    stage = construct_pipeline_stage(mods, plan, rank)
+    engine = PipeEngine(stage, plan, loss_fn=_loss_fn, device=torch.device("cpu"))
+    torch.manual_seed(23)
+    x = torch.randn(bs, d)
+    y = torch.randn(bs, d)
+
+    # explicit forward_only kwarg
+    loss = engine.evaluate((x, y), n_mb)
+    # under torch.no_grad() the engine must auto-switch too
+    with torch.no_grad():
+        loss2 = engine.forward_backward((x, y), n_mb)
+
+    # reference eval loss on a single device
+    ref_model = nn.Sequential(*_make_modules())
+    torch.manual_seed(23)
+    xr = torch.randn(bs, d)
+    yr = torch.randn(bs, d)
+    with torch.no_grad():
+        ref = sum(
+            float(_loss_fn(ref_model(xm), ym))
+            for xm, ym in zip(torch.chunk(xr, n_mb), torch.chunk(yr, n_mb))
+        )
+    loss_holder = 0 if sched == "zero_bubble_v" and virtual_chunks == 2 else ws - 1
+    if rank == loss_holder:
+        assert loss is not None and loss2 is not None
+        assert abs(float(loss) - ref) < 1e-5, (float(loss), ref)
+        assert abs(float(loss2) - ref) < 1e-5
+    # NO grads were produced anywhere
+    for _, p in stage.named_parameters():
+        assert p.grad is None
+
+
+@pytest.mark.parametrize(
+    "sched,vc",
+    [("1f1b", 1), ("gpipe", 1), ("interleaved_1f1b", 2), ("zero_bubble_v", 2)],
+)
+def test_pp_forward_only(sched, vc):
+    spawn(2, _t_pp_forward_only, sched, vc)

@@ -40,10 +40,26 @@ class PipeEngine:
         self,
         minibatch: Optional[Tuple[torch.Tensor, torch.Tensor]],
         n_microbatches: int,
+        *,
+        forward_only: bool = False,
     ):
         """Returns the summed minibatch loss on the LAST stage (None
-        elsewhere); gradients are left on stage parameters."""
-        return self.schedule_engine.execute(minibatch, n_microbatches)
+        elsewhere); gradients are left on stage parameters.
+
+        forward_only=True (or calling under torch.no_grad()) runs the
+        eval projection of the schedule: forwards and activation sends
+        only, no grads (reference _schedules forward_only)."""
+        return self.schedule_engine.execute(
+            minibatch, n_microbatches, forward_only=forward_only
+        )
+
+    def evaluate(
+        self,
+        minibatch: Optional[Tuple[torch.Tensor, torch.Tensor]],
+        n_microbatches: int,
+    ):
+        """Validation-loop convenience: forward-only, no autograd state."""
+        return self.forward_backward(minibatch, n_microbatches, forward_only=True)
 
     def parameters(self):
         return self.module.parameters()

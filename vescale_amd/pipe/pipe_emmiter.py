@@ -146,17 +146,43 @@ class ScheduleEngine:
         return self.next_rank
 
     # ------------------------------------------------------------------
+    @staticmethod
+    def _strip_backward(sched: List[Instr]) -> List[Instr]:
+        """Forward-only projection of a schedule (reference
+        looping_bfs.py:666,982 forward_only): drop every backward
+        instruction, reduce fused ops to their forward half.  The
+        resulting per-rank streams stay post-order matched because every
+        rank drops the same halves."""
+        out: List[Instr] = []
+        for ins in sched:
+            if ins.kind in ("BWD", "BWD_B", "BWD_W", "SEND_BWD", "RECV_BWD"):
+                continue
+            if ins.kind == "SEND_FWD_RECV_BWD":
+                out.append(Instr("SEND_FWD", ins.microbatch, ins.chunk))
+            elif ins.kind == "SEND_BWD_RECV_FWD":
+                out.append(Instr("RECV_FWD", ins.microbatch2, ins.chunk))
+            else:
+                out.append(ins)
+        return out
+
     def execute(
         self,
         minibatch: Optional[Tuple[torch.Tensor, torch.Tensor]],
         n_microbatches: int,
         *,
         grad_scale: Optional[float] = None,
+        forward_only: bool = False,
     ):
         """Run one forward_backward over the minibatch split into
         n_microbatches.  First stage consumes inputs; last stage consumes
         targets + computes loss via loss_fn(output, target) (mean over
-        microbatches)."""
+        microbatches).
+
+        forward_only=True (or calling under torch.no_grad(), reference
+        looping_bfs.py:892) runs just the forward halves — the PP eval /
+        validation loop."""
+        forward_only = forward_only or not torch.is_grad_enabled()
+        self._forward_only = forward_only
         xs = ys = None
         if minibatch is not None:
             x, y = minibatch
@@ -168,28 +194,39 @@ class ScheduleEngine:
         self._recv_grads.clear()
         scale = grad_scale if grad_scale is not None else 1.0 / n_microbatches
 
-        for ins in self.build_schedule(n_microbatches):
+        sched = self.build_schedule(n_microbatches)
+        if forward_only:
+            sched = self._strip_backward(sched)
+        for ins in sched:
             kind, m, ck = ins.kind, ins.microbatch, ins.chunk
             if kind == "RECV_FWD":
                 with ndtimeit_p2p(ndm.RECV_FORWARD, self._in_peer(ck)):
                     t = p2p.recv_forward(
                         self._in_peer(ck), self._pg_for(kind, ck), device=self.device
                     )
-                self._inputs[(ck, m)] = t.requires_grad_(True)
+                self._inputs[(ck, m)] = t if forward_only else t.requires_grad_(True)
             elif kind == "FWD":
                 with ndtimeit(ndm.FORWARD_COMPUTE):
-                    self._fwd(ck, m, xs, ys, scale)
+                    if forward_only:
+                        with torch.no_grad():
+                            self._fwd(ck, m, xs, ys, scale)
+                    else:
+                        self._fwd(ck, m, xs, ys, scale)
             elif kind == "SEND_FWD":
                 peer = self._out_peer(ck)
                 if peer == "local":
                     # ZB-V rank P-1: chunk 0 output feeds own chunk 1
+                    h = self._outputs[(ck, m)].detach()
                     self._inputs[(ck + 1, m)] = (
-                        self._outputs[(ck, m)].detach().requires_grad_(True)
+                        h if forward_only else h.requires_grad_(True)
                     )
                 else:
                     p2p.send_forward(
                         self._outputs[(ck, m)].detach(), peer, self._pg_for(kind, ck)
                     )
+                if forward_only and not self._is_last_global(ck):
+                    # no BWD will pop it — free the activation now
+                    self._outputs.pop((ck, m), None)
             elif kind == "SEND_FWD_RECV_BWD":
                 g = p2p.send_forward_recv_backward(
                     self._outputs[(ck, m)].detach(), self._out_peer(ck), self.pg,
