@@ -292,3 +292,25 @@ def test_ddp_tied_weight_parity():
 
 def test_ddp_zero2_tied_weight_parity():
     spawn(2, _t_tied_weight, 4, True)
+
+
+def _t_start_grad_sync_api(rank, ws):
+    """Explicit start_grad_sync launches pending bucket syncs; parity with
+    the implicit hook-driven path (reference ddp :277)."""
+    torch.manual_seed(3)
+    net = Net()
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("DP",))
+    ddp = DDP(net, mesh, use_distributed_optimizer=False)
+    opt = BasicOptimizer(torch.optim.AdamW(net.parameters(), lr=1e-2), [ddp])
+    x = _data(ws, 1)[0]
+    loss = ddp(torch.chunk(x, ws)[rank]).pow(2).mean()
+    loss.backward()
+    ddp.start_grad_sync()   # idempotent with whatever the hooks launched
+    ddp.finish_grad_sync()
+    opt.step()
+    sd = ddp.state_dict_for_save_checkpoint()
+    assert set(sd) == set(net.state_dict())
+
+
+def test_start_grad_sync_api():
+    spawn(2, _t_start_grad_sync_api)

@@ -122,11 +122,24 @@ class DistributedDataParallel(nn.Module):
         for p in self.module.parameters():
             p.grad = None
 
+    def start_grad_sync(self):
+        """Kick off the DP sync for any bucket not yet launched (reference
+        ddp :277 — normally the per-bucket hooks start syncs as grads
+        complete; this forces the remainder, e.g. before an early
+        finish_grad_sync outside a backward)."""
+        for gb in self.grad_buffers.values():
+            gb.start_grad_sync()
+
     def finish_grad_sync(self):
         for gb in self.grad_buffers.values():
             gb.finish_grad_sync()
 
     def state_dict(self, *args, **kwargs):
+        return self.module.state_dict(*args, **kwargs)
+
+    def state_dict_for_save_checkpoint(self, *args, **kwargs):
+        """Reference :324 naming parity — the checkpointable state dict
+        (same as state_dict: params are DTensors with their placements)."""
         return self.module.state_dict(*args, **kwargs)
 
     def load_state_dict(self, *args, **kwargs):

@@ -171,6 +171,11 @@ class GradBuffer:
         for b in self.buckets:
             b.reset()
 
+    def start_grad_sync(self):
+        for b in self.buckets:
+            if not b.comm_issued:
+                b.start_grad_sync()
+
     def finish_grad_sync(self):
         for b in self.buckets:
             b.finish_grad_sync()
