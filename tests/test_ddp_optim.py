@@ -314,3 +314,29 @@ def _t_start_grad_sync_api(rank, ws):
 
 def test_start_grad_sync_api():
     spawn(2, _t_start_grad_sync_api)
+
+
+def _t_do_accessors(rank, ws):
+    """DistributedOptimizer accessor parity: master shards, main grads,
+    one-off clip (reference :1223-1276)."""
+    torch.manual_seed(3)
+    net = Net()
+    mesh = init_device_mesh("cpu", (ws,), mesh_dim_names=("DP",))
+    ddp = DDP(net, mesh, use_distributed_optimizer=True)
+    opt = DistributedOptimizer(torch.optim.AdamW(net.parameters(), lr=1e-2), [ddp])
+    masters = opt.get_parameters()
+    assert masters and all(m.dtype == torch.float32 for m in masters)
+    x = _data(ws, 1)[0]
+    ddp(torch.chunk(x, ws)[rank]).pow(2).mean().backward()
+    ddp.finish_grad_sync()
+    opt._copy_model_grads_to_main_grads()
+    grads = opt.get_main_grads_for_grad_norm()
+    assert grads and all(g is not None for g in grads)
+    norm = opt.clip_grad_norm(1e-9)  # tiny clip scales everything
+    assert float(norm) > 0
+    for g in opt.get_main_grads_for_grad_norm():
+        assert g.abs().max() <= 1e-9 + 1e-12
+
+
+def test_do_accessors():
+    spawn(2, _t_do_accessors)

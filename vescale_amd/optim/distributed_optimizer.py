@@ -227,6 +227,22 @@ class DistributedOptimizer:
     def param_groups(self):
         return self.optimizer.param_groups
 
+    # -- reference API parity (distributed_optimizer.py:1223-1276) -----
+    def get_parameters(self) -> List[torch.Tensor]:
+        """The fp32 master shards this rank optimizes."""
+        return [s[7] for s in self._slices]
+
+    def get_main_grads_for_grad_norm(self) -> List[torch.Tensor]:
+        return [s[7].grad for s in self._slices if s[7].grad is not None]
+
+    def clip_grad_norm(self, clip_grad: float):
+        """One-off clip of the current main grads (step() already clips
+        when constructed with clip_grad>0)."""
+        pgs = [self.dp_group] + list(self.extra_norm_pgs)
+        return clip_grad_norm_fp32(
+            self.get_main_grads_for_grad_norm(), clip_grad, pgs
+        )
+
 
 def _local(p) -> torch.Tensor:
     d = p.data
