@@ -24,6 +24,9 @@ from .dtensor import (  # noqa: F401
     normalize_placements,
     redistribute_dtensor,
     to_local,
+    vescale_all_gather,
+    vescale_all_reduce,
+    vescale_reduce_scatter,
 )
 
 # top-level API parity with the reference's `import vescale` surface
