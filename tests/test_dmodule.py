@@ -274,3 +274,34 @@ def _t_state_dict_saveload(rank, ws):
 
 def test_state_dict_saveload():
     spawn(2, _t_state_dict_saveload)
+
+
+def _t_grad_placement_enforce(rank, ws):
+    """PlacementsInterface(grad=...) forces the parameter's gradient into
+    a chosen placement via a hook (reference PostHookGrad): here a
+    row-parallel weight's naturally-Partial grad is reduced to Replicate
+    at backward time instead of at finish_grad_sync."""
+    from vescale_amd.dmodule import PlacementsInterface as PI
+    from vescale_amd.dtensor import Partial, init_device_mesh
+
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(6)
+    net = nn.Linear(8, 8, bias=False)
+    torch.manual_seed(6)
+    ref = nn.Linear(8, 8, bias=False)
+    plan = {
+        "parameter": {r"weight": PI([Shard(1)], grad=[Replicate()])},
+        "forward": {"input": [[Replicate()]], "output": [[Replicate()]]},
+    }
+    net = parallelize_module(net, mesh, plan)
+    x = torch.randn(4, 8, generator=torch.Generator().manual_seed(7))
+    out = net(x)
+    (out.to_local() if isinstance(out, DTensor) else out).pow(2).mean().backward()
+    g = net.weight.grad
+    assert isinstance(g, DTensor) and g.placements[0].is_replicate(), g.placements
+    ref(x).pow(2).mean().backward()
+    assert torch.allclose(g.full_tensor(), ref.weight.grad, atol=1e-6)
+
+
+def test_grad_placement_enforce():
+    spawn(2, _t_grad_placement_enforce)

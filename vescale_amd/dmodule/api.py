@@ -19,7 +19,7 @@ from __future__ import annotations
 import re
 import os
 from dataclasses import dataclass
-from typing import Any, Dict, Sequence, Union
+from typing import Any, Dict, Optional, Sequence, Union
 
 import torch
 import torch.nn as nn
@@ -31,19 +31,38 @@ from ._hook import install_forward_hooks
 
 @dataclass
 class PlacementsInterface:
-    """Placement spec + conversion flags (reference PlacementsInterface)."""
+    """Placement spec + conversion flags (reference PlacementsInterface:
+    placements/run_check/grad — async_op/defer_reshard are not needed
+    here: redistribute is eager, and Partial arithmetic stays deferred by
+    the default dispatch rules)."""
 
     placements: Sequence[Placement]
     run_check: bool = False
     # if the local tensor is ALREADY laid out as a shard, wrap with
     # from_local instead of slicing a replicated global tensor
     is_local: bool = False
+    # placement to ENFORCE on this parameter's gradient (a hook
+    # redistributes the incoming grad; reference PostHookGrad)
+    grad: Optional[Sequence[Placement]] = None
 
     @classmethod
     def normalize(cls, v) -> "PlacementsInterface":
         if isinstance(v, PlacementsInterface):
             return v
         return cls(placements=tuple(v))
+
+
+def _install_grad_placement_hook(param: nn.Parameter, grad_placements):
+    """Enforce a placement on the param's incoming gradient (reference
+    PostHookGrad): e.g. keep an SP weight grad Partial for the bucketed
+    sync, or force Replicate for an immediately-consumed grad."""
+
+    def hook(g):
+        if isinstance(g, DTensor) and tuple(g.placements) != tuple(grad_placements):
+            return g.redistribute(placements=list(grad_placements))
+        return g
+
+    param.register_hook(hook)
 
 
 def _match_plan(fqn: str, plan: Dict[str, Any]):
@@ -87,6 +106,8 @@ def parallelize_module(
             else:
                 d = distribute_tensor(p.data, device_mesh, spec.placements)
             newp = nn.Parameter(d, requires_grad=p.requires_grad)
+            if spec.grad is not None:
+                _install_grad_placement_hook(newp, tuple(spec.grad))
             replaced[id(p)] = newp
             mod._parameters[pname] = newp
         for bname, b in list(mod.named_buffers(recurse=False)):
