@@ -305,3 +305,39 @@ def _t_grad_placement_enforce(rank, ws):
 
 def test_grad_placement_enforce():
     spawn(2, _t_grad_placement_enforce)
+
+
+def _t_dict_input_plan_binding(rank, ws):
+    """Dict input plans bind by PARAMETER NAME across calling conventions
+    (reference test_fwd_plan.py dict_fwd_plan): positional, keyword,
+    kw-only, and default-valued args all convert."""
+    from vescale_amd.dtensor import init_device_mesh
+
+    mesh = init_device_mesh("cpu", (ws,))
+
+    class Pos(nn.Module):
+        def forward(self, a, b=None):
+            return a if b is None else a + b
+
+    class KwOnly(nn.Module):
+        def forward(self, *, a):
+            return a
+
+    plan = {"parameter": {}, "forward": {"input": {"a": [Shard(0)]}}}
+    a = torch.ones(2 * ws, 2)
+
+    m = parallelize_module(Pos(), mesh, dict(plan))
+    out = m(a)                      # positional
+    assert isinstance(out, DTensor) and out.placements[0].is_shard(0)
+    out = m(a=a)                    # keyword
+    assert isinstance(out, DTensor) and out.placements[0].is_shard(0)
+    out = m(a, b=torch.zeros(2 * ws, 2))  # unplanned arg passes through (replicate)
+    assert isinstance(out, DTensor)
+
+    m2 = parallelize_module(KwOnly(), mesh, dict(plan))
+    out = m2(a=a)                   # kw-only binding
+    assert isinstance(out, DTensor) and out.placements[0].is_shard(0)
+
+
+def test_dict_input_plan_binding():
+    spawn(2, _t_dict_input_plan_binding)
