@@ -697,3 +697,40 @@ def _t_pp_forward_only(rank, ws, sched, virtual_chunks):
 )
 def test_pp_forward_only(sched, vc):
     spawn(2, _t_pp_forward_only, sched, vc)
+
+
+def _t_single_stage_engine(rank, ws):
+    """Degenerate P=1 pipeline: the engine runs pure compute, no p2p
+    (reference pipeline/api/test_pipe_single_stage_ops.py)."""
+    from vescale_amd.engine import PipeEngine
+    from vescale_amd.pipe.pipe_stage import construct_pipeline_stage
+
+    torch.manual_seed(1)
+    mods = _make_modules()
+    plan = PipelineParallelPlan(
+        num_stages=1,
+        virtual_chunks=1,
+        schedule_type=PipelineScheduleType("1f1b"),
+        split_method=PipelineSplitMethodType.UNIFORM,
+    )
+    stage = construct_pipeline_stage(mods, plan, rank)
+    eng = PipeEngine(stage, plan, loss_fn=_loss_fn, device=torch.device("cpu"))
+    torch.manual_seed(23)
+    x = torch.randn(8, 16)
+    y = torch.randn(8, 16)
+    loss = eng.forward_backward((x, y), 4)
+    # equals a plain n_mb-chunked loop on the same module stack
+    ref_model = nn.Sequential(*_make_modules())
+    torch.manual_seed(23)
+    xr = torch.randn(8, 16)
+    yr = torch.randn(8, 16)
+    ref = sum(float(_loss_fn(ref_model(xm), ym))
+              for xm, ym in zip(torch.chunk(xr, 4), torch.chunk(yr, 4)))
+    assert abs(float(loss) - ref) < 1e-5
+    for _, p in stage.named_parameters():
+        assert p.grad is not None
+    assert abs(float(eng.evaluate((x, y), 4)) - ref) < 1e-5
+
+
+def test_single_stage_engine():
+    spawn(1, _t_single_stage_engine)
