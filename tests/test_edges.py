@@ -335,3 +335,24 @@ def _t_meta_device_dtensor(rank, ws):
 
 def test_meta_device_dtensor():
     spawn(2, _t_meta_device_dtensor)
+
+
+def _t_shard_to_shard_single_a2a(rank, ws):
+    """Shard(a) -> Shard(b) is ONE all_to_all (the Ulysses
+    head<->sequence switch must be an all-to-all transition, never
+    allgather+reshard)."""
+    from vescale_amd.debug import CommDebugMode
+    from vescale_amd.dtensor import distribute_tensor
+
+    mesh = init_device_mesh("cpu", (ws,))
+    g = torch.arange(4 * 6, dtype=torch.float32).reshape(4, 6)
+    d = distribute_tensor(g, mesh, [Shard(0)])
+    with CommDebugMode() as cm:
+        e = d.redistribute(placements=[Shard(1)])
+    assert cm.get_comm_counts() == {"mesh_all_to_all": 1}, cm.get_comm_counts()
+    assert e.placements[0].is_shard(1)
+    assert torch.equal(e.full_tensor(), g)
+
+
+def test_shard_to_shard_single_a2a():
+    spawn(2, _t_shard_to_shard_single_a2a)
