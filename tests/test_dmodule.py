@@ -430,3 +430,31 @@ def _t_ring_sdpa(rank, ws):
 
 def test_ring_sdpa():
     spawn(2, _t_ring_sdpa)
+
+
+def _t_ring_sdpa_ws4(rank, ws):
+    """Ring attention at p=4 (multi-hop ring, mixed masked/unmasked)."""
+    from vescale_amd.dmodule.ring_attention import ring_sdpa
+    from vescale_amd.dtensor import distribute_tensor, init_device_mesh
+    import torch.nn.functional as F
+
+    mesh = init_device_mesh("cpu", (ws,))
+    B, H, S, D = 1, 2, 16, 8
+    g = torch.Generator().manual_seed(5)
+    qg = torch.randn(B, H, S, D, generator=g, requires_grad=True)
+    kg = torch.randn(B, H, S, D, generator=g, requires_grad=True)
+    vg = torch.randn(B, H, S, D, generator=g, requires_grad=True)
+    q = distribute_tensor(qg.detach().clone().requires_grad_(True), mesh, [Shard(2)])
+    k = distribute_tensor(kg.detach().clone().requires_grad_(True), mesh, [Shard(2)])
+    v = distribute_tensor(vg.detach().clone().requires_grad_(True), mesh, [Shard(2)])
+    out = ring_sdpa(q, k, v, is_causal=True)
+    ref = F.scaled_dot_product_attention(qg, kg, vg, is_causal=True)
+    assert torch.allclose(out.full_tensor(), ref, atol=1e-5)
+    out.to_local().pow(2).sum().backward()
+    ref.pow(2).sum().backward()
+    gq = q.grad.full_tensor() if hasattr(q.grad, "full_tensor") else q.grad
+    assert torch.allclose(gq, qg.grad, atol=1e-5)
+
+
+def test_ring_sdpa_ws4():
+    spawn(4, _t_ring_sdpa_ws4)
