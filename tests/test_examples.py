@@ -81,3 +81,19 @@ def test_example_checkpoint_reshard(tmp_path):
     out = _torchrun(2, "examples/checkpoint_reshard.py", "--load", ck,
                     "--steps", "1", port=29717)
     assert "loss" in out.lower() or "loaded" in out.lower() or out
+
+
+def test_example_long_context_sp():
+    # ring and ulysses must print IDENTICAL losses (both exact)
+    out_r = _torchrun(
+        2, "examples/long_context_sp.py", "--algo", "ring", "--steps", "2",
+        port=29717,
+    )
+    out_u = _torchrun(
+        2, "examples/long_context_sp.py", "--algo", "ulysses", "--steps", "2",
+        port=29718,
+    )
+    lr = [l.split("loss")[-1] for l in out_r.splitlines() if "loss" in l]
+    lu = [l.split("loss")[-1] for l in out_u.splitlines() if "loss" in l]
+    assert lr and lr == lu, (lr, lu)
+    assert "DONE" in out_r and "DONE" in out_u

@@ -61,9 +61,10 @@ def _shift(t: torch.Tensor, group, *, forward: bool) -> torch.Tensor:
     nxt = dist.get_global_rank(group, (me + 1) % ws)
     prv = dist.get_global_rank(group, (me - 1) % ws)
     dst, src = (nxt, prv) if forward else (prv, nxt)
-    recv = torch.empty_like(t)
+    t = t.contiguous()
+    recv = torch.empty_like(t)  # empty_like of a contiguous t IS contiguous
     ops = [
-        dist.P2POp(dist.isend, t.contiguous(), peer=dst, group=group),
+        dist.P2POp(dist.isend, t, peer=dst, group=group),
         dist.P2POp(dist.irecv, recv, peer=src, group=group),
     ]
     for w in dist.batch_isend_irecv(ops):
