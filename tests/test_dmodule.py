@@ -458,3 +458,33 @@ def _t_ring_sdpa_ws4(rank, ws):
 
 def test_ring_sdpa_ws4():
     spawn(4, _t_ring_sdpa_ws4)
+
+
+def _t_sp_on_submesh(rank, ws):
+    """Ring + Ulysses over the SP dim of a 2-D (DP, SP) mesh: subgroup
+    global-rank mapping in the ring shift, per-DP-group independent data."""
+    from vescale_amd.dmodule.ring_attention import ring_sdpa
+    from vescale_amd.dmodule.ulysses import ulysses_sdpa
+    from vescale_amd.dtensor import distribute_tensor, init_device_mesh
+    import torch.nn.functional as F
+
+    mesh = init_device_mesh("cpu", (2, 2), mesh_dim_names=("DP", "SP"))
+    sp = mesh["SP"]
+    dp_idx = mesh.get_coordinate()[0]
+    B, H, S, D = 1, 2, 8, 4
+    g = torch.Generator().manual_seed(100 + dp_idx)  # different per DP group
+    qg = torch.randn(B, H, S, D, generator=g)
+    kg = torch.randn(B, H, S, D, generator=g)
+    vg = torch.randn(B, H, S, D, generator=g)
+    q = distribute_tensor(qg, sp, [Shard(2)])
+    k = distribute_tensor(kg, sp, [Shard(2)])
+    v = distribute_tensor(vg, sp, [Shard(2)])
+    ref = F.scaled_dot_product_attention(qg, kg, vg, is_causal=True)
+    out_r = ring_sdpa(q, k, v, is_causal=True)
+    assert torch.allclose(out_r.full_tensor(), ref, atol=1e-5)
+    out_u = ulysses_sdpa(q, k, v, is_causal=True)
+    assert torch.allclose(out_u.full_tensor(), ref, atol=1e-5)
+
+
+def test_sp_on_submesh():
+    spawn(4, _t_sp_on_submesh)
