@@ -539,3 +539,35 @@ def _t_deferred_parallelize_materialize(rank, ws):
 
 def test_deferred_parallelize_materialize():
     spawn(2, _t_deferred_parallelize_materialize)
+
+
+def _t_deferred_tied_weights(rank, ws):
+    """Weight ties survive deferred_init -> parallelize -> materialize
+    (one allocation, both modules point at it)."""
+    import torch.nn as nn
+
+    from vescale_amd.dmodule import parallelize_module
+    from vescale_amd.dtensor import init_device_mesh
+    from vescale_amd.initialize import deferred_init, materialize_dmodule
+
+    class Tied(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb = nn.Embedding(10, 8)
+            self.out = nn.Linear(8, 10, bias=False)
+            self.out.weight = self.emb.weight
+
+    mesh = init_device_mesh("cpu", (ws,))
+    m = deferred_init(Tied)
+    m = parallelize_module(m, mesh, {"parameter": {}, "forward": {}})
+    assert m.out.weight is m.emb.weight
+    m = materialize_dmodule(
+        m, device=torch.device("cpu"),
+        init_weights=lambda mod: [p.data.fill_(0.5) for p in mod.parameters()],
+    )
+    assert m.out.weight is m.emb.weight
+    assert not m.out.weight.data._local_tensor.is_meta
+
+
+def test_deferred_tied_weights():
+    spawn(2, _t_deferred_tied_weights)
