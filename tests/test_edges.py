@@ -356,3 +356,29 @@ def _t_shard_to_shard_single_a2a(rank, ws):
 
 def test_shard_to_shard_single_a2a():
     spawn(2, _t_shard_to_shard_single_a2a)
+
+
+def _t_mesh_unpickle_lazy_groups(rank, ws):
+    """Unpickling a DTensor when NO identical live mesh exists: groups
+    rebuild lazily on first use (lockstep across ranks)."""
+    import io
+
+    from vescale_amd.dtensor import device_mesh as dm
+    from vescale_amd.dtensor import distribute_tensor
+
+    mesh = init_device_mesh("cpu", (ws,))
+    d = distribute_tensor(torch.arange(8, dtype=torch.float32), mesh, [Shard(0)])
+    buf = io.BytesIO()
+    torch.save(d, buf)
+    # simulate a fresh process: registry wiped, original mesh forgotten
+    dm._LIVE_MESHES.clear()
+    buf.seek(0)
+    d2 = torch.load(buf, weights_only=False)
+    assert d2.device_mesh._dim_groups is None  # not yet rebound
+    # first collective use rebuilds the dim groups in lockstep
+    full = d2.full_tensor()
+    assert torch.equal(full, torch.arange(8, dtype=torch.float32))
+
+
+def test_mesh_unpickle_lazy_groups():
+    spawn(2, _t_mesh_unpickle_lazy_groups)
