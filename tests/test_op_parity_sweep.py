@@ -52,6 +52,16 @@ UNARY = [
     ("type_f64", lambda t: t.double()),
     ("gelu", F.gelu),
     ("layer_norm", lambda t: F.layer_norm(t, (8,))),
+    ("logaddexp_shift", lambda t: torch.logaddexp(t, t - 1)),
+    ("masked_fill", lambda t: t.masked_fill(t > 0.5, 0.0)),
+    ("repeat_interleave_d1", lambda t: t.repeat_interleave(2, dim=1)),
+    ("flip_d1", lambda t: t.flip(1)),
+    ("roll_d1", lambda t: t.roll(3, dims=1)),
+    ("cummax_d1", lambda t: t.cummax(1)[0]),
+    ("logsumexp_d1", lambda t: t.logsumexp(1)),
+    ("clip_grad_style_sq", lambda t: (t * t).sum(-1, keepdim=True).sqrt()),
+    ("erf", torch.erf),
+    ("sinh_cosh", lambda t: t.sinh() + t.cosh()),
 ]
 
 BINARY = [
