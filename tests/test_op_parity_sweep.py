@@ -128,6 +128,12 @@ GRAD_OPS = [
     ("layer_norm", lambda t: F.layer_norm(t, (8,)).pow(2).sum()),
     ("sliced", lambda t: t[:, 2:6].sum()),
     ("gelu", lambda t: F.gelu(t).sum()),
+    ("logsumexp", lambda t: t.logsumexp(1).sum()),
+    ("masked_fill", lambda t: t.masked_fill(t > 0.5, 0.0).pow(2).sum()),
+    ("cumsum_tanh", lambda t: t.cumsum(1).tanh().sum()),
+    ("flip_mul", lambda t: (t.flip(1) * t).sum()),
+    ("softplus", lambda t: F.softplus(t).sum()),
+    ("var", lambda t: t.var(1).sum()),
 ]
 
 
