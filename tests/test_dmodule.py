@@ -488,3 +488,43 @@ def _t_sp_on_submesh(rank, ws):
 
 def test_sp_on_submesh():
     spawn(4, _t_sp_on_submesh)
+
+
+def _t_sp_shape_sweep(rank, ws):
+    """Ring + Ulysses across diverse (B,H,S,D,causal,gqa) shapes."""
+    from vescale_amd.dmodule.ring_attention import ring_sdpa
+    from vescale_amd.dmodule.ulysses import ulysses_sdpa
+    from vescale_amd.dtensor import distribute_tensor, init_device_mesh
+    import torch.nn.functional as F
+
+    mesh = init_device_mesh("cpu", (ws,))
+    cases = [
+        (1, 2, 4, 8, False, 1),
+        (3, 4, 8, 16, True, 1),
+        (2, 8, 16, 4, True, 2),    # GQA 8q/4kv
+        (1, 2, 32, 32, False, 2),  # GQA 2q/1kv
+        (2, 6, 12, 8, True, 3),    # GQA 6q/2kv, S not power of 2
+    ]
+    for idx, (B, H, S, D, causal, gqa) in enumerate(cases):
+        if H % ws or S % ws or (H // gqa) % ws:
+            continue
+        g = torch.Generator().manual_seed(50 + idx)
+        qg = torch.randn(B, H, S, D, generator=g)
+        kg = torch.randn(B, H // gqa, S, D, generator=g)
+        vg = torch.randn(B, H // gqa, S, D, generator=g)
+        ref = F.scaled_dot_product_attention(
+            qg, kg.repeat_interleave(gqa, 1), vg.repeat_interleave(gqa, 1),
+            is_causal=causal,
+        )
+        q = distribute_tensor(qg, mesh, [Shard(2)])
+        k = distribute_tensor(kg, mesh, [Shard(2)])
+        v = distribute_tensor(vg, mesh, [Shard(2)])
+        out_r = ring_sdpa(q, k, v, is_causal=causal)
+        assert torch.allclose(out_r.full_tensor(), ref, atol=1e-5), (idx, "ring")
+        if (H // gqa) % ws == 0 and H % ws == 0:
+            out_u = ulysses_sdpa(q, k, v, is_causal=causal)
+            assert torch.allclose(out_u.full_tensor(), ref, atol=1e-5), (idx, "uly")
+
+
+def test_sp_shape_sweep():
+    spawn(2, _t_sp_shape_sweep)

@@ -64,9 +64,11 @@ def ulysses_sdpa(
     kh = k.redistribute(placements=[Shard(head_dim)])
     vh = v.redistribute(placements=[Shard(head_dim)])
 
+    ql, kl, vl = qh.to_local(), kh.to_local(), vh.to_local()
     o = F.scaled_dot_product_attention(
-        qh.to_local(), kh.to_local(), vh.to_local(),
+        ql, kl, vl,
         is_causal=is_causal, scale=scale,
+        enable_gqa=(ql.shape[1] != kl.shape[1]),
     )
     oh = DTensor.from_local(o, mesh, [Shard(head_dim)])
     # head -> seq: the fourth all_to_all
