@@ -155,3 +155,41 @@ def _t_hf_mixtral_tp_parity(rank, ws):
 def test_hf_mixtral_tp_parity():
     pytest.importorskip("transformers")
     spawn(2, _t_hf_mixtral_tp_parity)
+
+
+def _t_hf_modeloutput_dict_plan(rank, ws):
+    """Dict output plans convert fields of a REAL HF ModelOutput return
+    (reference test_obj_return.py's dict_like/mixed cases)."""
+    import torch.nn as nn
+    from transformers.modeling_outputs import BaseModelOutput
+
+    from vescale_amd.dmodule import parallelize_module
+    from vescale_amd.dtensor import DTensor, Replicate, Shard, init_device_mesh
+
+    class Net(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = nn.Linear(8, 8)
+
+        def forward(self, x):
+            h = self.fc(x)
+            return BaseModelOutput(last_hidden_state=h, hidden_states=(h + 1,))
+
+    mesh = init_device_mesh("cpu", (ws,))
+    net = parallelize_module(
+        Net(), mesh,
+        {"parameter": {}, "forward": {"output": {"last_hidden_state": [Shard(0)]}}},
+    )
+    out = net(torch.randn(2 * ws, 8))
+    assert isinstance(out, BaseModelOutput)
+    assert isinstance(out.last_hidden_state, DTensor)
+    assert out.last_hidden_state.placements[0].is_shard(0)
+    # unplanned field untouched by the plan: params default to Replicate
+    # DTensors, so it's a DTensor too — but NOT redistributed to Shard
+    h0 = out.hidden_states[0]
+    assert isinstance(h0, DTensor) and h0.placements[0].is_replicate()
+
+
+def test_hf_modeloutput_dict_plan():
+    pytest.importorskip("transformers")
+    spawn(2, _t_hf_modeloutput_dict_plan)
