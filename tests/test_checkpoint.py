@@ -472,3 +472,40 @@ def test_mem_checkpoint_server_roundtrip(tmp_path):
             assert torch.equal(a, b)
     finally:
         srv.stop()
+
+
+def _t_tied_model_ckpt(rank, ws, path):
+    """A model with tied weights saves/loads cleanly: the tie means ONE
+    logical tensor in the state dict (both fqns map to the same storage);
+    on load both modules see the restored values."""
+    import torch.nn as nn
+
+    import vescale_amd.checkpoint as ckpt
+    from vescale_amd.dmodule import parallelize_module
+    from vescale_amd.dtensor import Shard as S
+
+    class Tied(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb = nn.Embedding(8, 4)
+            self.out = nn.Linear(4, 8, bias=False)
+            self.out.weight = self.emb.weight
+
+    mesh = init_device_mesh("cpu", (ws,))
+    torch.manual_seed(21)
+    m = parallelize_module(Tied(), mesh, {"parameter": {r".*weight": [S(0)]}, "forward": {}})
+    assert m.out.weight is m.emb.weight
+    want = m.emb.weight.data._local_tensor.clone()
+    ckpt.save(path, {"model": m})
+    with torch.no_grad():
+        for p in m.parameters():
+            p.add_(7.0)
+    ckpt.load(path, {"model": m})
+    assert torch.allclose(m.emb.weight.data._local_tensor, want)
+    assert torch.allclose(m.out.weight.data._local_tensor, want)
+    assert m.out.weight is m.emb.weight  # tie survives the load
+
+
+def test_tied_model_checkpoint():
+    with tempfile.TemporaryDirectory() as td:
+        spawn(2, _t_tied_model_ckpt, td)
